@@ -1,0 +1,133 @@
+"""Data pipeline.
+
+Reference surface: ``Partition`` / ``DataPartitioner``
+deterministically shuffle dataset indices with seed 1234 and split them into
+``world_size`` equal fractions; per-rank batch size is global_batch //
+world_size (global default 128).
+
+torchvision is not installed and there is no network, so the MNIST source is a
+synthetic stand-in of the same shape and normalization statistics
+(1x28x28 f32, mean 0.1307 / std 0.3081 applied — i.e. we generate already-
+normalized tensors), with deterministic per-index content so every rank and
+every run sees identical data for a given seed (BASELINE.json:5: synthetic
+data / random-init weights).
+"""
+
+from __future__ import annotations
+
+from random import Random
+
+import torch
+
+
+class SyntheticMNIST(torch.utils.data.Dataset):
+    """Deterministic synthetic MNIST-shaped dataset.
+
+    Each index yields (image [1,28,28] f32, label int64). Content is a pure
+    function of (seed, index) — cheap blockwise generation from a
+    counter-based generator, no stored tensors.
+    """
+
+    def __init__(self, n: int = 60000, seed: int = 1234):
+        self.n = int(n)
+        self.seed = int(seed)
+
+    def __len__(self):
+        return self.n
+
+    def __getitem__(self, idx):
+        g = torch.Generator().manual_seed((self.seed << 20) ^ idx)
+        img = torch.randn(1, 28, 28, generator=g)
+        label = int(torch.randint(0, 10, (1,), generator=g))
+        return img, label
+
+    def batch(self, indices, device=None, dtype=torch.float32):
+        """Vectorized batch generation (fast path used by the loaders)."""
+        imgs = torch.empty(len(indices), 1, 28, 28, dtype=torch.float32)
+        labels = torch.empty(len(indices), dtype=torch.int64)
+        for i, idx in enumerate(indices):
+            im, lb = self[idx]
+            imgs[i] = im
+            labels[i] = lb
+        if device is not None:
+            imgs = imgs.to(device, dtype)
+            labels = labels.to(device)
+        return imgs, labels
+
+
+class Partition:
+    """A slice of a dataset given by an index list (reference R3)."""
+
+    def __init__(self, data, index):
+        self.data = data
+        self.index = index
+
+    def __len__(self):
+        return len(self.index)
+
+    def __getitem__(self, i):
+        return self.data[self.index[i]]
+
+
+class DataPartitioner:
+    """Deterministic shuffle (seed 1234) + fractional split (reference R3)."""
+
+    def __init__(self, data, sizes=(0.7, 0.2, 0.1), seed: int = 1234):
+        self.data = data
+        self.partitions = []
+        rng = Random()
+        rng.seed(seed)
+        data_len = len(data)
+        indexes = list(range(data_len))
+        rng.shuffle(indexes)
+        for frac in sizes:
+            part_len = int(frac * data_len)
+            self.partitions.append(indexes[0:part_len])
+            indexes = indexes[part_len:]
+
+    def use(self, partition):
+        return Partition(self.data, self.partitions[partition])
+
+
+GLOBAL_BATCH = 128
+
+
+def partition_dataset(rank: int, world_size: int, global_batch: int = GLOBAL_BATCH,
+                      n: int = 60000, seed: int = 1234):
+    """Reference-compatible rank sharding.
+
+    Returns (DataLoader, per_rank_batch_size).
+    """
+    dataset = SyntheticMNIST(n=n, seed=seed)
+    bsz = int(global_batch / float(world_size))
+    partition_sizes = [1.0 / world_size for _ in range(world_size)]
+    partition = DataPartitioner(dataset, partition_sizes, seed=seed).use(rank)
+    loader = torch.utils.data.DataLoader(partition, batch_size=bsz, shuffle=True)
+    return loader, bsz
+
+
+class SyntheticBatchStream:
+    """Infinite synthetic batch stream for benchmarking: pre-generates a pool
+    of batches once, then cycles. Pairs with runtime.Prefetcher for pinned
+    async H2D staging on a side stream."""
+
+    def __init__(self, batch_size: int, n_pool: int = 4, seed: int = 1234,
+                 dtype=torch.float32, pin: bool = False):
+        g = torch.Generator().manual_seed(seed)
+        self.pool = []
+        for _ in range(n_pool):
+            x = torch.randn(batch_size, 1, 28, 28, generator=g, dtype=torch.float32).to(dtype)
+            y = torch.randint(0, 10, (batch_size,), generator=g)
+            if pin:
+                x = x.pin_memory()
+                y = y.pin_memory()
+            self.pool.append((x, y))
+        self.i = 0
+
+    def __iter__(self):
+        return self
+
+    def __next__(self):
+        b = self.pool[self.i % len(self.pool)]
+        self.i += 1
+        return b

@@ -1,0 +1,11 @@
+from . import native  # noqa: F401
+from .functional import (  # noqa: F401
+    conv2d,
+    dropout,
+    linear,
+    log_softmax,
+    logsoftmax_nll,
+    maxpool2x2_relu,
+    relu,
+)
+from .sgd import SGD  # noqa: F401

@@ -1,0 +1,81 @@
+"""SGD with momentum — fused multi-tensor HIP step on device.
+
+Reference semantics (torch.optim.SGD, lr=0.01 momentum=0.5 — SURVEY.md §2a R8):
+    buf = momentum * buf + grad ; param -= lr * buf
+(no dampening, no nesterov, no weight decay — the reference uses none).
+
+On GPU the whole parameter set updates in ONE kernel launch: a chunk table
+(param/grad/momentum pointers + lengths) is built once at init and the kernel
+grid-strides across the concatenated element range. On CPU it falls back to
+plain torch ops (used by the gloo plumbing config and the no-GPU tests).
+"""
+
+from __future__ import annotations
+
+import torch
+
+from . import native
+
+
+class SGD(torch.optim.Optimizer):
+    def __init__(self, params, lr=0.01, momentum=0.0):
+        defaults = dict(lr=lr, momentum=momentum)
+        super().__init__(params, defaults)
+        self._table = None  # lazily built device pointer table
+
+    def _build_table(self, group):
+        """Device chunk table: for each param i, (param_ptr, grad_ptr, mom_ptr, numel).
+
+        Pointers are stable for the life of the training run: grads are bucket
+        views installed once by the DDP engine (parallel/ddp.py) or allocated
+        here on first step; momentum buffers are allocated here.
+        """
+        ptrs = []
+        total = 0
+        params = [p for p in group["params"] if p.requires_grad]
+        for p in params:
+            st = self.state[p]
+            if "momentum_buffer" not in st:
+                st["momentum_buffer"] = torch.zeros_like(p, memory_format=torch.contiguous_format)
+            if p.grad is None:
+                p.grad = torch.zeros_like(p, memory_format=torch.contiguous_format)
+            assert p.is_contiguous() and p.grad.is_contiguous()
+            ptrs.append(
+                (p.data_ptr(), p.grad.data_ptr(), st["momentum_buffer"].data_ptr(), p.numel())
+            )
+            total += p.numel()
+        table = torch.tensor(ptrs, dtype=torch.int64, device=params[0].device)
+        return table, total, [p.grad.data_ptr() for p in params]
+
+    @torch.no_grad()
+    def step(self, closure=None):
+        loss = None
+        if closure is not None:
+            with torch.enable_grad():
+                loss = closure()
+        for group in self.param_groups:
+            params = [p for p in group["params"] if p.requires_grad and p.grad is not None]
+            if not params:
+                continue
+            if params[0].is_cuda:
+                cached = self._table
+                if (
+                    cached is None
+                    or cached[2] != [p.grad.data_ptr() for p in params if p.grad is not None]
+                ):
+                    self._table = self._build_table(group)
+                table, total, _ = self._table
+                native.ext().fused_sgd_step(table, total, group["lr"], group["momentum"])
+            else:
+                for p in params:
+                    buf = self.state[p].get("momentum_buffer")
+                    if group["momentum"] != 0.0:
+                        if buf is None:
+                            buf = torch.clone(p.grad).detach()
+                            self.state[p]["momentum_buffer"] = buf
+                        else:
+                            buf.mul_(group["momentum"]).add_(p.grad)
+                        p.add_(buf, alpha=-group["lr"])
+                    else:
+                        p.add_(p.grad, alpha=-group["lr"])
+        return loss

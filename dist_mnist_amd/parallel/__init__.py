@@ -1,0 +1,2 @@
+from .comm import init_distributed  # noqa: F401
+from .ddp import DistributedDataParallel, average_gradients  # noqa: F401

@@ -1,0 +1,129 @@
+"""Distributed plumbing tests — gloo world_size=2 on 127.0.0.1, no GPU
+(SURVEY.md §4 "Distributed plumbing"; BASELINE.json config 1).
+
+Verifies the bucketed overlapped DDP engine reproduces the reference's
+per-parameter SUM+divide semantics bitwise-comparable, and
+that 2-rank data-parallel training equals single-rank large-batch training.
+"""
+
+import os
+
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+from dist_mnist_amd.models import Net
+
+
+def _init(rank, world, port):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+
+
+def _worker_bucketed_vs_reference(rank, world, port, q):
+    from dist_mnist_amd.parallel import DistributedDataParallel, average_gradients
+
+    _init(rank, world, port)
+    try:
+        torch.manual_seed(100)  # same params on both ranks
+        base = Net().eval()
+
+        # per-rank distinct batch
+        g = torch.Generator().manual_seed(1000 + rank)
+        x = torch.randn(8, 1, 28, 28, generator=g)
+        t = torch.randint(0, 10, (8,), generator=g)
+
+        # --- path A: reference semantics (per-param blocking allreduce)
+        ref = Net().eval()
+        ref.load_state_dict(base.state_dict())
+        _, loss = ref(x, t)
+        loss.backward()
+        average_gradients(ref)
+        ref_grads = {k: p.grad.clone() for k, p in ref.named_parameters()}
+
+        # --- path B: bucketed overlapped engine
+        ddp = DistributedDataParallel(Net().eval(), broadcast_params=False)
+        ddp.module.load_state_dict(base.state_dict())
+        ddp.zero_grad()
+        _, loss = ddp(x, t)
+        loss.backward()
+        ddp.sync()
+
+        for k, p in ddp.module.named_parameters():
+            torch.testing.assert_close(p.grad, ref_grads[k], rtol=1e-6, atol=1e-7,
+                                       msg=f"bucketed != reference semantics: {k}")
+        q.put((rank, "ok"))
+    except Exception as e:  # pragma: no cover
+        q.put((rank, f"FAIL: {type(e).__name__}: {e}"))
+    finally:
+        dist.destroy_process_group()
+
+
+def _worker_dp_equals_large_batch(rank, world, port, q):
+    from dist_mnist_amd.ops import SGD
+    from dist_mnist_amd.parallel import DistributedDataParallel
+
+    _init(rank, world, port)
+    try:
+        torch.manual_seed(200)
+        ddp = DistributedDataParallel(Net().eval(), broadcast_params=False)
+        opt = SGD(ddp.module.parameters(), lr=0.05, momentum=0.5)
+
+        # global batch 16 split as 8+8; single-rank oracle runs all 16
+        g = torch.Generator().manual_seed(42)
+        X = torch.randn(16, 1, 28, 28, generator=g)
+        T = torch.randint(0, 10, (16,), generator=g)
+        xs = X[rank * 8:(rank + 1) * 8]
+        ts = T[rank * 8:(rank + 1) * 8]
+
+        oracle = Net().eval()
+        oracle.load_state_dict(ddp.module.state_dict())
+        oopt = SGD(oracle.parameters(), lr=0.05, momentum=0.5)
+
+        for _ in range(3):
+            ddp.zero_grad()
+            _, loss = ddp(xs, ts)
+            loss.backward()
+            ddp.sync()
+            opt.step()
+
+            oopt.zero_grad()
+            _, oloss = oracle(X, T)
+            oloss.backward()
+            oopt.step()
+
+        for (k, p), (_, op) in zip(ddp.module.named_parameters(), oracle.named_parameters()):
+            torch.testing.assert_close(p, op, rtol=1e-5, atol=1e-6,
+                                       msg=f"DP != large-batch: {k}")
+        q.put((rank, "ok"))
+    except Exception as e:  # pragma: no cover
+        q.put((rank, f"FAIL: {type(e).__name__}: {e}"))
+    finally:
+        dist.destroy_process_group()
+
+
+def _run_world(fn, world, port):
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=fn, args=(r, world, port, q)) for r in range(world)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(world):
+        rank, status = q.get()
+        results[rank] = status
+    for p in procs:
+        p.join(timeout=60)
+    assert all(s == "ok" for s in results.values()), results
+
+
+@pytest.mark.timeout(180)
+def test_bucketed_matches_reference_semantics(free_port):
+    _run_world(_worker_bucketed_vs_reference, 2, free_port)
+
+
+@pytest.mark.timeout(180)
+def test_dp_training_equals_large_batch(free_port):
+    _run_world(_worker_dp_equals_large_batch, 2, free_port)
