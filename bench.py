@@ -1,0 +1,142 @@
+#!/usr/bin/env python3
+"""Flagship training-step benchmark (driver contract — see task brief).
+
+    python bench.py --gpus N --steps K --warmup W
+
+For N > 1 the driver launches this under torch.distributed.run with one rank
+per GPU (RCCL over xGMI); RANK/LOCAL_RANK/WORLD_SIZE/MASTER_* come from the
+env. Measures the BASELINE.json metric — images/sec (whole node) of the
+LeNet-style CNN on synthetic MNIST-shaped data (BASELINE.json:2, configs 2-4):
+full training semantics per step (fwd incl. dropout, fused loss, bwd, DDP
+gradient averaging, fused SGD), W untimed warmup steps, then exactly K timed
+steps bracketed by barrier+synchronize on both sides, MAX step time over
+ranks, one JSON line from rank 0.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+import torch.distributed as dist
+
+from dist_mnist_amd.data import SyntheticBatchStream
+from dist_mnist_amd.models import Net
+from dist_mnist_amd.ops import SGD
+from dist_mnist_amd.parallel import DistributedDataParallel, init_distributed
+from dist_mnist_amd.runtime import Prefetcher
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=50)
+    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--batch", type=int, default=None,
+                   help="per-GPU batch (default 8192 on GPU — BASELINE config 4 shape)")
+    p.add_argument("--dtype", type=str, default="fp32", choices=["fp32"],
+                   help="compute dtype of the hot path (bf16 path: later milestone)")
+    p.add_argument("--bucket-mb", type=float, default=25.0)
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    use_cuda = torch.cuda.is_available()
+    if world > 1:
+        rank, world = init_distributed("nccl" if use_cuda else "gloo")
+    else:
+        rank = 0
+    local_rank = int(os.environ.get("LOCAL_RANK", rank))
+    device = torch.device(f"cuda:{local_rank}" if use_cuda else "cpu")
+    if use_cuda:
+        torch.cuda.set_device(device)
+
+    batch = args.batch or (8192 if use_cuda else 256)
+    torch.manual_seed(1234)
+
+    model = Net().to(device).train()
+    if world > 1:
+        model = DistributedDataParallel(model, bucket_bytes=int(args.bucket_mb * 2**20))
+    opt = SGD(model.parameters(), lr=0.01, momentum=0.5)
+
+    stream = SyntheticBatchStream(batch_size=batch, n_pool=4, seed=1234 + rank,
+                                  pin=use_cuda)
+    batches = Prefetcher(stream, device)
+
+    def one_step():
+        x, t = next(batches)
+        if isinstance(model, DistributedDataParallel):
+            model.zero_grad()
+        else:
+            opt.zero_grad()
+        _, loss = model(x, t)
+        loss.backward()
+        if isinstance(model, DistributedDataParallel):
+            model.sync()
+        opt.step()
+        return loss
+
+    def barrier_sync():
+        if world > 1:
+            dist.barrier()
+        if use_cuda:
+            torch.cuda.synchronize(device)
+
+    for _ in range(args.warmup):
+        one_step()
+    barrier_sync()
+
+    t0 = time.perf_counter()
+    last_loss = None
+    for _ in range(args.steps):
+        last_loss = one_step()
+    barrier_sync()
+    dt = time.perf_counter() - t0
+
+    # MAX step time over ranks
+    if world > 1:
+        dt_t = torch.tensor([dt], dtype=torch.float64,
+                            device=device if dist.get_backend() == "nccl" else "cpu")
+        dist.all_reduce(dt_t, op=dist.ReduceOp.MAX)
+        dt = float(dt_t.item())
+
+    n_gpus = world if use_cuda else world  # one rank per GPU
+    images_per_sec = world * batch * args.steps / dt
+
+    if rank == 0:
+        result = {
+            "metric": "images/sec (whole node), LeNet-style CNN on MNIST",
+            "value": images_per_sec,
+            "unit": "images/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": dt / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,  # reference publishes no numbers
+            "dtype": args.dtype,
+            "data": "synthetic",
+            "final_loss": float(last_loss.item()) if last_loss is not None else None,
+            "config": {
+                "model": "lenet5-style-2conv (SURVEY.md R5)",
+                "global_batch": world * batch,
+                "per_gpu_batch": batch,
+                "seq_len": None,
+                "parallelism": f"dp{world}",
+                "device": str(device),
+            },
+        }
+        print(json.dumps(result))
+
+    if world > 1 and dist.is_initialized():
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -23,17 +23,21 @@ class SGD(torch.optim.Optimizer):
         super().__init__(params, defaults)
         self._table = None  # lazily built device pointer table
 
+    CHUNK = 4096  # elements per block in the fused kernel
+
     def _build_table(self, group):
-        """Device chunk table: for each param i, (param_ptr, grad_ptr, mom_ptr, numel).
+        """Device tables for the one-launch fused step:
+        ptrs[t] = (param_ptr, grad_ptr, momentum_ptr, numel);
+        chunks[b] = (tensor_idx, elem_offset) — one block per chunk.
 
         Pointers are stable for the life of the training run: grads are bucket
         views installed once by the DDP engine (parallel/ddp.py) or allocated
         here on first step; momentum buffers are allocated here.
         """
         ptrs = []
-        total = 0
+        chunks = []
         params = [p for p in group["params"] if p.requires_grad]
-        for p in params:
+        for ti, p in enumerate(params):
             st = self.state[p]
             if "momentum_buffer" not in st:
                 st["momentum_buffer"] = torch.zeros_like(p, memory_format=torch.contiguous_format)
@@ -43,9 +47,12 @@ class SGD(torch.optim.Optimizer):
             ptrs.append(
                 (p.data_ptr(), p.grad.data_ptr(), st["momentum_buffer"].data_ptr(), p.numel())
             )
-            total += p.numel()
-        table = torch.tensor(ptrs, dtype=torch.int64, device=params[0].device)
-        return table, total, [p.grad.data_ptr() for p in params]
+            for off in range(0, p.numel(), self.CHUNK):
+                chunks.append((ti, off))
+        dev = params[0].device
+        ptrs_t = torch.tensor(ptrs, dtype=torch.int64, device=dev)
+        chunks_t = torch.tensor(chunks, dtype=torch.int32, device=dev)
+        return ptrs_t, chunks_t, [p.grad.data_ptr() for p in params]
 
     @torch.no_grad()
     def step(self, closure=None):
@@ -64,8 +71,10 @@ class SGD(torch.optim.Optimizer):
                     or cached[2] != [p.grad.data_ptr() for p in params if p.grad is not None]
                 ):
                     self._table = self._build_table(group)
-                table, total, _ = self._table
-                native.ext().fused_sgd_step(table, total, group["lr"], group["momentum"])
+                ptrs_t, chunks_t, _ = self._table
+                native.ext().fused_sgd_step(
+                    ptrs_t, chunks_t, self.CHUNK, group["lr"], group["momentum"]
+                )
             else:
                 for p in params:
                     buf = self.state[p].get("momentum_buffer")

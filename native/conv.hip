@@ -1,0 +1,361 @@
+// Conv2d fwd / bwd-input / bwd-weight as fused-im2col implicit-GEMM on MFMA
+// (BASELINE.json:5 "Conv2d fwd/bwd via im2col-GEMM ...
+// hand-written HIP for CDNA4 with MFMA/LDS tiling").
+//
+// fp32 path, NCHW, exact-f32 MFMA (v_mfma_f32_16x16x4_f32 — guide §3: the f32
+// vector-rate matrix path, ~2.4x a VALU GEMM at identical numerics).
+//
+// GEMM views (M = N*OH*OW output pixels, K = CI*KH*KW, canonical k order
+// (ci,kh,kw) == NCHW patch order, so weights need NO repack):
+//   fwd   : D[co][m]  = sum_k  W[co][k]      * im2col(x)[m][k]
+//   dgrad : D[ci][m'] = sum_k' Wflip[ci][k'] * col(dy)[m'][k']   (k'=(co,kh,kw))
+//   wgrad : D[k][co]  = sum_m  im2col(x)[m][k] * dy[m][co]       (split-M + atomicAdd)
+//
+// Operand/store orientation is chosen so every global 16-lane access group
+// walks the fastest (ow/iw) NCHW axis => coalesced gathers and stores.
+#include "common.h"
+
+// ---------------------------------------------------------------------------
+// Forward. Block = 256 threads = 4 waves; each wave owns MW=2 m-frags of 16
+// rows => BM = 128 output pixels per block; one 16-wide co tile per block.
+// Weights + k-decode staged in LDS per BK=32 chunk.
+// ---------------------------------------------------------------------------
+template <bool HAS_BIAS, bool RELU, bool BOUNDS>
+__global__ __launch_bounds__(256) void conv2d_fwd_f32_k(
+    const float* __restrict__ x, const float* __restrict__ w,
+    const float* __restrict__ bias, float* __restrict__ y,
+    int CI, int H, int W, int CO, int KH, int KW, int OH, int OW,
+    int stride, int pad, int M, int K) {
+  constexpr int BM = 128, BK = 32, MW = 2;
+  const int m0 = blockIdx.x * BM;
+  const int co0 = blockIdx.y * 16;
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6;
+  const int l = tid & 63;
+  const int lm = l & 15;   // B col (m) / A row (co)
+  const int lk = l >> 4;   // k slice 0..3
+
+  __shared__ float lds_w[BK][16];
+  __shared__ int lds_koff[BK];    // ci*H*W + kh*W + kw
+  __shared__ int lds_khw[BK];     // (kh<<16)|kw, only used if BOUNDS
+
+  const int HW = H * W;
+  const int OHOW = OH * OW;
+
+  // per-(lane, mi) output-pixel decode — fixed across the whole K loop
+  int xbase[MW];    // n*CI*HW + ih0*W + iw0
+  int ybase[MW];    // n*CO*OHOW + oh*OW + ow
+  int ih0[MW], iw0[MW];
+  bool mval[MW];
+#pragma unroll
+  for (int mi = 0; mi < MW; ++mi) {
+    int m = m0 + (wid * MW + mi) * 16 + lm;
+    mval[mi] = m < M;
+    int mm = mval[mi] ? m : 0;
+    int n = mm / OHOW, r = mm % OHOW;
+    int oh = r / OW, ow = r % OW;
+    ih0[mi] = oh * stride - pad;
+    iw0[mi] = ow * stride - pad;
+    xbase[mi] = n * CI * HW + ih0[mi] * W + iw0[mi];
+    ybase[mi] = n * CO * OHOW + r;
+  }
+
+  f32x4 acc[MW] = {};
+
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    // stage W^T tile [BK][16co] + k decode
+    __syncthreads();
+    for (int idx = tid; idx < BK * 16; idx += 256) {
+      int kk = idx >> 4, co = idx & 15;
+      int k = k0 + kk;
+      lds_w[kk][co] = (k < K && co0 + co < CO) ? w[(co0 + co) * K + k] : 0.f;
+    }
+    if (tid < BK) {
+      int k = k0 + tid;
+      int kcl = k < K ? k : 0;
+      int ci = kcl / (KH * KW), r = kcl % (KH * KW);
+      int kh = r / KW, kw = r % KW;
+      lds_koff[tid] = ci * HW + kh * W + kw;
+      lds_khw[tid] = (kh << 16) | kw;
+    }
+    __syncthreads();
+
+    int klim = min(BK, K - k0);
+#pragma unroll 4
+    for (int kk = 0; kk + 4 <= klim; kk += 4) {
+      float a = lds_w[kk + lk][lm];
+      int koff = lds_koff[kk + lk];
+      bool kin = true;
+      if (BOUNDS) {
+        int khw = lds_khw[kk + lk];
+        int kh = khw >> 16, kw = khw & 0xffff;
+        // per-mi bound check folded below (ih0/iw0 differ per mi)
+        (void)kh; (void)kw;
+      }
+#pragma unroll
+      for (int mi = 0; mi < MW; ++mi) {
+        bool ok = mval[mi] & kin;
+        if (BOUNDS) {
+          int khw = lds_khw[kk + lk];
+          int kh = khw >> 16, kw = khw & 0xffff;
+          int ih = ih0[mi] + kh, iw = iw0[mi] + kw;
+          ok = ok && ih >= 0 && ih < H && iw >= 0 && iw < W;
+        }
+        float b = ok ? x[xbase[mi] + koff] : 0.f;
+        acc[mi] = mfma16x16x4(a, b, acc[mi]);
+      }
+      // tail k (klim not multiple of 4) handled below
+    }
+    int ktail = klim & 3;
+    if (ktail) {
+      int kk = klim - ktail;  // process remaining 1..3 slices: lanes with lk < ktail
+      float a = (lk < ktail) ? lds_w[kk + lk][lm] : 0.f;
+      int koff = (lk < ktail) ? lds_koff[kk + lk] : 0;
+#pragma unroll
+      for (int mi = 0; mi < MW; ++mi) {
+        bool ok = mval[mi] && (lk < ktail);
+        if (BOUNDS && ok) {
+          int khw = lds_khw[kk + lk];
+          int kh = khw >> 16, kw = khw & 0xffff;
+          int ih = ih0[mi] + kh, iw = iw0[mi] + kw;
+          ok = ih >= 0 && ih < H && iw >= 0 && iw < W;
+        }
+        float b = ok ? x[xbase[mi] + koff] : 0.f;
+        acc[mi] = mfma16x16x4(a, b, acc[mi]);
+      }
+    }
+  }
+
+  // epilogue: D col = lm -> m (coalesced over ow), row = lk*4+reg -> co
+#pragma unroll
+  for (int mi = 0; mi < MW; ++mi) {
+    if (!mval[mi]) continue;
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      int co = co0 + lk * 4 + reg;
+      if (co >= CO) continue;
+      float v = acc[mi][reg];
+      if (HAS_BIAS) v += bias[co];
+      if (RELU) v = fmaxf(v, 0.f);
+      y[ybase[mi] + co * OHOW] = v;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Backward-input (dgrad). Same skeleton; m' = (n, ih, iw), k' = (co, kh, kw),
+// A = flipped weights staged as [k'][ci-tile], B = dy gathered with the
+// transposed-conv index map (zero where oh/ow fall outside or off-stride).
+// ---------------------------------------------------------------------------
+template <bool S1>
+__global__ __launch_bounds__(256) void conv2d_dgrad_f32_k(
+    const float* __restrict__ dy, const float* __restrict__ w,
+    float* __restrict__ dx,
+    int CI, int H, int W, int CO, int KH, int KW, int OH, int OW,
+    int stride, int pad, int Mp, int Kp) {
+  constexpr int BM = 128, BK = 32, MW = 2;
+  const int m0 = blockIdx.x * BM;
+  const int ci0 = blockIdx.y * 16;
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6;
+  const int l = tid & 63;
+  const int lm = l & 15;
+  const int lk = l >> 4;
+
+  __shared__ float lds_w[BK][16];   // Wflip^T [k'][ci]
+  __shared__ int lds_kdec[BK];      // (co<<16)|(kh<<8)|kw
+
+  const int HW = H * W;
+  const int OHOW = OH * OW;
+  const int KHKW = KH * KW;
+
+  int dybase[MW];  // n*CO*OHOW
+  int ih[MW], iw[MW];
+  int dxbase[MW];
+  bool mval[MW];
+#pragma unroll
+  for (int mi = 0; mi < MW; ++mi) {
+    int m = m0 + (wid * MW + mi) * 16 + lm;
+    mval[mi] = m < Mp;
+    int mm = mval[mi] ? m : 0;
+    int n = mm / HW, r = mm % HW;
+    ih[mi] = r / W;
+    iw[mi] = r % W;
+    dybase[mi] = n * CO * OHOW;
+    dxbase[mi] = n * CI * HW + r;
+  }
+
+  f32x4 acc[MW] = {};
+
+  for (int k0 = 0; k0 < Kp; k0 += BK) {
+    __syncthreads();
+    for (int idx = tid; idx < BK * 16; idx += 256) {
+      int kk = idx >> 4, ci = idx & 15;
+      int k = k0 + kk;
+      float v = 0.f;
+      if (k < Kp && ci0 + ci < CI) {
+        int co = k / KHKW, r = k % KHKW;
+        int kh = r / KW, kw = r % KW;
+        // flipped kernel: w[co][ci][KH-1-kh][KW-1-kw]
+        v = w[((co * CI + ci0 + ci) * KH + (KH - 1 - kh)) * KW + (KW - 1 - kw)];
+      }
+      lds_w[kk][ci] = v;
+    }
+    if (tid < BK) {
+      int k = k0 + tid;
+      int kcl = k < Kp ? k : 0;
+      int co = kcl / KHKW, r = kcl % KHKW;
+      int kh = r / KW, kw = r % KW;
+      lds_kdec[tid] = (co << 16) | (kh << 8) | kw;
+    }
+    __syncthreads();
+
+    int klim = min(BK, Kp - k0);
+    for (int kk = 0; kk < klim; kk += 4) {
+      bool kin = lk < (klim - kk);
+      int kdec = lds_kdec[kin ? kk + lk : 0];
+      float a = kin ? lds_w[kk + lk][lm] : 0.f;
+      int co = kdec >> 16, kh = (kdec >> 8) & 0xff, kw = kdec & 0xff;
+      // flipped map: dx[ih] gets dy[oh] where ih = oh*s - p + (KH-1-kh)' ...
+      // with the flip above: oh = (ih + pad - (KH-1-kh)) / s; using kdec's
+      // (kh,kw) as the FLIPPED position directly:
+#pragma unroll
+      for (int mi = 0; mi < MW; ++mi) {
+        float b = 0.f;
+        if (mval[mi] & kin) {
+          int tH = ih[mi] + pad - (KH - 1 - kh);
+          int tW = iw[mi] + pad - (KW - 1 - kw);
+          if (S1) {
+            if (tH >= 0 && tH < OH && tW >= 0 && tW < OW)
+              b = dy[dybase[mi] + co * OHOW + tH * OW + tW];
+          } else {
+            if (tH >= 0 && tW >= 0 && tH % stride == 0 && tW % stride == 0) {
+              int oh = tH / stride, ow = tW / stride;
+              if (oh < OH && ow < OW)
+                b = dy[dybase[mi] + co * OHOW + oh * OW + ow];
+            }
+          }
+        }
+        acc[mi] = mfma16x16x4(a, b, acc[mi]);
+      }
+    }
+  }
+
+#pragma unroll
+  for (int mi = 0; mi < MW; ++mi) {
+    if (!mval[mi]) continue;
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      int ci = ci0 + lk * 4 + reg;
+      if (ci >= CI) continue;
+      dx[dxbase[mi] + ci * HW] = acc[mi][reg];
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Backward-weight (wgrad) + bias. D[k-tile][co-tile], reduce over an M-split
+// chunk per block, atomicAdd f32 into dw/db (zero-initialized by the caller).
+// dy chunk is LDS-transposed so its MFMA reads are conflict-free; per-chunk
+// pixel decode is precomputed into LDS once (m -> x base addr / bounds).
+// ---------------------------------------------------------------------------
+template <bool BOUNDS, bool WANT_BIAS>
+__global__ __launch_bounds__(256) void conv2d_wgrad_f32_k(
+    const float* __restrict__ dy, const float* __restrict__ x,
+    float* __restrict__ dw, float* __restrict__ db,
+    int CI, int H, int W, int CO, int KH, int KW, int OH, int OW,
+    int stride, int pad, int M, int K, int m_per_split) {
+  constexpr int MC = 64;  // m chunk
+  const int k0 = blockIdx.x * 64;  // 4 waves x 16 k rows
+  const int co0 = blockIdx.y * 16;
+  const int ms = blockIdx.z;
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6;
+  const int l = tid & 63;
+  const int lm = l & 15;
+  const int lk = l >> 4;
+
+  __shared__ float lds_dy[MC][17];  // [m][co], padded
+  __shared__ int lds_mbase[MC];     // n*CI*HW + ih0*W + iw0
+  __shared__ int lds_mhw[MC];       // (ih0+pad)<<16 | (iw0+pad)  (for bounds)
+
+  const int HW = H * W;
+  const int OHOW = OH * OW;
+
+  // this wave's k row (A row = lm): fixed over the whole reduction
+  int k = k0 + wid * 16 + lm;
+  bool kval = k < K;
+  int kcl = kval ? k : 0;
+  int ci = kcl / (KH * KW), kr = kcl % (KH * KW);
+  int kh = kr / KW, kw = kr % KW;
+  int koff = ci * HW + kh * W + kw;
+
+  f32x4 acc = {};
+  float db_part = 0.f;
+
+  int m_begin = ms * m_per_split;
+  int m_end = min(M, m_begin + m_per_split);
+
+  for (int mc = m_begin; mc < m_end; mc += MC) {
+    __syncthreads();
+    // stage dy[mc:mc+64][co0:co0+16] transposed into LDS (coalesced over m)
+    for (int idx = tid; idx < MC * 16; idx += 256) {
+      int co = idx >> 6;        // 0..15? MC*16/64: idx/64 -> co? careful:
+      int mloc = idx & 63;      // we want m fastest for coalescing
+      co = idx >> 6;            // idx = co*64 + mloc
+      int m = mc + mloc;
+      float v = 0.f;
+      if (m < m_end && co0 + co < CO) {
+        int n = m / OHOW, r = m % OHOW;
+        v = dy[(n * CO + co0 + co) * OHOW + r];
+      }
+      lds_dy[mloc][co] = v;
+    }
+    // stage per-m x-base decode
+    if (tid < MC) {
+      int m = mc + tid;
+      int mm = m < m_end ? m : 0;
+      int n = mm / OHOW, r = mm % OHOW;
+      int oh = r / OW, ow = r % OW;
+      int ih0 = oh * stride - pad, iw0 = ow * stride - pad;
+      lds_mbase[tid] = n * CI * HW + ih0 * W + iw0;
+      lds_mhw[tid] = ((ih0 + 16384) << 16) | (iw0 + 16384);  // biased for bounds
+    }
+    __syncthreads();
+
+    int mlim = min(MC, m_end - mc);
+    for (int mm = 0; mm < mlim; mm += 4) {
+      bool min_ok = (mm + lk) < mlim;
+      int msel = min_ok ? mm + lk : 0;
+      float b = lds_dy[msel][lm];            // B[k=m][col=co]
+      float a = 0.f;
+      if (kval & min_ok) {
+        bool ok = true;
+        if (BOUNDS) {
+          int hw = lds_mhw[msel];
+          int ih = (hw >> 16) - 16384 + kh;
+          int iw = (hw & 0xffff) - 16384 + kw;
+          ok = ih >= 0 && ih < H && iw >= 0 && iw < W;
+        }
+        if (ok) a = x[lds_mbase[msel] + koff];
+      }
+      if (!min_ok) b = 0.f;
+      acc = mfma16x16x4(a, b, acc);
+      if (WANT_BIAS) db_part += b;  // sum over m of dy[m][co=lm] (lk-partial)
+    }
+  }
+
+  // dw[co][k] atomic accumulate: D row = lk*4+reg -> k-local, col = lm -> co
+#pragma unroll
+  for (int reg = 0; reg < 4; ++reg) {
+    int kk = k0 + wid * 16 + lk * 4 + reg;
+    int co = co0 + lm;
+    if (kk < K && co < CO) atomicAdd(&dw[co * K + kk], acc[reg]);
+  }
+  if (WANT_BIAS && blockIdx.x == 0 && wid == 0) {
+    // reduce db_part across the 4 lk groups of wave 0
+    db_part += __shfl_down(db_part, 16);
+    db_part += __shfl_down(db_part, 32);
+    if (lk == 0 && co0 + lm < CO) atomicAdd(&db[co0 + lm], db_part);
+  }
+}

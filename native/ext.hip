@@ -1,0 +1,353 @@
+// dist_mnist_amd native extension — host bindings for the CDNA4 kernel core.
+// Single TU: includes the kernel files and exposes torch-tensor entry points.
+// Build: hipcc --offload-arch=gfx950 (in-tree, see setup.py / __graft_entry__).
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+
+#include "conv.hip"
+#include "linear.hip"
+#include "elementwise.hip"
+
+namespace {
+
+#define CHECK_F32(t) \
+  TORCH_CHECK((t).is_cuda() && (t).is_contiguous() && (t).scalar_type() == at::kFloat, \
+              #t " must be contiguous f32 CUDA tensor")
+#define CHECK_I64(t) \
+  TORCH_CHECK((t).is_cuda() && (t).is_contiguous() && (t).scalar_type() == at::kLong, \
+              #t " must be contiguous i64 CUDA tensor")
+
+inline hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+// ---------------------------------------------------------------------------
+at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias, long stride,
+                      long pad) {
+  CHECK_F32(x);
+  CHECK_F32(w);
+  bool has_bias = bias.numel() > 0;
+  if (has_bias) CHECK_F32(bias);
+  int N = x.size(0), CI = x.size(1), H = x.size(2), W = x.size(3);
+  int CO = w.size(0), KH = w.size(2), KW = w.size(3);
+  TORCH_CHECK(w.size(1) == CI, "conv2d channel mismatch");
+  int OH = (H + 2 * pad - KH) / stride + 1;
+  int OW = (W + 2 * pad - KW) / stride + 1;
+  auto y = at::empty({N, CO, OH, OW}, x.options());
+  int M = N * OH * OW, K = CI * KH * KW;
+  dim3 grid(cdiv(M, 128), cdiv(CO, 16));
+  bool bounds = pad > 0;
+  auto launch = [&](auto hb, auto rl, auto bd) {
+    hipLaunchKernelGGL((conv2d_fwd_f32_k<decltype(hb)::value, decltype(rl)::value,
+                                         decltype(bd)::value>),
+                       grid, dim3(256), 0, cur_stream(), x.data_ptr<float>(),
+                       w.data_ptr<float>(),
+                       has_bias ? bias.data_ptr<float>() : nullptr,
+                       y.data_ptr<float>(), CI, H, W, CO, KH, KW, OH, OW,
+                       (int)stride, (int)pad, M, K);
+  };
+  using T = std::true_type;
+  using F = std::false_type;
+  if (has_bias && bounds) launch(T{}, F{}, T{});
+  else if (has_bias) launch(T{}, F{}, F{});
+  else if (bounds) launch(F{}, F{}, T{});
+  else launch(F{}, F{}, F{});
+  return y;
+}
+
+at::Tensor conv2d_bwd_input(at::Tensor dy, at::Tensor w, long H, long W,
+                            long stride, long pad) {
+  CHECK_F32(dy);
+  CHECK_F32(w);
+  int N = dy.size(0), CO = dy.size(1), OH = dy.size(2), OW = dy.size(3);
+  int CI = w.size(1), KH = w.size(2), KW = w.size(3);
+  auto dx = at::empty({N, CI, H, W}, dy.options());
+  int Mp = N * H * W, Kp = CO * KH * KW;
+  dim3 grid(cdiv(Mp, 128), cdiv(CI, 16));
+  if (stride == 1)
+    hipLaunchKernelGGL((conv2d_dgrad_f32_k<true>), grid, dim3(256), 0,
+                       cur_stream(), dy.data_ptr<float>(), w.data_ptr<float>(),
+                       dx.data_ptr<float>(), CI, H, W, CO, KH, KW, OH, OW,
+                       (int)stride, (int)pad, Mp, Kp);
+  else
+    hipLaunchKernelGGL((conv2d_dgrad_f32_k<false>), grid, dim3(256), 0,
+                       cur_stream(), dy.data_ptr<float>(), w.data_ptr<float>(),
+                       dx.data_ptr<float>(), CI, H, W, CO, KH, KW, OH, OW,
+                       (int)stride, (int)pad, Mp, Kp);
+  return dx;
+}
+
+std::vector<at::Tensor> conv2d_bwd_weight(at::Tensor dy, at::Tensor x, long KH,
+                                          long KW, long stride, long pad,
+                                          bool want_bias) {
+  CHECK_F32(dy);
+  CHECK_F32(x);
+  int N = x.size(0), CI = x.size(1), H = x.size(2), W = x.size(3);
+  int CO = dy.size(1), OH = dy.size(2), OW = dy.size(3);
+  auto dw = at::zeros({CO, CI, KH, KW}, x.options());
+  auto db = at::zeros({want_bias ? CO : 0}, x.options());
+  int M = N * OH * OW, K = CI * KH * KW;
+  int kt = cdiv(K, 64), ct = cdiv(CO, 16);
+  int ms = std::max(1, std::min(2048 / std::max(1, kt * ct), cdiv(M, 64)));
+  int m_per_split = cdiv(cdiv(M, ms), 64) * 64;
+  ms = cdiv(M, m_per_split);
+  dim3 grid(kt, ct, ms);
+  bool bounds = pad > 0;
+  auto launch = [&](auto bd, auto wb) {
+    hipLaunchKernelGGL((conv2d_wgrad_f32_k<decltype(bd)::value, decltype(wb)::value>),
+                       grid, dim3(256), 0, cur_stream(), dy.data_ptr<float>(),
+                       x.data_ptr<float>(), dw.data_ptr<float>(),
+                       want_bias ? db.data_ptr<float>() : nullptr, CI, H, W, CO,
+                       KH, KW, OH, OW, (int)stride, (int)pad, M, K, m_per_split);
+  };
+  using T = std::true_type;
+  using F = std::false_type;
+  if (bounds && want_bias) launch(T{}, T{});
+  else if (bounds) launch(T{}, F{});
+  else if (want_bias) launch(F{}, T{});
+  else launch(F{}, F{});
+  return {dw, db};
+}
+
+// ---------------------------------------------------------------------------
+at::Tensor linear_fwd(at::Tensor x, at::Tensor w, at::Tensor bias, bool relu) {
+  CHECK_F32(x);
+  CHECK_F32(w);
+  bool has_bias = bias.numel() > 0;
+  if (has_bias) CHECK_F32(bias);
+  int M = x.size(0), CI = x.size(1), CO = w.size(0);
+  TORCH_CHECK(w.size(1) == CI, "linear shape mismatch");
+  auto y = at::empty({M, CO}, x.options());
+  dim3 grid(cdiv(M, 64), cdiv(CO, 64));
+  auto launch = [&](auto hb, auto rl) {
+    hipLaunchKernelGGL((linear_fwd_f32_k<decltype(hb)::value, decltype(rl)::value>),
+                       grid, dim3(256), 0, cur_stream(), x.data_ptr<float>(),
+                       w.data_ptr<float>(),
+                       has_bias ? bias.data_ptr<float>() : nullptr,
+                       y.data_ptr<float>(), M, CI, CO);
+  };
+  using T = std::true_type;
+  using F = std::false_type;
+  if (has_bias && relu) launch(T{}, T{});
+  else if (has_bias) launch(T{}, F{});
+  else if (relu) launch(F{}, T{});
+  else launch(F{}, F{});
+  return y;
+}
+
+std::vector<at::Tensor> linear_bwd(at::Tensor dy, at::Tensor x, at::Tensor w,
+                                   at::Tensor yact, bool relu, bool want_bias) {
+  CHECK_F32(dy);
+  CHECK_F32(x);
+  CHECK_F32(w);
+  int M = x.size(0), CI = x.size(1), CO = w.size(0);
+  auto dx = at::empty({M, CI}, x.options());
+  auto dw = at::zeros({CO, CI}, x.options());
+  auto db = at::zeros({want_bias ? CO : 0}, x.options());
+  {
+    dim3 grid(cdiv(M, 64), cdiv(CI, 64));
+    if (relu)
+      hipLaunchKernelGGL((linear_dx_f32_k<true>), grid, dim3(256), 0,
+                         cur_stream(), dy.data_ptr<float>(), w.data_ptr<float>(),
+                         yact.data_ptr<float>(), dx.data_ptr<float>(), M, CI, CO);
+    else
+      hipLaunchKernelGGL((linear_dx_f32_k<false>), grid, dim3(256), 0,
+                         cur_stream(), dy.data_ptr<float>(), w.data_ptr<float>(),
+                         yact.data_ptr<float>(), dx.data_ptr<float>(), M, CI, CO);
+  }
+  {
+    int ct = cdiv(CO, 16), it = cdiv(CI, 64);
+    int ms = std::max(1, std::min(2048 / std::max(1, ct * it), cdiv(M, 64)));
+    int m_per_split = cdiv(cdiv(M, ms), 4) * 4;
+    ms = cdiv(M, m_per_split);
+    dim3 grid(ct, it, ms);
+    auto launch = [&](auto rl, auto wb) {
+      hipLaunchKernelGGL((linear_dw_f32_k<decltype(rl)::value, decltype(wb)::value>),
+                         grid, dim3(256), 0, cur_stream(), dy.data_ptr<float>(),
+                         x.data_ptr<float>(), yact.data_ptr<float>(),
+                         dw.data_ptr<float>(),
+                         want_bias ? db.data_ptr<float>() : nullptr, M, CI, CO,
+                         m_per_split);
+    };
+    using T = std::true_type;
+    using F = std::false_type;
+    if (relu && want_bias) launch(T{}, T{});
+    else if (relu) launch(T{}, F{});
+    else if (want_bias) launch(F{}, T{});
+    else launch(F{}, F{});
+  }
+  return {dx, dw, db};
+}
+
+// ---------------------------------------------------------------------------
+std::vector<at::Tensor> maxpool2x2_relu_fwd(at::Tensor x) {
+  CHECK_F32(x);
+  int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  TORCH_CHECK(H % 2 == 0 && W % 2 == 0, "maxpool2x2 needs even H, W");
+  int OH = H / 2, OW = W / 2;
+  auto y = at::empty({N, C, OH, OW}, x.options());
+  auto idx = at::empty({N, C, OH, OW}, x.options().dtype(at::kByte));
+  long total = (long)N * C * OH * OW;
+  int blocks = (int)std::min<long>(2048, cdiv(total, 256));
+  hipLaunchKernelGGL(maxpool2x2_relu_fwd_k, dim3(blocks), dim3(256), 0,
+                     cur_stream(), x.data_ptr<float>(), y.data_ptr<float>(),
+                     idx.data_ptr<unsigned char>(), N * C, H, W, OH, OW);
+  return {y, idx};
+}
+
+at::Tensor maxpool2x2_relu_bwd(at::Tensor dy, at::Tensor y, at::Tensor idx,
+                               long H, long W) {
+  CHECK_F32(dy);
+  CHECK_F32(y);
+  int N = dy.size(0), C = dy.size(1), OH = dy.size(2), OW = dy.size(3);
+  auto dx = at::empty({N, C, H, W}, dy.options());
+  long total = (long)N * C * OH * OW;
+  int blocks = (int)std::min<long>(2048, cdiv(total, 256));
+  hipLaunchKernelGGL(maxpool2x2_relu_bwd_k, dim3(blocks), dim3(256), 0,
+                     cur_stream(), dy.data_ptr<float>(), y.data_ptr<float>(),
+                     idx.data_ptr<unsigned char>(), dx.data_ptr<float>(), N * C,
+                     (int)H, (int)W, OH, OW);
+  return dx;
+}
+
+// ---------------------------------------------------------------------------
+at::Tensor relu_fwd(at::Tensor x) {
+  CHECK_F32(x);
+  auto y = at::empty_like(x);
+  long n = x.numel();
+  int blocks = (int)std::min<long>(2048, cdiv(std::max<long>(n / 4, 1), 256));
+  hipLaunchKernelGGL(relu_fwd_k, dim3(blocks), dim3(256), 0, cur_stream(),
+                     x.data_ptr<float>(), y.data_ptr<float>(), n);
+  return y;
+}
+
+at::Tensor relu_bwd(at::Tensor dy, at::Tensor y) {
+  CHECK_F32(dy);
+  CHECK_F32(y);
+  auto dx = at::empty_like(dy);
+  long n = dy.numel();
+  int blocks = (int)std::min<long>(2048, cdiv(std::max<long>(n / 4, 1), 256));
+  hipLaunchKernelGGL(relu_bwd_k, dim3(blocks), dim3(256), 0, cur_stream(),
+                     dy.data_ptr<float>(), y.data_ptr<float>(),
+                     dx.data_ptr<float>(), n);
+  return dx;
+}
+
+// ---------------------------------------------------------------------------
+std::vector<at::Tensor> logsoftmax_nll_fwd(at::Tensor x, at::Tensor target) {
+  CHECK_F32(x);
+  CHECK_I64(target);
+  int M = x.size(0), C = x.size(1);
+  auto logp = at::empty_like(x);
+  auto loss = at::zeros({}, x.options());
+  int blocks = (int)std::min<long>(1024, cdiv(M, 256));
+  hipLaunchKernelGGL(logsoftmax_nll_fwd_k, dim3(blocks), dim3(256), 0,
+                     cur_stream(), x.data_ptr<float>(), target.data_ptr<long>(),
+                     logp.data_ptr<float>(), loss.data_ptr<float>(), M, C, 1);
+  return {logp, loss};
+}
+
+at::Tensor logsoftmax_nll_bwd(at::Tensor logp, at::Tensor target,
+                              at::Tensor dloss) {
+  CHECK_F32(logp);
+  CHECK_I64(target);
+  CHECK_F32(dloss);
+  int M = logp.size(0), C = logp.size(1);
+  auto dx = at::empty_like(logp);
+  int blocks = (int)std::min<long>(1024, cdiv(M, 256));
+  hipLaunchKernelGGL(logsoftmax_nll_bwd_k, dim3(blocks), dim3(256), 0,
+                     cur_stream(), logp.data_ptr<float>(),
+                     target.data_ptr<long>(), dloss.data_ptr<float>(),
+                     dx.data_ptr<float>(), M, C);
+  return dx;
+}
+
+at::Tensor logsoftmax_fwd(at::Tensor x) {
+  CHECK_F32(x);
+  int M = x.size(0), C = x.size(1);
+  auto logp = at::empty_like(x);
+  int blocks = (int)std::min<long>(1024, cdiv(M, 256));
+  hipLaunchKernelGGL(logsoftmax_nll_fwd_k, dim3(blocks), dim3(256), 0,
+                     cur_stream(), x.data_ptr<float>(), nullptr,
+                     logp.data_ptr<float>(), nullptr, M, C, 0);
+  return logp;
+}
+
+at::Tensor logsoftmax_bwd(at::Tensor dy, at::Tensor logp) {
+  CHECK_F32(dy);
+  CHECK_F32(logp);
+  int M = dy.size(0), C = dy.size(1);
+  auto dx = at::empty_like(dy);
+  int blocks = (int)std::min<long>(1024, cdiv(M, 256));
+  hipLaunchKernelGGL(logsoftmax_bwd_k, dim3(blocks), dim3(256), 0, cur_stream(),
+                     dy.data_ptr<float>(), logp.data_ptr<float>(),
+                     dx.data_ptr<float>(), M, C);
+  return dx;
+}
+
+// ---------------------------------------------------------------------------
+std::vector<at::Tensor> dropout_fwd(at::Tensor x, double p, bool channelwise,
+                                    long seed, long ctr) {
+  CHECK_F32(x);
+  long n = x.numel();
+  long hw = 1;
+  long mask_n = n;
+  if (channelwise) {
+    TORCH_CHECK(x.dim() == 4, "channelwise dropout needs NCHW");
+    hw = (long)x.size(2) * x.size(3);
+    mask_n = (long)x.size(0) * x.size(1);
+  }
+  auto y = at::empty_like(x);
+  auto mask = at::empty({mask_n}, x.options().dtype(at::kByte));
+  int blocks = (int)std::min<long>(2048, cdiv(n, 256));
+  hipLaunchKernelGGL(dropout_fwd_k, dim3(blocks), dim3(256), 0, cur_stream(),
+                     x.data_ptr<float>(), y.data_ptr<float>(),
+                     mask.data_ptr<unsigned char>(), n, hw, (float)p,
+                     (unsigned long long)seed, (unsigned long long)ctr);
+  return {y, mask};
+}
+
+at::Tensor dropout_bwd(at::Tensor dy, at::Tensor mask, double p,
+                       bool channelwise) {
+  CHECK_F32(dy);
+  long n = dy.numel();
+  long hw = channelwise ? (long)dy.size(2) * dy.size(3) : 1;
+  auto dx = at::empty_like(dy);
+  int blocks = (int)std::min<long>(2048, cdiv(n, 256));
+  hipLaunchKernelGGL(dropout_bwd_k, dim3(blocks), dim3(256), 0, cur_stream(),
+                     dy.data_ptr<float>(), mask.data_ptr<unsigned char>(),
+                     dx.data_ptr<float>(), n, hw, (float)p);
+  return dx;
+}
+
+// ---------------------------------------------------------------------------
+void fused_sgd_step(at::Tensor ptrs, at::Tensor chunks, long chunk_elems,
+                    double lr, double momentum) {
+  CHECK_I64(ptrs);
+  TORCH_CHECK(chunks.is_cuda() && chunks.scalar_type() == at::kInt);
+  int nc = chunks.size(0);
+  hipLaunchKernelGGL(fused_sgd_k, dim3(nc), dim3(256), 0, cur_stream(),
+                     ptrs.data_ptr<long>(), chunks.data_ptr<int>(),
+                     (int)chunk_elems, (float)lr, (float)momentum);
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("conv2d_fwd", &conv2d_fwd);
+  m.def("conv2d_bwd_input", &conv2d_bwd_input);
+  m.def("conv2d_bwd_weight", &conv2d_bwd_weight);
+  m.def("linear_fwd", &linear_fwd);
+  m.def("linear_bwd", &linear_bwd);
+  m.def("maxpool2x2_relu_fwd", &maxpool2x2_relu_fwd);
+  m.def("maxpool2x2_relu_bwd", &maxpool2x2_relu_bwd);
+  m.def("relu_fwd", &relu_fwd);
+  m.def("relu_bwd", &relu_bwd);
+  m.def("logsoftmax_nll_fwd", &logsoftmax_nll_fwd);
+  m.def("logsoftmax_nll_bwd", &logsoftmax_nll_bwd);
+  m.def("logsoftmax_fwd", &logsoftmax_fwd);
+  m.def("logsoftmax_bwd", &logsoftmax_bwd);
+  m.def("dropout_fwd", &dropout_fwd);
+  m.def("dropout_bwd", &dropout_bwd);
+  m.def("fused_sgd_step", &fused_sgd_step);
+}
