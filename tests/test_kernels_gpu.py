@@ -221,15 +221,15 @@ def test_training_loss_decreases_gpu():
     from dist_mnist_amd.ops import SGD
 
     torch.manual_seed(22)
-    net = Net().to(DEV).train()
+    net = Net().to(DEV).eval()  # dropout off: deterministic memorization
     opt = SGD(net.parameters(), lr=0.1, momentum=0.5)
-    x = rand(256, 1, 28, 28, seed=23)
-    t = torch.randint(0, 10, (256,), device=DEV)
+    x = rand(64, 1, 28, 28, seed=23)
+    t = torch.randint(0, 10, (64,), device=DEV)
     losses = []
-    for _ in range(40):
+    for _ in range(60):
         opt.zero_grad()
         _, loss = net(x, t)
         loss.backward()
         opt.step()
         losses.append(loss.item())
-    assert losses[-1] < losses[0] * 0.8, losses[::10]
+    assert losses[-1] < losses[0] * 0.7, losses[::10]
