@@ -254,19 +254,24 @@ __global__ __launch_bounds__(256) void conv2d_dgrad_f32_k(
 }
 
 // ---------------------------------------------------------------------------
-// Backward-weight (wgrad) + bias. D[k-tile][co-tile], reduce over an M-split
-// chunk per block, atomicAdd f32 into dw/db (zero-initialized by the caller).
-// dy chunk is LDS-transposed so its MFMA reads are conflict-free; per-chunk
-// pixel decode is precomputed into LDS once (m -> x base addr / bounds).
+// Backward-weight (wgrad) + bias. Each block owns a [k-tile x 16co] output
+// tile and reduces an M-split; the 4 waves stream DISJOINT m-chunks in
+// parallel (full wave utilization at any K, unlike a waves-on-k split that
+// idles 3/4 of the block for K < 64), each accumulating the full k-tile
+// (KT 16-row frags). Wave partials are reduced through LDS so each block
+// issues ONE atomicAdd per dw element (atomic traffic matters: dw is tiny and
+// thousands of M-split blocks hit the same cache lines).
+// dy chunk is LDS-transposed (coalesced loads, conflict-free MFMA reads);
+// per-m pixel decode is precomputed into LDS once per chunk.
 // ---------------------------------------------------------------------------
-template <bool BOUNDS, bool WANT_BIAS>
+template <bool BOUNDS, bool WANT_BIAS, int KT>
 __global__ __launch_bounds__(256) void conv2d_wgrad_f32_k(
     const float* __restrict__ dy, const float* __restrict__ x,
     float* __restrict__ dw, float* __restrict__ db,
     int CI, int H, int W, int CO, int KH, int KW, int OH, int OW,
     int stride, int pad, int M, int K, int m_per_split) {
-  constexpr int MC = 64;  // m chunk
-  const int k0 = blockIdx.x * 64;  // 4 waves x 16 k rows
+  constexpr int MC = 256;  // super-chunk: 4 waves x 64 m
+  const int k0 = blockIdx.x * (KT * 16);
   const int co0 = blockIdx.y * 16;
   const int ms = blockIdx.z;
   const int tid = threadIdx.x;
@@ -275,22 +280,28 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_f32_k(
   const int lm = l & 15;
   const int lk = l >> 4;
 
-  __shared__ float lds_dy[MC][17];  // [m][co], padded
+  __shared__ float lds_dy[MC][17];  // [m][co] transposed, padded
   __shared__ int lds_mbase[MC];     // n*CI*HW + ih0*W + iw0
-  __shared__ int lds_mhw[MC];       // (ih0+pad)<<16 | (iw0+pad)  (for bounds)
+  __shared__ int lds_mhw[MC];       // biased (ih0, iw0) for bounds
 
   const int HW = H * W;
   const int OHOW = OH * OW;
 
-  // this wave's k row (A row = lm): fixed over the whole reduction
-  int k = k0 + wid * 16 + lm;
-  bool kval = k < K;
-  int kcl = kval ? k : 0;
-  int ci = kcl / (KH * KW), kr = kcl % (KH * KW);
-  int kh = kr / KW, kw = kr % KW;
-  int koff = ci * HW + kh * W + kw;
+  // per-lane k decode (A row = lm), fixed over the whole reduction
+  int koff[KT], khh[KT], kww[KT];
+  bool kval[KT];
+#pragma unroll
+  for (int kf = 0; kf < KT; ++kf) {
+    int k = k0 + kf * 16 + lm;
+    kval[kf] = k < K;
+    int kcl = kval[kf] ? k : 0;
+    int ci = kcl / (KH * KW), kr = kcl % (KH * KW);
+    khh[kf] = kr / KW;
+    kww[kf] = kr % KW;
+    koff[kf] = ci * HW + khh[kf] * W + kww[kf];
+  }
 
-  f32x4 acc = {};
+  f32x4 acc[KT] = {};
   float db_part = 0.f;
 
   int m_begin = ms * m_per_split;
@@ -298,11 +309,10 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_f32_k(
 
   for (int mc = m_begin; mc < m_end; mc += MC) {
     __syncthreads();
-    // stage dy[mc:mc+64][co0:co0+16] transposed into LDS (coalesced over m)
+    // stage dy[mc:mc+256][co0:+16] transposed (reads coalesced over m)
     for (int idx = tid; idx < MC * 16; idx += 256) {
-      int co = idx >> 6;        // 0..15? MC*16/64: idx/64 -> co? careful:
-      int mloc = idx & 63;      // we want m fastest for coalescing
-      co = idx >> 6;            // idx = co*64 + mloc
+      int co = idx >> 8;  // idx = co*256 + mloc
+      int mloc = idx & 255;
       int m = mc + mloc;
       float v = 0.f;
       if (m < m_end && co0 + co < CO) {
@@ -319,43 +329,72 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_f32_k(
       int oh = r / OW, ow = r % OW;
       int ih0 = oh * stride - pad, iw0 = ow * stride - pad;
       lds_mbase[tid] = n * CI * HW + ih0 * W + iw0;
-      lds_mhw[tid] = ((ih0 + 16384) << 16) | (iw0 + 16384);  // biased for bounds
+      lds_mhw[tid] = ((ih0 + 16384) << 16) | (iw0 + 16384);
     }
     __syncthreads();
 
-    int mlim = min(MC, m_end - mc);
+    // wave `wid` consumes m rows [wid*64, wid*64+64)
+    int wbase = wid * 64;
+    int mlim = min(64, m_end - mc - wbase);  // may be negative on tail
     for (int mm = 0; mm < mlim; mm += 4) {
       bool min_ok = (mm + lk) < mlim;
-      int msel = min_ok ? mm + lk : 0;
-      float b = lds_dy[msel][lm];            // B[k=m][col=co]
-      float a = 0.f;
-      if (kval & min_ok) {
-        bool ok = true;
-        if (BOUNDS) {
-          int hw = lds_mhw[msel];
-          int ih = (hw >> 16) - 16384 + kh;
-          int iw = (hw & 0xffff) - 16384 + kw;
-          ok = ih >= 0 && ih < H && iw >= 0 && iw < W;
+      int msel = wbase + (min_ok ? mm + lk : 0);
+      float b = min_ok ? lds_dy[msel][lm] : 0.f;  // B[k=m][col=co]
+      if (WANT_BIAS) db_part += b;
+      int mb = lds_mbase[msel];
+      int hw = lds_mhw[msel];
+#pragma unroll
+      for (int kf = 0; kf < KT; ++kf) {
+        float a = 0.f;
+        if (kval[kf] & min_ok) {
+          bool ok = true;
+          if (BOUNDS) {
+            int ih = (hw >> 16) - 16384 + khh[kf];
+            int iw = (hw & 0xffff) - 16384 + kww[kf];
+            ok = ih >= 0 && ih < H && iw >= 0 && iw < W;
+          }
+          if (ok) a = x[mb + koff[kf]];
         }
-        if (ok) a = x[lds_mbase[msel] + koff];
+        acc[kf] = mfma16x16x4(a, b, acc[kf]);
       }
-      if (!min_ok) b = 0.f;
-      acc = mfma16x16x4(a, b, acc);
-      if (WANT_BIAS) db_part += b;  // sum over m of dy[m][co=lm] (lk-partial)
     }
   }
 
-  // dw[co][k] atomic accumulate: D row = lk*4+reg -> k-local, col = lm -> co
+  // cross-wave reduction in LDS (reuse lds_dy: 4 waves x KT x 256 floats)
+  float* red = &lds_dy[0][0];  // >= 4*KT*256 floats available (KT<=4)
+  __syncthreads();
 #pragma unroll
-  for (int reg = 0; reg < 4; ++reg) {
-    int kk = k0 + wid * 16 + lk * 4 + reg;
-    int co = co0 + lm;
-    if (kk < K && co < CO) atomicAdd(&dw[co * K + kk], acc[reg]);
+  for (int kf = 0; kf < KT; ++kf) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      red[(wid * KT + kf) * 256 + (lk * 4 + reg) * 16 + lm] = acc[kf][reg];
+    }
   }
-  if (WANT_BIAS && blockIdx.x == 0 && wid == 0) {
-    // reduce db_part across the 4 lk groups of wave 0
+  __syncthreads();
+  // wave 0 folds the other 3 waves and does the single atomicAdd set
+  if (wid == 0) {
+#pragma unroll
+    for (int kf = 0; kf < KT; ++kf) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        int e = (lk * 4 + reg) * 16 + lm;
+        float v = red[kf * 256 + e] + red[(KT + kf) * 256 + e] +
+                  red[(2 * KT + kf) * 256 + e] + red[(3 * KT + kf) * 256 + e];
+        int kk = k0 + kf * 16 + lk * 4 + reg;
+        int co = co0 + lm;
+        if (kk < K && co < CO) atomicAdd(&dw[co * K + kk], v);
+      }
+    }
+  }
+  if (WANT_BIAS && blockIdx.x == 0) {
+    // db: every wave summed its own m rows; fold lk groups then waves
     db_part += __shfl_down(db_part, 16);
     db_part += __shfl_down(db_part, 32);
-    if (lk == 0 && co0 + lm < CO) atomicAdd(&db[co0 + lm], db_part);
+    __syncthreads();
+    if (lk == 0) red[wid * 16 + lm] = db_part;
+    __syncthreads();
+    if (tid < 16 && co0 + tid < CO)
+      atomicAdd(&db[co0 + tid],
+                red[tid] + red[16 + tid] + red[32 + tid] + red[48 + tid]);
   }
 }

@@ -87,14 +87,17 @@ std::vector<at::Tensor> conv2d_bwd_weight(at::Tensor dy, at::Tensor x, long KH,
   auto dw = at::zeros({CO, CI, KH, KW}, x.options());
   auto db = at::zeros({want_bias ? CO : 0}, x.options());
   int M = N * OH * OW, K = CI * KH * KW;
-  int kt = cdiv(K, 64), ct = cdiv(CO, 16);
-  int ms = std::max(1, std::min(2048 / std::max(1, kt * ct), cdiv(M, 64)));
-  int m_per_split = cdiv(cdiv(M, ms), 64) * 64;
+  bool small_k = K <= 32;
+  int ktile = small_k ? 32 : 64;
+  int kt = cdiv(K, ktile), ct = cdiv(CO, 16);
+  int ms = std::max(1, std::min(2048 / std::max(1, kt * ct), cdiv(M, 256)));
+  int m_per_split = cdiv(cdiv(M, ms), 256) * 256;
   ms = cdiv(M, m_per_split);
   dim3 grid(kt, ct, ms);
   bool bounds = pad > 0;
-  auto launch = [&](auto bd, auto wb) {
-    hipLaunchKernelGGL((conv2d_wgrad_f32_k<decltype(bd)::value, decltype(wb)::value>),
+  auto launch = [&](auto bd, auto wb, auto ktc) {
+    hipLaunchKernelGGL((conv2d_wgrad_f32_k<decltype(bd)::value, decltype(wb)::value,
+                                           decltype(ktc)::value>),
                        grid, dim3(256), 0, cur_stream(), dy.data_ptr<float>(),
                        x.data_ptr<float>(), dw.data_ptr<float>(),
                        want_bias ? db.data_ptr<float>() : nullptr, CI, H, W, CO,
@@ -102,10 +105,19 @@ std::vector<at::Tensor> conv2d_bwd_weight(at::Tensor dy, at::Tensor x, long KH,
   };
   using T = std::true_type;
   using F = std::false_type;
-  if (bounds && want_bias) launch(T{}, T{});
-  else if (bounds) launch(T{}, F{});
-  else if (want_bias) launch(F{}, T{});
-  else launch(F{}, F{});
+  using K2 = std::integral_constant<int, 2>;
+  using K4 = std::integral_constant<int, 4>;
+  if (small_k) {
+    if (bounds && want_bias) launch(T{}, T{}, K2{});
+    else if (bounds) launch(T{}, F{}, K2{});
+    else if (want_bias) launch(F{}, T{}, K2{});
+    else launch(F{}, F{}, K2{});
+  } else {
+    if (bounds && want_bias) launch(T{}, T{}, K4{});
+    else if (bounds) launch(T{}, F{}, K4{});
+    else if (want_bias) launch(F{}, T{}, K4{});
+    else launch(F{}, F{}, K4{});
+  }
   return {dw, db};
 }
 
