@@ -371,8 +371,11 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_f32_k(
     }
   }
   __syncthreads();
-  // wave 0 folds the other 3 waves and does the single atomicAdd set
+  // wave 0 folds the other 3 waves; result goes to this split's PARTIAL slab
+  // (dw points at workspace [n_splits][CO*K] — plain stores, no atomics: the
+  // deterministic reduce kernel below folds the split axis in fixed order).
   if (wid == 0) {
+    float* slab = dw + (long)ms * (CO * (long)K);
 #pragma unroll
     for (int kf = 0; kf < KT; ++kf) {
 #pragma unroll
@@ -382,7 +385,7 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_f32_k(
                   red[(2 * KT + kf) * 256 + e] + red[(3 * KT + kf) * 256 + e];
         int kk = k0 + kf * 16 + lk * 4 + reg;
         int co = co0 + lm;
-        if (kk < K && co < CO) atomicAdd(&dw[co * K + kk], v);
+        if (kk < K && co < CO) slab[co * K + kk] = v;
       }
     }
   }
@@ -394,7 +397,19 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_f32_k(
     if (lk == 0) red[wid * 16 + lm] = db_part;
     __syncthreads();
     if (tid < 16 && co0 + tid < CO)
-      atomicAdd(&db[co0 + tid],
-                red[tid] + red[16 + tid] + red[32 + tid] + red[48 + tid]);
+      db[(long)ms * CO + co0 + tid] =
+          red[tid] + red[16 + tid] + red[32 + tid] + red[48 + tid];
+  }
+}
+
+// Fold the split axis of the wgrad partial slabs in fixed order
+// (deterministic — no float-atomic reassociation): out[e] = sum_s in[s*E+e].
+__global__ void reduce_splits_k(const float* __restrict__ in,
+                                float* __restrict__ out, long E, int S) {
+  for (long e = (long)blockIdx.x * blockDim.x + threadIdx.x; e < E;
+       e += (long)gridDim.x * blockDim.x) {
+    float v = 0.f;
+    for (int s = 0; s < S; ++s) v += in[s * E + e];
+    out[e] = v;
   }
 }

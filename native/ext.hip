@@ -84,23 +84,28 @@ std::vector<at::Tensor> conv2d_bwd_weight(at::Tensor dy, at::Tensor x, long KH,
   CHECK_F32(x);
   int N = x.size(0), CI = x.size(1), H = x.size(2), W = x.size(3);
   int CO = dy.size(1), OH = dy.size(2), OW = dy.size(3);
-  auto dw = at::zeros({CO, CI, KH, KW}, x.options());
-  auto db = at::zeros({want_bias ? CO : 0}, x.options());
+  auto dw = at::empty({CO, CI, KH, KW}, x.options());
+  auto db = at::empty({want_bias ? CO : 0}, x.options());
   int M = N * OH * OW, K = CI * KH * KW;
   bool small_k = K <= 32;
   int ktile = small_k ? 32 : 64;
   int kt = cdiv(K, ktile), ct = cdiv(CO, 16);
-  int ms = std::max(1, std::min(2048 / std::max(1, kt * ct), cdiv(M, 256)));
+  long dw_elems = (long)CO * K;
+  int ms_cap = (int)std::max<long>(1, (256L << 20) / (4 * dw_elems));
+  int ms = std::max(1, std::min({2048 / std::max(1, kt * ct), cdiv(M, 256), ms_cap}));
   int m_per_split = cdiv(cdiv(M, ms), 256) * 256;
   ms = cdiv(M, m_per_split);
   dim3 grid(kt, ct, ms);
   bool bounds = pad > 0;
+  // partial slabs per M-split; folded deterministically below
+  auto dw_ws = at::empty({ms, dw_elems}, x.options());
+  auto db_ws = at::empty({want_bias ? ms : 0, want_bias ? CO : 0}, x.options());
   auto launch = [&](auto bd, auto wb, auto ktc) {
     hipLaunchKernelGGL((conv2d_wgrad_f32_k<decltype(bd)::value, decltype(wb)::value,
                                            decltype(ktc)::value>),
                        grid, dim3(256), 0, cur_stream(), dy.data_ptr<float>(),
-                       x.data_ptr<float>(), dw.data_ptr<float>(),
-                       want_bias ? db.data_ptr<float>() : nullptr, CI, H, W, CO,
+                       x.data_ptr<float>(), dw_ws.data_ptr<float>(),
+                       want_bias ? db_ws.data_ptr<float>() : nullptr, CI, H, W, CO,
                        KH, KW, OH, OW, (int)stride, (int)pad, M, K, m_per_split);
   };
   using T = std::true_type;
@@ -117,6 +122,14 @@ std::vector<at::Tensor> conv2d_bwd_weight(at::Tensor dy, at::Tensor x, long KH,
     else if (bounds) launch(T{}, F{}, K4{});
     else if (want_bias) launch(F{}, T{}, K4{});
     else launch(F{}, F{}, K4{});
+  }
+  {
+    int blocks = (int)std::min<long>(2048, cdiv(dw_elems, 256));
+    hipLaunchKernelGGL(reduce_splits_k, dim3(blocks), dim3(256), 0, cur_stream(),
+                       dw_ws.data_ptr<float>(), dw.data_ptr<float>(), dw_elems, ms);
+    if (want_bias)
+      hipLaunchKernelGGL(reduce_splits_k, dim3(1), dim3(256), 0, cur_stream(),
+                         db_ws.data_ptr<float>(), db.data_ptr<float>(), CO, ms);
   }
   return {dw, db};
 }

@@ -222,14 +222,15 @@ def test_training_loss_decreases_gpu():
 
     torch.manual_seed(22)
     net = Net().to(DEV).eval()  # dropout off: deterministic memorization
-    opt = SGD(net.parameters(), lr=0.1, momentum=0.5)
+    opt = SGD(net.parameters(), lr=0.05, momentum=0.5)
     x = rand(64, 1, 28, 28, seed=23)
     t = torch.randint(0, 10, (64,), device=DEV)
     losses = []
-    for _ in range(60):
+    for _ in range(150):
         opt.zero_grad()
         _, loss = net(x, t)
         loss.backward()
         opt.step()
         losses.append(loss.item())
-    assert losses[-1] < losses[0] * 0.7, losses[::10]
+    tail = sum(losses[-10:]) / 10
+    assert tail < losses[0] * 0.7, (losses[0], tail, losses[::30])
