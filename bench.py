@@ -37,8 +37,9 @@ def parse_args():
     p.add_argument("--warmup", type=int, default=10)
     p.add_argument("--batch", type=int, default=None,
                    help="per-GPU batch (default 8192 on GPU — BASELINE config 4 shape)")
-    p.add_argument("--dtype", type=str, default="fp32", choices=["fp32"],
-                   help="compute dtype of the hot path (bf16 path: later milestone)")
+    p.add_argument("--dtype", type=str, default="bf16", choices=["fp32", "bf16"],
+                   help="compute dtype of the hot path (bf16 = NHWC native engine, "
+                        "the BASELINE config-4 headline; fp32 = NCHW engine)")
     p.add_argument("--bucket-mb", type=float, default=25.0)
     return p.parse_args()
 
@@ -57,16 +58,21 @@ def main():
         torch.cuda.set_device(device)
 
     batch = args.batch or (8192 if use_cuda else 256)
+    use_bf16 = args.dtype == "bf16" and use_cuda
     torch.manual_seed(1234)
 
     model = Net().to(device).train()
+    if use_bf16:
+        model.bf16_native_()
     if world > 1:
         model = DistributedDataParallel(model, bucket_bytes=int(args.bucket_mb * 2**20))
     opt = SGD(model.parameters(), lr=0.01, momentum=0.5)
 
     stream = SyntheticBatchStream(batch_size=batch, n_pool=4, seed=1234 + rank,
-                                  pin=use_cuda)
-    batches = Prefetcher(stream, device)
+                                  pin=use_cuda,
+                                  layout="nhwc_bf16" if use_bf16 else "nchw_f32")
+    batches = Prefetcher(stream, device,
+                         dtype=torch.bfloat16 if use_bf16 else torch.float32)
 
     def one_step():
         x, t = next(batches)
@@ -120,7 +126,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,  # reference publishes no numbers
-            "dtype": args.dtype,
+            "dtype": "bf16" if use_bf16 else "fp32",
             "data": "synthetic",
             "final_loss": float(last_loss.item()) if last_loss is not None else None,
             "config": {

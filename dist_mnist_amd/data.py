@@ -109,15 +109,27 @@ def partition_dataset(rank: int, world_size: int, global_batch: int = GLOBAL_BAT
 class SyntheticBatchStream:
     """Infinite synthetic batch stream for benchmarking: pre-generates a pool
     of batches once, then cycles. Pairs with runtime.Prefetcher for pinned
-    async H2D staging on a side stream."""
+    async H2D staging on a side stream.
+
+    layout="nchw_f32": reference layout [N,1,28,28] fp32.
+    layout="nhwc_bf16": the native bf16 engine's input layout — NHWC bf16
+    with channels padded to 8 ([N,28,28,8], image in channel 0) — built once
+    on the host so the device never runs a repack kernel.
+    """
 
     def __init__(self, batch_size: int, n_pool: int = 4, seed: int = 1234,
-                 dtype=torch.float32, pin: bool = False):
+                 dtype=torch.float32, pin: bool = False, layout: str = "nchw_f32"):
         g = torch.Generator().manual_seed(seed)
         self.pool = []
         for _ in range(n_pool):
-            x = torch.randn(batch_size, 1, 28, 28, generator=g, dtype=torch.float32).to(dtype)
+            x = torch.randn(batch_size, 1, 28, 28, generator=g, dtype=torch.float32)
             y = torch.randint(0, 10, (batch_size,), generator=g)
+            if layout == "nhwc_bf16":
+                xp = torch.zeros(batch_size, 28, 28, 8, dtype=torch.bfloat16)
+                xp[..., 0] = x[:, 0].to(torch.bfloat16)
+                x = xp
+            else:
+                x = x.to(dtype)
             if pin:
                 x = x.pin_memory()
                 y = y.pin_memory()

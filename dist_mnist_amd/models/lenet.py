@@ -36,8 +36,20 @@ class Net(nn.Module):
         self.conv2_drop = nn.Dropout2d()
         self.fc1 = nn.Linear(320, 50)
         self.fc2 = nn.Linear(50, 10)
+        self._bf16_native = False
+        self._fc1_perm = None
+        self._fc2_perm = None
+
+    def bf16_native_(self, enable: bool = True):
+        """Switch forward to the bf16 NHWC kernel path (GPU only). Inputs
+        must then be NHWC bf16 [N, 28, 28, 8] (see ops.functional_bf16).
+        Master weights stay fp32 canonical — checkpoints are unchanged."""
+        self._bf16_native = enable
+        return self
 
     def forward(self, x, target=None):
+        if self._bf16_native and x.is_cuda:
+            return self._forward_bf16(x, target)
         x = ops.conv2d(x, self.conv1.weight, self.conv1.bias)
         x = ops.maxpool2x2_relu(x)
         x = ops.conv2d(x, self.conv2.weight, self.conv2.bias)
@@ -51,3 +63,32 @@ class Net(nn.Module):
             # fused LogSoftmax+NLLLoss: returns (log_probs, loss)
             return ops.logsoftmax_nll(x, target)
         return ops.log_softmax(x)
+
+    def _forward_bf16(self, x, target=None):
+        from ..ops import functional_bf16 as fb
+
+        if self._fc1_perm is None or self._fc1_perm.device != x.device:
+            # pool2 output is NHWC [4, 4, 24]; reference flatten order is
+            # canonical NCHW (c, h, w) -> perm maps padded k -> canonical ci
+            self._fc1_perm = fb.build_flatten_perm(4, 4, 24, 20, x.device)
+            self._fc2_perm = fb.build_identity_perm(56, 50, x.device)
+        x = fb.conv2d_bf16(x, self.conv1.weight, self.conv1.bias, cop=16)
+        x = fb.maxpool2x2_relu_bf16(x)
+        x = fb.conv2d_bf16(x, self.conv2.weight, self.conv2.bias, cop=24)
+        x = fb.dropout_bf16(x, p=self.conv2_drop.p, training=self.training,
+                            channelwise=True)
+        x = fb.maxpool2x2_relu_bf16(x)
+        x = x.reshape(-1, 4 * 4 * 24)
+        x = fb.linear_bf16(x, self.fc1.weight, self.fc1.bias,
+                           perm=self._fc1_perm, cop=56, relu=True)
+        x = fb.dropout_bf16(x, p=0.5, training=self.training)
+        x = fb.linear_bf16(x, self.fc2.weight, self.fc2.bias,
+                           perm=self._fc2_perm, cop=16)
+        if target is not None:
+            from ..ops.functional_bf16 import logsoftmax_nll_bf16
+
+            return logsoftmax_nll_bf16(x, target, c=10)
+        # standalone log-probs (eval): reuse the fused kernel's fwd only
+        logp, _ = fb.logsoftmax_nll_bf16(x, torch.zeros(x.shape[0], dtype=torch.long,
+                                                        device=x.device), c=10)
+        return logp

@@ -1,0 +1,921 @@
+// bf16 NHWC engine — the MI355X-first hot path (BASELINE.json config 4).
+//
+// Layout contract:
+//  - activations: NHWC bf16, channels padded to a multiple of 8 ("Cp") so an
+//    MFMA A-fragment (8 contiguous k) is ONE 16-byte global load;
+//  - GEMM k order for conv: k = kh*(KW*Cp) + kw*Cp + ci  — contiguous in x
+//    within each kh row because Cp % 8 == 0;
+//  - master weights fp32 canonical NCHW (checkpoint-compatible); converted to
+//    bf16 + repacked during LDS staging (weights are tiny and L2-resident);
+//  - accumulation fp32 (MFMA C/D); weight grads fp32 canonical.
+//
+// MFMA: v_mfma_f32_16x16x32_bf16 (guide §3):
+//   A: lane l holds A[l&15][(l>>4)*8 + j], j=0..7 (8 contiguous k, 4 VGPRs)
+//   B: lane l holds B[(l>>4)*8 + j][l&15]
+//   C/D: col = l&15, row = (l>>4)*4 + reg  (f32x4)
+// (verified on-device by mfma_probe_16x16x32, tests/test_kernels_gpu.py)
+#include "common.h"
+#include <hip/hip_bf16.h>
+
+using bf16 = __bf16;
+using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
+using bf16x4 = __attribute__((ext_vector_type(4))) __bf16;
+
+DEV f32x4 mfma16x16x32bf(bf16x8 a, bf16x8 b, f32x4 c) {
+  return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+}
+
+DEV bf16 to_bf16(float f) { return (bf16)f; }
+DEV float to_f32(bf16 h) { return (float)h; }
+
+// zero-filled bf16x8
+DEV bf16x8 bf8_zero() {
+  bf16x8 z = {};
+  return z;
+}
+
+// ---------------------------------------------------------------------------
+// Mapping probe: one wave computes D = A(16x32) @ B(32x16) via the assumed
+// fragment mapping; the GPU test compares against torch.matmul.
+// ---------------------------------------------------------------------------
+__global__ void mfma_probe_16x16x32_k(const float* __restrict__ A,
+                                      const float* __restrict__ B,
+                                      float* __restrict__ D) {
+  int l = threadIdx.x;
+  if (l >= 64) return;
+  int lm = l & 15, lk = l >> 4;
+  bf16x8 a, b;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    a[j] = to_bf16(A[lm * 32 + lk * 8 + j]);      // A[row][k]
+    b[j] = to_bf16(B[(lk * 8 + j) * 16 + lm]);    // B[k][col]
+  }
+  f32x4 c = {};
+  c = mfma16x16x32bf(a, b, c);
+#pragma unroll
+  for (int reg = 0; reg < 4; ++reg) D[(lk * 4 + reg) * 16 + lm] = c[reg];
+}
+
+// ---------------------------------------------------------------------------
+// Conv2d forward, NHWC bf16.
+//   D[m][co]: A = x patches (ONE predicated 16B load per lane per k-step),
+//   B = weights repacked fp32->bf16 into LDS per BK=32 chunk.
+// Block = 256 threads = 4 waves, each wave MW m-frags => BM = 4*MW*16 rows;
+// grid.y covers 16-wide co tiles. Stores NHWC (co fastest), bf16.
+// Bias fp32; optional fused ReLU (ResNet); co >= CO stores 0 (keeps channel
+// padding clean for the next layer's A loads).
+// ---------------------------------------------------------------------------
+template <bool HAS_BIAS, bool RELU, bool BOUNDS, int MW>
+__global__ __launch_bounds__(256) void conv2d_fwd_bf16_k(
+    const bf16* __restrict__ x, const float* __restrict__ w,
+    const float* __restrict__ bias, bf16* __restrict__ y,
+    int CIp, int CI, int H, int W, int COp, int CO, int KH, int KW,
+    int OH, int OW, int stride, int pad, int M, int K) {
+  constexpr int BK = 32;
+  const int BM = 4 * MW * 16;
+  const int m0 = blockIdx.x * BM;
+  const int co0 = blockIdx.y * 16;
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6;
+  const int l = tid & 63;
+  const int lm = l & 15;
+  const int lk = l >> 4;
+
+  // lds_w[co][k-chunk]: row stride 40 bf16 (80B) => conflict-free b128 reads
+  __shared__ bf16 lds_w[16][BK + 8];
+  __shared__ int lds_koff[BK / 8];   // element offset of each 8-k group
+  __shared__ int lds_khw[BK / 8];    // (kh<<16)|kw of each group (uniform in group)
+
+  const int KWC = KW * CIp;          // k-row length (multiple of 8)
+  const int KHKW = KH * KW;
+
+  // per-(lane, mi) A-side pixel decode (m = ... + lm), fixed across K loop
+  int xbase[MW];   // element offset of x[n][ih0][iw0][0]
+  int ih0[MW], iw0[MW];
+  bool mval[MW];
+#pragma unroll
+  for (int mi = 0; mi < MW; ++mi) {
+    int m = m0 + (wid * MW + mi) * 16 + lm;
+    mval[mi] = m < M;
+    int mm = mval[mi] ? m : 0;
+    int n = mm / (OH * OW), r = mm % (OH * OW);
+    int oh = r / OW, ow = r % OW;
+    ih0[mi] = oh * stride - pad;
+    iw0[mi] = ow * stride - pad;
+    xbase[mi] = ((n * H + ih0[mi]) * W + iw0[mi]) * CIp;
+  }
+
+  f32x4 acc[MW] = {};
+
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    __syncthreads();
+    // stage weights: lds_w[co][kk] = bf16(w[co0+co][ci][kh][kw]) for k=k0+kk
+    for (int idx = tid; idx < 16 * BK; idx += 256) {
+      int co = idx >> 5;       // 0..15 (idx = co*32 + kk)
+      int kk = idx & 31;
+      int k = k0 + kk;
+      float v = 0.f;
+      if (k < K && co0 + co < CO) {
+        int kh = k / KWC, r = k % KWC;
+        int kw = r / CIp, ci = r % CIp;
+        if (ci < CI)
+          v = w[((co0 + co) * CI + ci) * KHKW + kh * KW + kw];
+      }
+      lds_w[co][kk] = to_bf16(v);
+    }
+    if (tid < BK / 8) {
+      int k = k0 + tid * 8;
+      int kcl = k < K ? k : 0;
+      int kh = kcl / KWC, r = kcl % KWC;
+      int kw = r / CIp, ci0 = r % CIp;
+      lds_koff[tid] = (kh * W + kw) * CIp + ci0;
+      lds_khw[tid] = (kh << 16) | kw;
+    }
+    __syncthreads();
+
+    int kgroups = min(BK, K - k0) / 8;  // K is always a multiple of 8
+    for (int kg = 0; kg < kgroups; kg += 4) {
+      bool gin = lk < (kgroups - kg);
+      int gsel = gin ? kg + lk : 0;
+      // B-frag: 8 contiguous k of one co row from LDS (16B read)
+      bf16x8 b = gin ? *reinterpret_cast<const bf16x8*>(&lds_w[lm][gsel * 8])
+                     : bf8_zero();
+      int koff = lds_koff[gsel];
+      int khw = lds_khw[gsel];
+      int kh = khw >> 16, kw = khw & 0xffff;
+#pragma unroll
+      for (int mi = 0; mi < MW; ++mi) {
+        bool ok = mval[mi] & gin;
+        if (BOUNDS) {
+          int ih = ih0[mi] + kh, iw = iw0[mi] + kw;
+          ok = ok && ih >= 0 && ih < H && iw >= 0 && iw < W;
+        }
+        bf16x8 a = ok ? *reinterpret_cast<const bf16x8*>(&x[xbase[mi] + koff])
+                      : bf8_zero();
+        // D[row=m][col=co]: patch supplies rows, weights supply columns.
+        acc[mi] = mfma16x16x32bf(a, b, acc[mi]);
+      }
+    }
+  }
+
+  // epilogue: D row = lk*4+reg -> m, col = lm -> co. A 16-lane group writes
+  // 16 consecutive co of one pixel => contiguous 32B NHWC stores.
+#pragma unroll
+  for (int mi = 0; mi < MW; ++mi) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      int m = m0 + (wid * MW + mi) * 16 + lk * 4 + reg;
+      if (m >= M) continue;
+      int co = co0 + lm;
+      if (co >= COp) continue;
+      int n = m / (OH * OW), r = m % (OH * OW);
+      float v = (co < CO) ? acc[mi][reg] : 0.f;
+      if (HAS_BIAS && co < CO) v += bias[co];
+      if (RELU) v = fmaxf(v, 0.f);
+      y[(long)(n * (OH * OW) + r) * COp + co] = to_bf16(v);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Conv2d backward-input (dgrad), NHWC bf16.
+//   D[m'=(n,ih,iw)][ci]: A = dy patches (k' = kh*(KW*COp)+kw*COp+co, so an
+//   A-frag is 8 contiguous co of one (kh,kw) tap — ONE predicated 16B load),
+//   B = weights staged fp32->bf16 into LDS as [ci][k'] rows.
+//   oh = (ih + pad - kh)/stride when divisible & in range, else 0.
+// ---------------------------------------------------------------------------
+template <bool S1, int MW>
+__global__ __launch_bounds__(256) void conv2d_dgrad_bf16_k(
+    const bf16* __restrict__ dy, const float* __restrict__ w,
+    bf16* __restrict__ dx,
+    int CIp, int CI, int H, int W, int COp, int CO, int KH, int KW,
+    int OH, int OW, int stride, int pad, int Mp, int Kp) {
+  constexpr int BK = 32;
+  const int BM = 4 * MW * 16;
+  const int m0 = blockIdx.x * BM;
+  const int ci0 = blockIdx.y * 16;
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6;
+  const int l = tid & 63;
+  const int lm = l & 15;
+  const int lk = l >> 4;
+
+  __shared__ bf16 lds_w[16][BK + 8];   // [ci][k'] rows
+  __shared__ int lds_kdec[BK / 8];     // (kh<<20)|(kw<<8)|co0_of_group... packed
+
+  const int KWC = KW * COp;
+  const int KHKW = KH * KW;
+
+  int ih[MW], iw[MW], nn[MW];
+  bool mval[MW];
+#pragma unroll
+  for (int mi = 0; mi < MW; ++mi) {
+    int m = m0 + (wid * MW + mi) * 16 + lm;
+    mval[mi] = m < Mp;
+    int mm = mval[mi] ? m : 0;
+    nn[mi] = mm / (H * W);
+    int r = mm % (H * W);
+    ih[mi] = r / W;
+    iw[mi] = r % W;
+  }
+
+  f32x4 acc[MW] = {};
+
+  for (int k0 = 0; k0 < Kp; k0 += BK) {
+    __syncthreads();
+    for (int idx = tid; idx < 16 * BK; idx += 256) {
+      int ci = idx >> 5;
+      int kk = idx & 31;
+      int k = k0 + kk;
+      float v = 0.f;
+      if (k < Kp && ci0 + ci < CI) {
+        int kh = k / KWC, r = k % KWC;
+        int kw = r / COp, co = r % COp;
+        if (co < CO)
+          v = w[((co * CI) + ci0 + ci) * KHKW + kh * KW + kw];
+      }
+      lds_w[ci][kk] = to_bf16(v);
+    }
+    if (tid < BK / 8) {
+      int k = k0 + tid * 8;
+      int kcl = k < Kp ? k : 0;
+      int kh = kcl / KWC, r = kcl % KWC;
+      int kw = r / COp, co0g = r % COp;
+      lds_kdec[tid] = (kh << 20) | (kw << 10) | co0g;
+    }
+    __syncthreads();
+
+    int kgroups = min(BK, Kp - k0) / 8;
+    for (int kg = 0; kg < kgroups; kg += 4) {
+      bool gin = lk < (kgroups - kg);
+      int gsel = gin ? kg + lk : 0;
+      bf16x8 b = gin ? *reinterpret_cast<const bf16x8*>(&lds_w[lm][gsel * 8])
+                     : bf8_zero();
+      int kdec = lds_kdec[gsel];
+      int kh = kdec >> 20, kw = (kdec >> 10) & 0x3ff, co0g = kdec & 0x3ff;
+#pragma unroll
+      for (int mi = 0; mi < MW; ++mi) {
+        bf16x8 a = bf8_zero();
+        if (mval[mi] & gin) {
+          int tH = ih[mi] + pad - kh;
+          int tW = iw[mi] + pad - kw;
+          if (S1) {
+            if (tH >= 0 && tH < OH && tW >= 0 && tW < OW)
+              a = *reinterpret_cast<const bf16x8*>(
+                  &dy[((long)(nn[mi] * OH + tH) * OW + tW) * COp + co0g]);
+          } else {
+            if (tH >= 0 && tW >= 0 && tH % stride == 0 && tW % stride == 0) {
+              int oh = tH / stride, ow = tW / stride;
+              if (oh < OH && ow < OW)
+                a = *reinterpret_cast<const bf16x8*>(
+                    &dy[((long)(nn[mi] * OH + oh) * OW + ow) * COp + co0g]);
+            }
+          }
+        }
+        acc[mi] = mfma16x16x32bf(a, b, acc[mi]);
+      }
+    }
+  }
+
+#pragma unroll
+  for (int mi = 0; mi < MW; ++mi) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      int m = m0 + (wid * MW + mi) * 16 + lk * 4 + reg;
+      if (m >= Mp) continue;
+      int ci = ci0 + lm;
+      if (ci >= CIp) continue;
+      dx[(long)m * CIp + ci] = to_bf16(ci < CI ? acc[mi][reg] : 0.f);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Conv2d backward-weight (wgrad) + bias, NHWC bf16 -> fp32 canonical dw/db.
+// Reduction over m: both MFMA fragments need 8-contiguous-m elements, so dy
+// and im2col(x) chunks are staged TRANSPOSED into LDS ([k][m] / [co][m])
+// with 16B-vector global reads (contiguous over k / co). 4 waves stream
+// disjoint m-chunks of a super-chunk; wave partials fold in LDS; each block
+// writes ONE partial slab (deterministic reduce_splits_k folds splits).
+// ---------------------------------------------------------------------------
+template <bool BOUNDS, bool WANT_BIAS, int KT>
+__global__ __launch_bounds__(256) void conv2d_wgrad_bf16_k(
+    const bf16* __restrict__ dy, const bf16* __restrict__ x,
+    float* __restrict__ dw_ws, float* __restrict__ db_ws,
+    int CIp, int CI, int H, int W, int COp, int CO, int KH, int KW,
+    int OH, int OW, int stride, int pad, int M, int K, int Kcan,
+    int m_per_split) {
+  constexpr int MC = 64;     // m per wave per super-chunk
+  const int k0 = blockIdx.x * (KT * 16);
+  const int co0 = blockIdx.y * 16;
+  const int ms = blockIdx.z;
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6;
+  const int l = tid & 63;
+  const int lm = l & 15;
+  const int lk = l >> 4;
+
+  // transposed chunk images for this block's 4 m-chunks (one per wave)
+  __shared__ bf16 lds_xT[4][KT * 16][MC + 8];   // [wave][k][m]
+  __shared__ bf16 lds_dyT[4][16][MC + 8];       // [wave][co][m]
+  __shared__ float lds_red[4 * KT * 256];       // wave-fold scratch
+
+  const int KWC = KW * CIp;
+  const int OHOW = OH * OW;
+
+  f32x4 acc[KT] = {};
+  float db_part = 0.f;
+
+  int m_begin = ms * m_per_split;
+  int m_end = min(M, m_begin + m_per_split);
+
+  for (int mc = m_begin; mc < m_end; mc += 4 * MC) {
+    __syncthreads();
+    // ---- stage x_col^T: entries (wave w, k row, m col) -------------------
+    // reads: bf16x8 over k (contiguous within a kh row); writes: 8 scalars.
+    // thread mapping: idx = (w*MC + mloc)*(KT*2) + kg  covers all k-groups.
+    for (int idx = tid; idx < 4 * MC * (KT * 2); idx += 256) {
+      int kg = idx % (KT * 2);
+      int rest = idx / (KT * 2);
+      int mloc = rest % MC;
+      int w = rest / MC;
+      int m = mc + w * MC + mloc;
+      int k = k0 + kg * 8;
+      bf16x8 v = bf8_zero();
+      if (m < m_end && k < K) {
+        int n = m / OHOW, r = m % OHOW;
+        int oh = r / OW, ow = r % OW;
+        int ih0 = oh * stride - pad, iw0 = ow * stride - pad;
+        int kh = k / KWC, kr = k % KWC;
+        int kw = kr / CIp, ci0 = kr % CIp;
+        bool ok = true;
+        if (BOUNDS) {
+          int ih = ih0 + kh, iw = iw0 + kw;
+          ok = ih >= 0 && ih < H && iw >= 0 && iw < W;
+        }
+        if (ok)
+          v = *reinterpret_cast<const bf16x8*>(
+              &x[((long)(n * H + ih0 + kh) * W + iw0 + kw) * CIp + ci0]);
+      }
+#pragma unroll
+      for (int j = 0; j < 8; ++j) lds_xT[w][kg * 8 + j][mloc] = v[j];
+    }
+    // ---- stage dy^T ------------------------------------------------------
+    for (int idx = tid; idx < 4 * MC * 2; idx += 256) {
+      int cg = idx & 1;          // two 8-co groups cover the 16-co tile
+      int rest = idx >> 1;
+      int mloc = rest % MC;
+      int w = rest / MC;
+      int m = mc + w * MC + mloc;
+      bf16x8 v = bf8_zero();
+      if (m < m_end && co0 + cg * 8 < COp)
+        v = *reinterpret_cast<const bf16x8*>(&dy[(long)m * COp + co0 + cg * 8]);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) lds_dyT[w][cg * 8 + j][mloc] = v[j];
+    }
+    __syncthreads();
+
+    // ---- MFMA over this wave's chunk ------------------------------------
+    int mlim = min(MC, m_end - mc - wid * MC);
+    for (int mm = 0; mm < mlim; mm += 32) {
+      // tail: zero-fill handled because staging zero-filled beyond m_end
+      bf16x8 b = *reinterpret_cast<const bf16x8*>(&lds_dyT[wid][lm][mm + lk * 8]);
+      if (WANT_BIAS) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) db_part += to_f32(b[j]);
+      }
+#pragma unroll
+      for (int kf = 0; kf < KT; ++kf) {
+        bf16x8 a = *reinterpret_cast<const bf16x8*>(
+            &lds_xT[wid][kf * 16 + lm][mm + lk * 8]);
+        acc[kf] = mfma16x16x32bf(a, b, acc[kf]);
+      }
+    }
+  }
+
+  // ---- fold waves in LDS, write partial slab (fp32 canonical layout) -----
+  __syncthreads();
+#pragma unroll
+  for (int kf = 0; kf < KT; ++kf)
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg)
+      lds_red[((wid * KT + kf) * 256) + (lk * 4 + reg) * 16 + lm] = acc[kf][reg];
+  __syncthreads();
+  if (wid == 0) {
+    float* slab = dw_ws + (long)ms * (CO * (long)Kcan);
+#pragma unroll
+    for (int kf = 0; kf < KT; ++kf) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        int e = (lk * 4 + reg) * 16 + lm;
+        float v = lds_red[kf * 256 + e] + lds_red[(KT + kf) * 256 + e] +
+                  lds_red[(2 * KT + kf) * 256 + e] + lds_red[(3 * KT + kf) * 256 + e];
+        int kk = k0 + kf * 16 + lk * 4 + reg;   // padded k space
+        int co = co0 + lm;
+        if (kk < K && co < CO) {
+          int kh = kk / KWC, r = kk % KWC;
+          int kw = r / CIp, ci = r % CIp;
+          if (ci < CI)
+            slab[((co * CI) + ci) * (KH * KW) + kh * KW + kw] = v;
+        }
+      }
+    }
+  }
+  if (WANT_BIAS && blockIdx.x == 0) {
+    // NB: D row/col note — b's lm indexes co here (B-operand col space);
+    // db_part sums dy over this lane's m-slices for co = co0+lm.
+    db_part += __shfl_down(db_part, 16);
+    db_part += __shfl_down(db_part, 32);
+    __syncthreads();
+    if (lk == 0) lds_red[wid * 16 + lm] = db_part;
+    __syncthreads();
+    if (tid < 16 && co0 + tid < CO)
+      db_ws[(long)ms * CO + co0 + tid] = lds_red[tid] + lds_red[16 + tid] +
+                                         lds_red[32 + tid] + lds_red[48 + tid];
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Linear forward, bf16: y[M][COp] = x[M][CIk] . w^T + b (optional ReLU).
+// w is fp32 canonical [CO][CIcan]; perm[k] maps the (padded, possibly
+// NHWC-flattened) k index to the canonical column (-1 => zero pad), so the
+// checkpoint-ordered weights never need a materialized repack.
+// ---------------------------------------------------------------------------
+template <bool HAS_BIAS, bool RELU, int MW>
+__global__ __launch_bounds__(256) void linear_fwd_bf16_k(
+    const bf16* __restrict__ x, const float* __restrict__ w,
+    const float* __restrict__ bias, const int* __restrict__ perm,
+    bf16* __restrict__ y, int M, int CIk, int CIcan, int COp, int CO) {
+  constexpr int BK = 32;
+  const int BM = 4 * MW * 16;
+  const int m0 = blockIdx.x * BM;
+  const int co0 = blockIdx.y * 16;
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6;
+  const int l = tid & 63;
+  const int lm = l & 15;
+  const int lk = l >> 4;
+
+  __shared__ bf16 lds_w[16][BK + 8];
+
+  f32x4 acc[MW] = {};
+
+  for (int k0 = 0; k0 < CIk; k0 += BK) {
+    __syncthreads();
+    for (int idx = tid; idx < 16 * BK; idx += 256) {
+      int co = idx >> 5;
+      int kk = idx & 31;
+      int k = k0 + kk;
+      float v = 0.f;
+      if (k < CIk && co0 + co < CO) {
+        int ci = perm[k];
+        if (ci >= 0) v = w[(co0 + co) * CIcan + ci];
+      }
+      lds_w[co][kk] = to_bf16(v);
+    }
+    __syncthreads();
+
+    int kgroups = min(BK, CIk - k0) / 8;
+    for (int kg = 0; kg < kgroups; kg += 4) {
+      bool gin = lk < (kgroups - kg);
+      int gsel = gin ? kg + lk : 0;
+      bf16x8 b = gin ? *reinterpret_cast<const bf16x8*>(&lds_w[lm][gsel * 8])
+                     : bf8_zero();
+#pragma unroll
+      for (int mi = 0; mi < MW; ++mi) {
+        int m = m0 + (wid * MW + mi) * 16 + lm;
+        bf16x8 a = (m < M && gin)
+                       ? *reinterpret_cast<const bf16x8*>(
+                             &x[(long)m * CIk + k0 + gsel * 8])
+                       : bf8_zero();
+        acc[mi] = mfma16x16x32bf(a, b, acc[mi]);
+      }
+    }
+  }
+
+#pragma unroll
+  for (int mi = 0; mi < MW; ++mi) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      int m = m0 + (wid * MW + mi) * 16 + lk * 4 + reg;
+      if (m >= M) continue;
+      int co = co0 + lm;
+      if (co >= COp) continue;
+      float v = (co < CO) ? acc[mi][reg] : 0.f;
+      if (HAS_BIAS && co < CO) v += bias[co];
+      if (RELU) v = fmaxf(v, 0.f);
+      y[(long)m * COp + co] = to_bf16(v);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Linear backward-input, bf16: dx[M][CIk] = dyr[M][COp] . w (relu-masked).
+//   A = dyr rows (8 contiguous co; mask from saved y at identical offsets),
+//   B = w^T staged via perm into LDS as [ci_k][co] rows.
+// ---------------------------------------------------------------------------
+template <bool RELU, int MW>
+__global__ __launch_bounds__(256) void linear_dx_bf16_k(
+    const bf16* __restrict__ dy, const float* __restrict__ w,
+    const bf16* __restrict__ yact, const int* __restrict__ perm,
+    bf16* __restrict__ dx, int M, int CIk, int CIcan, int COp, int CO) {
+  constexpr int BK = 32;
+  const int BM = 4 * MW * 16;
+  const int m0 = blockIdx.x * BM;
+  const int ci0 = blockIdx.y * 16;
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6;
+  const int l = tid & 63;
+  const int lm = l & 15;
+  const int lk = l >> 4;
+
+  __shared__ bf16 lds_w[16][BK + 8];   // [ci_k][co-chunk]
+
+  f32x4 acc[MW] = {};
+
+  for (int k0 = 0; k0 < COp; k0 += BK) {
+    __syncthreads();
+    for (int idx = tid; idx < 16 * BK; idx += 256) {
+      int ci = idx >> 5;
+      int kk = idx & 31;
+      int co = k0 + kk;
+      float v = 0.f;
+      if (co < CO && ci0 + ci < CIk) {
+        int cic = perm[ci0 + ci];
+        if (cic >= 0) v = w[co * CIcan + cic];
+      }
+      lds_w[ci][kk] = to_bf16(v);
+    }
+    __syncthreads();
+
+    int kgroups = min(BK, COp - k0) / 8;
+    for (int kg = 0; kg < kgroups; kg += 4) {
+      bool gin = lk < (kgroups - kg);
+      int gsel = gin ? kg + lk : 0;
+      bf16x8 b = gin ? *reinterpret_cast<const bf16x8*>(&lds_w[lm][gsel * 8])
+                     : bf8_zero();
+#pragma unroll
+      for (int mi = 0; mi < MW; ++mi) {
+        int m = m0 + (wid * MW + mi) * 16 + lm;
+        bf16x8 a = bf8_zero();
+        if (m < M && gin) {
+          long off = (long)m * COp + k0 + gsel * 8;
+          a = *reinterpret_cast<const bf16x8*>(&dy[off]);
+          if (RELU) {
+            bf16x8 yv = *reinterpret_cast<const bf16x8*>(&yact[off]);
+#pragma unroll
+            for (int j = 0; j < 8; ++j)
+              if (!(to_f32(yv[j]) > 0.f)) a[j] = to_bf16(0.f);
+          }
+        }
+        acc[mi] = mfma16x16x32bf(a, b, acc[mi]);
+      }
+    }
+  }
+
+#pragma unroll
+  for (int mi = 0; mi < MW; ++mi) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      int m = m0 + (wid * MW + mi) * 16 + lk * 4 + reg;
+      if (m >= M) continue;
+      int ci = ci0 + lm;
+      if (ci >= CIk) continue;
+      dx[(long)m * CIk + ci] = to_bf16(acc[mi][reg]);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Linear backward-weight (+bias), bf16 -> fp32 canonical via perm.
+// Same structure as conv wgrad: transposed LDS chunks, waves on m, partial
+// slabs. dyr mask applied at staging time.
+// ---------------------------------------------------------------------------
+template <bool RELU, bool WANT_BIAS, int KT>
+__global__ __launch_bounds__(256) void linear_dw_bf16_k(
+    const bf16* __restrict__ dy, const bf16* __restrict__ x,
+    const bf16* __restrict__ yact, const int* __restrict__ perm,
+    float* __restrict__ dw_ws, float* __restrict__ db_ws,
+    int M, int CIk, int CIcan, int COp, int CO, int m_per_split) {
+  constexpr int MC = 64;
+  const int k0 = blockIdx.x * (KT * 16);   // ci_k tile
+  const int co0 = blockIdx.y * 16;
+  const int ms = blockIdx.z;
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6;
+  const int l = tid & 63;
+  const int lm = l & 15;
+  const int lk = l >> 4;
+
+  __shared__ bf16 lds_xT[4][KT * 16][MC + 8];
+  __shared__ bf16 lds_dyT[4][16][MC + 8];
+  __shared__ float lds_red[4 * KT * 256];
+
+  f32x4 acc[KT] = {};
+  float db_part = 0.f;
+
+  int m_begin = ms * m_per_split;
+  int m_end = min(M, m_begin + m_per_split);
+
+  for (int mc = m_begin; mc < m_end; mc += 4 * MC) {
+    __syncthreads();
+    for (int idx = tid; idx < 4 * MC * (KT * 2); idx += 256) {
+      int kg = idx % (KT * 2);
+      int rest = idx / (KT * 2);
+      int mloc = rest % MC;
+      int w = rest / MC;
+      int m = mc + w * MC + mloc;
+      int k = k0 + kg * 8;
+      bf16x8 v = bf8_zero();
+      if (m < m_end && k < CIk)
+        v = *reinterpret_cast<const bf16x8*>(&x[(long)m * CIk + k]);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) lds_xT[w][kg * 8 + j][mloc] = v[j];
+    }
+    for (int idx = tid; idx < 4 * MC * 2; idx += 256) {
+      int cg = idx & 1;
+      int rest = idx >> 1;
+      int mloc = rest % MC;
+      int w = rest / MC;
+      int m = mc + w * MC + mloc;
+      bf16x8 v = bf8_zero();
+      if (m < m_end && co0 + cg * 8 < COp) {
+        long off = (long)m * COp + co0 + cg * 8;
+        v = *reinterpret_cast<const bf16x8*>(&dy[off]);
+        if (RELU) {
+          bf16x8 yv = *reinterpret_cast<const bf16x8*>(&yact[off]);
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            if (!(to_f32(yv[j]) > 0.f)) v[j] = to_bf16(0.f);
+        }
+      }
+#pragma unroll
+      for (int j = 0; j < 8; ++j) lds_dyT[w][cg * 8 + j][mloc] = v[j];
+    }
+    __syncthreads();
+
+    int mlim = min(MC, m_end - mc - wid * MC);
+    for (int mm = 0; mm < mlim; mm += 32) {
+      bf16x8 b = *reinterpret_cast<const bf16x8*>(&lds_dyT[wid][lm][mm + lk * 8]);
+      if (WANT_BIAS) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) db_part += to_f32(b[j]);
+      }
+#pragma unroll
+      for (int kf = 0; kf < KT; ++kf) {
+        bf16x8 a = *reinterpret_cast<const bf16x8*>(
+            &lds_xT[wid][kf * 16 + lm][mm + lk * 8]);
+        acc[kf] = mfma16x16x32bf(a, b, acc[kf]);
+      }
+    }
+  }
+
+  __syncthreads();
+#pragma unroll
+  for (int kf = 0; kf < KT; ++kf)
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg)
+      lds_red[((wid * KT + kf) * 256) + (lk * 4 + reg) * 16 + lm] = acc[kf][reg];
+  __syncthreads();
+  if (wid == 0) {
+    float* slab = dw_ws + (long)ms * (CO * (long)CIcan);
+#pragma unroll
+    for (int kf = 0; kf < KT; ++kf) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        int e = (lk * 4 + reg) * 16 + lm;
+        float v = lds_red[kf * 256 + e] + lds_red[(KT + kf) * 256 + e] +
+                  lds_red[(2 * KT + kf) * 256 + e] + lds_red[(3 * KT + kf) * 256 + e];
+        int kk = k0 + kf * 16 + lk * 4 + reg;
+        int co = co0 + lm;
+        if (kk < CIk && co < CO) {
+          int cic = perm[kk];
+          if (cic >= 0) slab[co * CIcan + cic] = v;
+        }
+      }
+    }
+  }
+  if (WANT_BIAS && blockIdx.x == 0) {
+    db_part += __shfl_down(db_part, 16);
+    db_part += __shfl_down(db_part, 32);
+    __syncthreads();
+    if (lk == 0) lds_red[wid * 16 + lm] = db_part;
+    __syncthreads();
+    if (tid < 16 && co0 + tid < CO)
+      db_ws[(long)ms * CO + co0 + tid] = lds_red[tid] + lds_red[16 + tid] +
+                                         lds_red[32 + tid] + lds_red[48 + tid];
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Fused MaxPool2d(2,2) + ReLU, NHWC bf16. One thread per (n,oh,ow,8-channel
+// group): 4 x 16B loads, packed max, relu; u8 argmax per channel (stored as
+// one 8-byte group). Backward scatters into the exclusively-owned 2x2 window.
+// ---------------------------------------------------------------------------
+__global__ void maxpool2x2_relu_bf16_fwd_k(const bf16* __restrict__ x,
+                                           bf16* __restrict__ y,
+                                           unsigned char* __restrict__ idx,
+                                           int N, int H, int W, int Cp,
+                                           int OH, int OW) {
+  long total = (long)N * OH * OW * (Cp / 8);
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int cg = i % (Cp / 8);
+    long r = i / (Cp / 8);
+    int ow = r % OW;
+    long r2 = r / OW;
+    int oh = r2 % OH;
+    int n = r2 / OH;
+    const bf16* p = x + (((long)(n * H + oh * 2) * W + ow * 2) * Cp) + cg * 8;
+    bf16x8 v0 = *reinterpret_cast<const bf16x8*>(p);
+    bf16x8 v1 = *reinterpret_cast<const bf16x8*>(p + Cp);
+    bf16x8 v2 = *reinterpret_cast<const bf16x8*>(p + (long)W * Cp);
+    bf16x8 v3 = *reinterpret_cast<const bf16x8*>(p + (long)W * Cp + Cp);
+    bf16x8 out;
+    unsigned char am[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f0 = to_f32(v0[j]), f1 = to_f32(v1[j]);
+      float f2 = to_f32(v2[j]), f3 = to_f32(v3[j]);
+      float m01 = fmaxf(f0, f1), m23 = fmaxf(f2, f3);
+      float m = fmaxf(m01, m23);
+      unsigned char a;
+      if (m == f0) a = 0; else if (m == f1) a = 1;
+      else if (m == f2) a = 2; else a = 3;
+      out[j] = to_bf16(fmaxf(m, 0.f));
+      am[j] = a;
+    }
+    long yoff = (((long)(n * OH + oh) * OW + ow) * Cp) + cg * 8;
+    *reinterpret_cast<bf16x8*>(y + yoff) = out;
+    *reinterpret_cast<unsigned long long*>(idx + yoff) =
+        *reinterpret_cast<unsigned long long*>(am);
+  }
+}
+
+__global__ void maxpool2x2_relu_bf16_bwd_k(const bf16* __restrict__ dy,
+                                           const bf16* __restrict__ y,
+                                           const unsigned char* __restrict__ idx,
+                                           bf16* __restrict__ dx,
+                                           int N, int H, int W, int Cp,
+                                           int OH, int OW) {
+  long total = (long)N * OH * OW * (Cp / 8);
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int cg = i % (Cp / 8);
+    long r = i / (Cp / 8);
+    int ow = r % OW;
+    long r2 = r / OW;
+    int oh = r2 % OH;
+    int n = r2 / OH;
+    long yoff = (((long)(n * OH + oh) * OW + ow) * Cp) + cg * 8;
+    bf16x8 g = *reinterpret_cast<const bf16x8*>(dy + yoff);
+    bf16x8 yv = *reinterpret_cast<const bf16x8*>(y + yoff);
+    unsigned long long am8 =
+        *reinterpret_cast<const unsigned long long*>(idx + yoff);
+    bf16x8 o0 = bf8_zero(), o1 = bf8_zero(), o2 = bf8_zero(), o3 = bf8_zero();
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float gv = (to_f32(yv[j]) > 0.f) ? to_f32(g[j]) : 0.f;
+      unsigned char a = (am8 >> (8 * j)) & 0xff;
+      bf16 bg = to_bf16(gv);
+      if (a == 0) o0[j] = bg; else if (a == 1) o1[j] = bg;
+      else if (a == 2) o2[j] = bg; else o3[j] = bg;
+    }
+    bf16* p = dx + (((long)(n * H + oh * 2) * W + ow * 2) * Cp) + cg * 8;
+    *reinterpret_cast<bf16x8*>(p) = o0;
+    *reinterpret_cast<bf16x8*>(p + Cp) = o1;
+    *reinterpret_cast<bf16x8*>(p + (long)W * Cp) = o2;
+    *reinterpret_cast<bf16x8*>(p + (long)W * Cp + Cp) = o3;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Dropout bf16, vectorized 8-wide; one mix64 hash yields 8 keep-bytes.
+// channelwise (Dropout2d on NHWC): rng unit = (n, c) — computed per element
+// of the vector (the 8 lanes are 8 consecutive channels of one pixel).
+// mask: u8 per element (elementwise) or per (n,c) (channelwise).
+// ---------------------------------------------------------------------------
+__global__ void dropout_bf16_fwd_k(const bf16* __restrict__ x,
+                                   bf16* __restrict__ y,
+                                   unsigned char* __restrict__ mask,
+                                   long n8, int Cp, long hwc, float p,
+                                   int channelwise,
+                                   unsigned long long seed,
+                                   unsigned long long ctr) {
+  float inv = 1.f / (1.f - p);
+  unsigned int pthresh = (unsigned int)(p * 256.f);
+  for (long i8 = (long)blockIdx.x * blockDim.x + threadIdx.x; i8 < n8;
+       i8 += (long)gridDim.x * blockDim.x) {
+    long i = i8 * 8;
+    bf16x8 v = *reinterpret_cast<const bf16x8*>(x + i);
+    unsigned long long h;
+    if (channelwise) {
+      // unit = (n, cgroup): one hash covers the 8 consecutive channels
+      long nimg = i / hwc;
+      int c8 = (int)((i % Cp) / 8);
+      h = mix64(seed ^ mix64(ctr ^ mix64((unsigned long long)(nimg * (Cp / 8) + c8))));
+    } else {
+      h = mix64(seed ^ mix64(ctr ^ mix64((unsigned long long)i8)));
+    }
+    bf16x8 o;
+    unsigned char mk[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      bool keep = ((h >> (8 * j)) & 0xff) >= pthresh;
+      mk[j] = keep ? 1 : 0;
+      o[j] = keep ? to_bf16(to_f32(v[j]) * inv) : to_bf16(0.f);
+    }
+    *reinterpret_cast<bf16x8*>(y + i) = o;
+    if (channelwise) {
+      // mask indexed [n][Cp]: only the first pixel of each image writes it
+      long nimg = i / hwc;
+      if (i % hwc < Cp) {
+        int c0 = (int)(i % Cp);
+        for (int j = 0; j < 8; ++j) mask[nimg * Cp + c0 + j] = mk[j];
+      }
+    } else {
+      *reinterpret_cast<unsigned long long*>(mask + i) =
+          *reinterpret_cast<unsigned long long*>(mk);
+    }
+  }
+}
+
+__global__ void dropout_bf16_bwd_k(const bf16* __restrict__ dy,
+                                   const unsigned char* __restrict__ mask,
+                                   bf16* __restrict__ dx, long n8, int Cp,
+                                   long hwc, float p, int channelwise) {
+  float inv = 1.f / (1.f - p);
+  for (long i8 = (long)blockIdx.x * blockDim.x + threadIdx.x; i8 < n8;
+       i8 += (long)gridDim.x * blockDim.x) {
+    long i = i8 * 8;
+    bf16x8 v = *reinterpret_cast<const bf16x8*>(dy + i);
+    unsigned char mk[8];
+    if (channelwise) {
+      long nimg = i / hwc;
+      int c0 = (int)(i % Cp);
+      for (int j = 0; j < 8; ++j) mk[j] = mask[nimg * Cp + c0 + j];
+    } else {
+      *reinterpret_cast<unsigned long long*>(mk) =
+          *reinterpret_cast<const unsigned long long*>(mask + i);
+    }
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      o[j] = mk[j] ? to_bf16(to_f32(v[j]) * inv) : to_bf16(0.f);
+    *reinterpret_cast<bf16x8*>(dx + i) = o;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// LogSoftmax + mean NLL, bf16 logits [M][COp] (only c < C are real columns).
+// logp fp32 [M][C] (the user-facing reference interface); bwd emits bf16
+// dx [M][COp] with zeroed padding columns.
+// ---------------------------------------------------------------------------
+__global__ void logsoftmax_nll_bf16_fwd_k(const bf16* __restrict__ x,
+                                          const long* __restrict__ target,
+                                          float* __restrict__ logp,
+                                          float* __restrict__ loss,
+                                          int M, int COp, int C, int with_loss) {
+  __shared__ float red[256];
+  float part = 0.f;
+  for (long m = (long)blockIdx.x * blockDim.x + threadIdx.x; m < M;
+       m += (long)gridDim.x * blockDim.x) {
+    const bf16* row = x + m * COp;
+    float mx = to_f32(row[0]);
+    for (int c = 1; c < C; ++c) mx = fmaxf(mx, to_f32(row[c]));
+    float s = 0.f;
+    for (int c = 0; c < C; ++c) s += __expf(to_f32(row[c]) - mx);
+    float lse = mx + __logf(s);
+    float* out = logp + m * C;
+    for (int c = 0; c < C; ++c) out[c] = to_f32(row[c]) - lse;
+    if (with_loss) part -= out[target[m]];
+  }
+  if (with_loss) {
+    red[threadIdx.x] = part;
+    __syncthreads();
+    for (int s = 128; s > 0; s >>= 1) {
+      if (threadIdx.x < s) red[threadIdx.x] += red[threadIdx.x + s];
+      __syncthreads();
+    }
+    if (threadIdx.x == 0) atomicAdd(loss, red[0] / (float)M);
+  }
+}
+
+__global__ void logsoftmax_nll_bf16_bwd_k(const float* __restrict__ logp,
+                                          const long* __restrict__ target,
+                                          const float* __restrict__ dloss,
+                                          bf16* __restrict__ dx, int M,
+                                          int COp, int C) {
+  float scale = dloss[0] / (float)M;
+  for (long m = (long)blockIdx.x * blockDim.x + threadIdx.x; m < M;
+       m += (long)gridDim.x * blockDim.x) {
+    const float* lp = logp + m * C;
+    bf16* out = dx + m * COp;
+    long t = target[m];
+    for (int c = 0; c < COp; ++c) {
+      float v = 0.f;
+      if (c < C) v = (__expf(lp[c]) - (c == t ? 1.f : 0.f)) * scale;
+      out[c] = to_bf16(v);
+    }
+  }
+}

@@ -7,6 +7,7 @@
 #include "conv.hip"
 #include "linear.hip"
 #include "elementwise.hip"
+#include "bf16.hip"
 
 namespace {
 
@@ -356,6 +357,8 @@ void fused_sgd_step(at::Tensor ptrs, at::Tensor chunks, long chunk_elems,
                      (int)chunk_elems, (float)lr, (float)momentum);
 }
 
+#include "ext_bf16.inc"
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -375,4 +378,18 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dropout_fwd", &dropout_fwd);
   m.def("dropout_bwd", &dropout_bwd);
   m.def("fused_sgd_step", &fused_sgd_step);
+  // bf16 NHWC engine
+  m.def("mfma_probe", &mfma_probe);
+  m.def("conv2d_fwd_bf16", &conv2d_fwd_bf16);
+  m.def("conv2d_dgrad_bf16", &conv2d_dgrad_bf16);
+  m.def("conv2d_wgrad_bf16", &conv2d_wgrad_bf16);
+  m.def("linear_fwd_bf16", &linear_fwd_bf16);
+  m.def("linear_dx_bf16", &linear_dx_bf16);
+  m.def("linear_dw_bf16", &linear_dw_bf16);
+  m.def("maxpool2x2_relu_bf16_fwd", &maxpool2x2_relu_bf16_fwd);
+  m.def("maxpool2x2_relu_bf16_bwd", &maxpool2x2_relu_bf16_bwd);
+  m.def("dropout_bf16_fwd", &dropout_bf16_fwd);
+  m.def("dropout_bf16_bwd", &dropout_bf16_bwd);
+  m.def("logsoftmax_nll_bf16_fwd", &logsoftmax_nll_bf16_fwd);
+  m.def("logsoftmax_nll_bf16_bwd", &logsoftmax_nll_bf16_bwd);
 }

@@ -1,0 +1,241 @@
+"""bf16 NHWC engine numerics vs plain PyTorch fp32 references (GPU only).
+
+Tolerances reflect bf16 storage of activations (rel ~1e-2) while weight-grad
+reductions accumulate fp32.
+"""
+
+import pytest
+import torch
+import torch.nn.functional as F
+
+from dist_mnist_amd.ops import functional_bf16 as fb
+from dist_mnist_amd.ops import native
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda:0"
+
+
+def randn(*s, seed=0, scale=1.0):
+    g = torch.Generator().manual_seed(seed)
+    return (torch.randn(*s, generator=g) * scale).to(DEV)
+
+
+def test_mfma_probe_mapping():
+    """Asymmetric A/B catch any transpose/mapping error (guide §3)."""
+    a = randn(16, 32, seed=1)
+    b = randn(32, 16, seed=2)
+    d = native.ext().mfma_probe(a, b)
+    ref = (a.to(torch.bfloat16).float() @ b.to(torch.bfloat16).float())
+    torch.testing.assert_close(d, ref, rtol=1e-3, atol=1e-3)
+
+
+@pytest.mark.parametrize(
+    "N,CI,H,W,CO,KH,KW,stride,pad",
+    [
+        (4, 1, 28, 28, 10, 5, 5, 1, 0),   # conv1
+        (4, 10, 12, 12, 20, 5, 5, 1, 0),  # conv2
+        (2, 3, 18, 18, 7, 3, 3, 1, 1),    # pad
+        (2, 8, 16, 16, 16, 3, 3, 2, 1),   # stride 2
+        (2, 3, 32, 32, 8, 7, 7, 2, 3),    # 7x7 s2 stem
+        (131, 1, 28, 28, 10, 5, 5, 1, 0), # tile remainder
+    ],
+)
+def test_conv2d_bf16_fwd(N, CI, H, W, CO, KH, KW, stride, pad):
+    x = randn(N, CI, H, W, seed=3)
+    w = randn(CO, CI, KH, KW, seed=4, scale=0.3)
+    b = randn(CO, seed=5)
+    xb = fb.nchw_to_nhwc_padded(x)
+    y = fb.conv2d_bf16(xb, w, b, stride=stride, pad=pad)
+    cop = fb.pad8(CO)
+    assert y.shape == (N, (H + 2 * pad - KH) // stride + 1,
+                       (W + 2 * pad - KW) // stride + 1, cop)
+    ref = F.conv2d(x, w, b, stride=stride, padding=pad)
+    got = fb.nhwc_to_nchw(y, CO)
+    torch.testing.assert_close(got, ref, rtol=2e-2, atol=3e-2)
+    # padded channels must be exactly zero
+    assert (y[..., CO:].float() == 0).all()
+
+
+@pytest.mark.parametrize(
+    "N,CI,H,W,CO,KH,KW,stride,pad",
+    [
+        (4, 10, 12, 12, 20, 5, 5, 1, 0),
+        (2, 3, 18, 18, 7, 3, 3, 1, 1),
+        (2, 8, 16, 16, 16, 3, 3, 2, 1),
+    ],
+)
+def test_conv2d_bf16_bwd(N, CI, H, W, CO, KH, KW, stride, pad):
+    OH = (H + 2 * pad - KH) // stride + 1
+    OW = (W + 2 * pad - KW) // stride + 1
+    x = randn(N, CI, H, W, seed=6)
+    w = randn(CO, CI, KH, KW, seed=7, scale=0.3).requires_grad_(True)
+    b = randn(CO, seed=8).requires_grad_(True)
+    dy = randn(N, CO, OH, OW, seed=9)
+
+    xb = fb.nchw_to_nhwc_padded(x).requires_grad_(True)
+    y = fb.conv2d_bf16(xb, w, b, stride=stride, pad=pad)
+    dyb = fb.nchw_to_nhwc_padded(dy, cop=y.shape[3])
+    y.backward(dyb)
+
+    xr = x.detach().clone().requires_grad_(True)
+    wr = w.detach().clone().requires_grad_(True)
+    br = b.detach().clone().requires_grad_(True)
+    F.conv2d(xr, wr, br, stride=stride, padding=pad).backward(dy)
+
+    got_dx = fb.nhwc_to_nchw(xb.grad, CI)
+    torch.testing.assert_close(got_dx, xr.grad, rtol=3e-2, atol=4e-2)
+    # weight grads are fp32-accumulated over bf16 products
+    torch.testing.assert_close(w.grad, wr.grad, rtol=2e-2,
+                               atol=2e-2 * max(1.0, wr.grad.abs().max().item()))
+    torch.testing.assert_close(b.grad, br.grad, rtol=2e-2,
+                               atol=2e-2 * max(1.0, br.grad.abs().max().item()))
+
+
+def test_maxpool_relu_bf16():
+    x = randn(3, 10, 12, 16, seed=10)
+    xb = fb.nchw_to_nhwc_padded(x).requires_grad_(True)
+    y = fb.maxpool2x2_relu_bf16(xb)
+    dy = randn(3, 10, 6, 8, seed=11)
+    dyb = fb.nchw_to_nhwc_padded(dy, cop=16)
+    y.backward(dyb)
+
+    xr = x.detach().to(torch.bfloat16).float().requires_grad_(True)
+    out = F.relu(F.max_pool2d(xr, 2))
+    out.backward(dy.to(torch.bfloat16).float())
+    torch.testing.assert_close(fb.nhwc_to_nchw(y, 10), out, rtol=1e-2, atol=1e-2)
+    torch.testing.assert_close(fb.nhwc_to_nchw(xb.grad, 10), xr.grad,
+                               rtol=1e-2, atol=1e-2)
+
+
+@pytest.mark.parametrize("M,CI,CO,relu", [(64, 320, 50, True), (64, 50, 10, False),
+                                          (257, 320, 50, True)])
+def test_linear_bf16(M, CI, CO, relu):
+    cik = fb.pad8(CI)
+    cop = fb.pad8(CO)
+    x = randn(M, CI, seed=12)
+    w = randn(CO, CI, seed=13, scale=0.1).requires_grad_(True)
+    b = randn(CO, seed=14).requires_grad_(True)
+    dy = randn(M, CO, seed=15)
+
+    xb = torch.zeros(M, cik, dtype=torch.bfloat16, device=DEV)
+    xb[:, :CI] = x.to(torch.bfloat16)
+    xb.requires_grad_(True)
+    y = fb.linear_bf16(xb, w, b, cop=cop, relu=relu)
+    dyb = torch.zeros(M, cop, dtype=torch.bfloat16, device=DEV)
+    dyb[:, :CO] = dy.to(torch.bfloat16)
+    y.backward(dyb)
+
+    xr = xb.detach()[:, :CI].float().requires_grad_(True)
+    wr = w.detach().clone().requires_grad_(True)
+    br = b.detach().clone().requires_grad_(True)
+    ref = F.linear(xr, wr, br)
+    if relu:
+        ref = F.relu(ref)
+    ref.backward(dyb.detach()[:, :CO].float())
+
+    torch.testing.assert_close(y[:, :CO].float(), ref, rtol=2e-2, atol=3e-2)
+    torch.testing.assert_close(xb.grad[:, :CI].float(), xr.grad, rtol=3e-2, atol=4e-2)
+    torch.testing.assert_close(w.grad, wr.grad, rtol=2e-2,
+                               atol=2e-2 * max(1.0, wr.grad.abs().max().item()))
+    torch.testing.assert_close(b.grad, br.grad, rtol=2e-2,
+                               atol=2e-2 * max(1.0, br.grad.abs().max().item()))
+
+
+def test_loss_bf16():
+    M = 130
+    x = randn(M, 10, seed=16)
+    xb = torch.zeros(M, 16, dtype=torch.bfloat16, device=DEV)
+    xb[:, :10] = x.to(torch.bfloat16)
+    xb.requires_grad_(True)
+    t = torch.randint(0, 10, (M,), device=DEV)
+    logp, loss = fb.logsoftmax_nll_bf16(xb, t, c=10)
+    loss.backward()
+
+    xr = xb.detach()[:, :10].float().requires_grad_(True)
+    lr = F.log_softmax(xr, dim=1)
+    F.nll_loss(lr, t).backward()
+    torch.testing.assert_close(logp, lr, rtol=1e-3, atol=1e-3)
+    torch.testing.assert_close(xb.grad[:, :10].float(), xr.grad, rtol=1e-2, atol=1e-3)
+    assert (xb.grad[:, 10:].float() == 0).all()
+
+
+def test_dropout_bf16():
+    x = torch.ones(100000 * 8, dtype=torch.bfloat16, device=DEV).requires_grad_(True)
+    y = fb.dropout_bf16(x, p=0.5, training=True)
+    keep = y.float() != 0
+    frac = keep.float().mean().item()
+    assert 0.47 < frac < 0.53, frac
+    y.backward(torch.ones_like(y))
+    assert (x.grad.float()[keep] == 2.0).all()
+    assert (x.grad.float()[~keep] == 0.0).all()
+
+
+def test_dropout2d_bf16_channelwise():
+    x = torch.ones(64, 8, 8, 24, dtype=torch.bfloat16, device=DEV)
+    y = fb.dropout_bf16(x, p=0.5, training=True, channelwise=True)
+    yc = y.float().permute(0, 3, 1, 2).reshape(64 * 24, 64)
+    zero = (yc == 0).all(dim=1)
+    live = (yc == 2.0).all(dim=1)
+    assert (zero | live).all()
+    assert 0.4 < live.float().mean().item() < 0.6
+
+
+def test_net_bf16_matches_fp32_reference():
+    from dist_mnist_amd.models import Net
+
+    torch.manual_seed(30)
+    net = Net().to(DEV).eval().bf16_native_()
+    x = randn(32, 1, 28, 28, seed=31)
+    t = torch.randint(0, 10, (32,), device=DEV)
+    xb = fb.nchw_to_nhwc_padded(x)
+    logp, loss = net(xb, t)
+
+    ref = Net().to(DEV).eval()
+    ref.load_state_dict(net.state_dict())
+    logp_ref, loss_ref = ref(x, t)
+    torch.testing.assert_close(loss, loss_ref, rtol=5e-2, atol=5e-2)
+    torch.testing.assert_close(logp, logp_ref, rtol=5e-2, atol=8e-2)
+
+
+def test_net_bf16_grads_match_fp32():
+    from dist_mnist_amd.models import Net
+
+    torch.manual_seed(32)
+    net = Net().to(DEV).eval().bf16_native_()
+    ref = Net().to(DEV).eval()
+    ref.load_state_dict(net.state_dict())
+    x = randn(64, 1, 28, 28, seed=33)
+    t = torch.randint(0, 10, (64,), device=DEV)
+
+    _, loss = net(fb.nchw_to_nhwc_padded(x), t)
+    loss.backward()
+    _, loss_ref = ref(x, t)
+    loss_ref.backward()
+
+    rsd = dict(ref.named_parameters())
+    for name, p in net.named_parameters():
+        r = rsd[name].grad
+        tol = max(0.05 * r.abs().max().item(), 2e-4)
+        torch.testing.assert_close(p.grad, r, rtol=5e-2, atol=tol,
+                                   msg=f"bf16 grad mismatch {name}")
+
+
+def test_training_loss_decreases_bf16():
+    from dist_mnist_amd.models import Net
+    from dist_mnist_amd.ops import SGD
+
+    torch.manual_seed(34)
+    net = Net().to(DEV).eval().bf16_native_()
+    opt = SGD(net.parameters(), lr=0.05, momentum=0.5)
+    x = fb.nchw_to_nhwc_padded(randn(64, 1, 28, 28, seed=35))
+    t = torch.randint(0, 10, (64,), device=DEV)
+    losses = []
+    for _ in range(150):
+        opt.zero_grad()
+        _, loss = net(x, t)
+        loss.backward()
+        opt.step()
+        losses.append(loss.item())
+    tail = sum(losses[-10:]) / 10
+    assert tail < losses[0] * 0.75, (losses[0], tail, losses[::30])
