@@ -292,12 +292,30 @@ __global__ __launch_bounds__(256) void conv2d_dgrad_bf16_k(
 
 // ---------------------------------------------------------------------------
 // Conv2d backward-weight (wgrad) + bias, NHWC bf16 -> fp32 canonical dw/db.
-// Reduction over m: both MFMA fragments need 8-contiguous-m elements, so dy
-// and im2col(x) chunks are staged TRANSPOSED into LDS ([k][m] / [co][m])
-// with 16B-vector global reads (contiguous over k / co). 4 waves stream
-// disjoint m-chunks of a super-chunk; wave partials fold in LDS; each block
-// writes ONE partial slab (deterministic reduce_splits_k folds splits).
+// Reduction over m. Waves split the k-tile (k-tile = 4*KT*16); the dy chunk
+// and the im2col(x) chunk are staged NON-transposed as [m][16] sub-images
+// (vectorized bf16x8 global reads -> ds_write_b128), and MFMA fragments come
+// out via gfx950's hardware transpose read ds_read_tr16_b64 (guide T10) —
+// no scalar LDS transpose pass. Each block writes one partial slab; the
+// deterministic reduce_splits_k folds the split axis.
+// Requires K >= ~128 for full wave utilization (LeNet: K=200/400; ResNet
+// 3x3: >=576). KT = per-wave 16-row k-frags (2 => k-tile 128).
 // ---------------------------------------------------------------------------
+using lds_bf4_ptr = __attribute__((address_space(3))) bf16x4*;
+
+DEV bf16x8 tr_read_8m(const bf16* lds_img, int lk) {
+  // image: [m][16] bf16 row-major; returns this lane's 8 m-elements at
+  // column (l&15), rows lk*8 .. lk*8+7 (two 4-row transpose reads).
+  bf16x4 r0 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+      (lds_bf4_ptr)(lds_img + lk * 128));
+  bf16x4 r1 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+      (lds_bf4_ptr)(lds_img + lk * 128 + 64));
+  bf16x8 a;
+#pragma unroll
+  for (int j = 0; j < 4; ++j) { a[j] = r0[j]; a[4 + j] = r1[j]; }
+  return a;
+}
+
 template <bool BOUNDS, bool WANT_BIAS, int KT>
 __global__ __launch_bounds__(256) void conv2d_wgrad_bf16_k(
     const bf16* __restrict__ dy, const bf16* __restrict__ x,
@@ -305,8 +323,9 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_bf16_k(
     int CIp, int CI, int H, int W, int COp, int CO, int KH, int KW,
     int OH, int OW, int stride, int pad, int M, int K, int Kcan,
     int m_per_split) {
-  constexpr int MC = 64;     // m per wave per super-chunk
-  const int k0 = blockIdx.x * (KT * 16);
+  constexpr int MC = 64;          // m per chunk (shared by all 4 waves)
+  constexpr int NKF = 4 * KT;     // 16-row k-frag images per block
+  const int k0 = blockIdx.x * (NKF * 16);
   const int co0 = blockIdx.y * 16;
   const int ms = blockIdx.z;
   const int tid = threadIdx.x;
@@ -315,10 +334,8 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_bf16_k(
   const int lm = l & 15;
   const int lk = l >> 4;
 
-  // transposed chunk images for this block's 4 m-chunks (one per wave)
-  __shared__ bf16 lds_xT[4][KT * 16][MC + 8];   // [wave][k][m]
-  __shared__ bf16 lds_dyT[4][16][MC + 8];       // [wave][co][m]
-  __shared__ float lds_red[4 * KT * 256];       // wave-fold scratch
+  __shared__ bf16 lds_x[NKF][MC][16];   // [kf][m][16k] sub-images
+  __shared__ bf16 lds_dy[MC][16];       // [m][16co]
 
   const int KWC = KW * CIp;
   const int OHOW = OH * OW;
@@ -329,18 +346,16 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_bf16_k(
   int m_begin = ms * m_per_split;
   int m_end = min(M, m_begin + m_per_split);
 
-  for (int mc = m_begin; mc < m_end; mc += 4 * MC) {
+  for (int mc = m_begin; mc < m_end; mc += MC) {
     __syncthreads();
-    // ---- stage x_col^T: entries (wave w, k row, m col) -------------------
-    // reads: bf16x8 over k (contiguous within a kh row); writes: 8 scalars.
-    // thread mapping: idx = (w*MC + mloc)*(KT*2) + kg  covers all k-groups.
-    for (int idx = tid; idx < 4 * MC * (KT * 2); idx += 256) {
-      int kg = idx % (KT * 2);
-      int rest = idx / (KT * 2);
+    // ---- stage x sub-images: entries (kf, m, 8k-half) -------------------
+    for (int idx = tid; idx < NKF * MC * 2; idx += 256) {
+      int half = idx & 1;
+      int rest = idx >> 1;
       int mloc = rest % MC;
-      int w = rest / MC;
-      int m = mc + w * MC + mloc;
-      int k = k0 + kg * 8;
+      int kf = rest / MC;
+      int m = mc + mloc;
+      int k = k0 + kf * 16 + half * 8;
       bf16x8 v = bf8_zero();
       if (m < m_end && k < K) {
         int n = m / OHOW, r = m % OHOW;
@@ -357,83 +372,63 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_bf16_k(
           v = *reinterpret_cast<const bf16x8*>(
               &x[((long)(n * H + ih0 + kh) * W + iw0 + kw) * CIp + ci0]);
       }
-#pragma unroll
-      for (int j = 0; j < 8; ++j) lds_xT[w][kg * 8 + j][mloc] = v[j];
+      *reinterpret_cast<bf16x8*>(&lds_x[kf][mloc][half * 8]) = v;
     }
-    // ---- stage dy^T ------------------------------------------------------
-    for (int idx = tid; idx < 4 * MC * 2; idx += 256) {
-      int cg = idx & 1;          // two 8-co groups cover the 16-co tile
-      int rest = idx >> 1;
-      int mloc = rest % MC;
-      int w = rest / MC;
-      int m = mc + w * MC + mloc;
+    // ---- stage dy chunk --------------------------------------------------
+    for (int idx = tid; idx < MC * 2; idx += 256) {
+      int half = idx & 1;
+      int mloc = idx >> 1;
+      int m = mc + mloc;
       bf16x8 v = bf8_zero();
-      if (m < m_end && co0 + cg * 8 < COp)
-        v = *reinterpret_cast<const bf16x8*>(&dy[(long)m * COp + co0 + cg * 8]);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) lds_dyT[w][cg * 8 + j][mloc] = v[j];
+      if (m < m_end && co0 + half * 8 < COp)
+        v = *reinterpret_cast<const bf16x8*>(&dy[(long)m * COp + co0 + half * 8]);
+      *reinterpret_cast<bf16x8*>(&lds_dy[mloc][half * 8]) = v;
     }
     __syncthreads();
 
-    // ---- MFMA over this wave's chunk ------------------------------------
-    int mlim = min(MC, m_end - mc - wid * MC);
-    for (int mm = 0; mm < mlim; mm += 32) {
-      // tail: zero-fill handled because staging zero-filled beyond m_end
-      bf16x8 b = *reinterpret_cast<const bf16x8*>(&lds_dyT[wid][lm][mm + lk * 8]);
-      if (WANT_BIAS) {
+    // ---- MFMA: all waves share the dy frags, each owns KT x-images ------
+#pragma unroll
+    for (int mm = 0; mm < MC; mm += 32) {
+      bf16x8 b = tr_read_8m(&lds_dy[mm][0], lk);
+      if (WANT_BIAS && wid == 0) {
 #pragma unroll
         for (int j = 0; j < 8; ++j) db_part += to_f32(b[j]);
       }
 #pragma unroll
       for (int kf = 0; kf < KT; ++kf) {
-        bf16x8 a = *reinterpret_cast<const bf16x8*>(
-            &lds_xT[wid][kf * 16 + lm][mm + lk * 8]);
+        bf16x8 a = tr_read_8m(&lds_x[wid * KT + kf][mm][0], lk);
         acc[kf] = mfma16x16x32bf(a, b, acc[kf]);
       }
     }
   }
 
-  // ---- fold waves in LDS, write partial slab (fp32 canonical layout) -----
-  __syncthreads();
-#pragma unroll
-  for (int kf = 0; kf < KT; ++kf)
-#pragma unroll
-    for (int reg = 0; reg < 4; ++reg)
-      lds_red[((wid * KT + kf) * 256) + (lk * 4 + reg) * 16 + lm] = acc[kf][reg];
-  __syncthreads();
-  if (wid == 0) {
+  // ---- epilogue: waves own disjoint k rows; write partial slab ----------
+  {
     float* slab = dw_ws + (long)ms * (CO * (long)Kcan);
 #pragma unroll
     for (int kf = 0; kf < KT; ++kf) {
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg) {
-        int e = (lk * 4 + reg) * 16 + lm;
-        float v = lds_red[kf * 256 + e] + lds_red[(KT + kf) * 256 + e] +
-                  lds_red[(2 * KT + kf) * 256 + e] + lds_red[(3 * KT + kf) * 256 + e];
-        int kk = k0 + kf * 16 + lk * 4 + reg;   // padded k space
+        int kk = k0 + (wid * KT + kf) * 16 + lk * 4 + reg;
         int co = co0 + lm;
         if (kk < K && co < CO) {
           int kh = kk / KWC, r = kk % KWC;
           int kw = r / CIp, ci = r % CIp;
           if (ci < CI)
-            slab[((co * CI) + ci) * (KH * KW) + kh * KW + kw] = v;
+            slab[((co * CI) + ci) * (KH * KW) + kh * KW + kw] = acc[kf][reg];
         }
       }
     }
   }
-  if (WANT_BIAS && blockIdx.x == 0) {
-    // NB: D row/col note — b's lm indexes co here (B-operand col space);
-    // db_part sums dy over this lane's m-slices for co = co0+lm.
+  if (WANT_BIAS && blockIdx.x == 0 && wid == 0) {
     db_part += __shfl_down(db_part, 16);
     db_part += __shfl_down(db_part, 32);
-    __syncthreads();
-    if (lk == 0) lds_red[wid * 16 + lm] = db_part;
-    __syncthreads();
-    if (tid < 16 && co0 + tid < CO)
-      db_ws[(long)ms * CO + co0 + tid] = lds_red[tid] + lds_red[16 + tid] +
-                                         lds_red[32 + tid] + lds_red[48 + tid];
+    if (lk == 0 && co0 + lm < CO) db_ws[(long)ms * CO + co0 + lm] = db_part;
   }
 }
+
+// k-tile for the bf16 wgrad above (element rows covered per block)
+constexpr int WGRAD_BF16_KTILE = 4 * 2 * 16;  // KT=2 -> 128
 
 // ---------------------------------------------------------------------------
 // Linear forward, bf16: y[M][COp] = x[M][CIk] . w^T + b (optional ReLU).
@@ -918,4 +913,19 @@ __global__ void logsoftmax_nll_bf16_bwd_k(const float* __restrict__ logp,
       out[c] = to_bf16(v);
     }
   }
+}
+
+// tr-read mapping probe: img [64][16] f32 -> out[l][j] should be
+// img[lk*8+j][lm] if tr_read_8m implements A[row=col-idx][m=lk*8+j].
+__global__ void tr16_probe_k(const float* __restrict__ img,
+                             float* __restrict__ out) {
+  __shared__ bf16 lds[64][16];
+  int l = threadIdx.x;
+  if (l >= 64) return;
+  for (int i = l; i < 64 * 16; i += 64) lds[i / 16][i % 16] = to_bf16(img[i]);
+  __syncthreads();
+  int lm = l & 15, lk = l >> 4;
+  bf16x8 a = tr_read_8m(&lds[0][0], lk);
+  (void)lm;
+  for (int j = 0; j < 8; ++j) out[l * 8 + j] = to_f32(a[j]);
 }

@@ -404,12 +404,24 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_f32_k(
 
 // Fold the split axis of the wgrad partial slabs in fixed order
 // (deterministic — no float-atomic reassociation): out[e] = sum_s in[s*E+e].
+// Tiled so small E with large S still fills the chip: a block owns a 16-wide
+// e-tile; its 16 thread-columns stride the s axis and fold through LDS in a
+// FIXED tree order (column c holds s ≡ c (mod 16), folded c..c+8..).
 __global__ void reduce_splits_k(const float* __restrict__ in,
                                 float* __restrict__ out, long E, int S) {
-  for (long e = (long)blockIdx.x * blockDim.x + threadIdx.x; e < E;
-       e += (long)gridDim.x * blockDim.x) {
-    float v = 0.f;
-    for (int s = 0; s < S; ++s) v += in[s * E + e];
-    out[e] = v;
+  __shared__ float red[16][17];
+  int e0 = blockIdx.x * 16;
+  int el = threadIdx.x & 15;   // e within tile
+  int sl = threadIdx.x >> 4;   // s lane 0..15
+  long e = e0 + el;
+  float v = 0.f;
+  if (e < E)
+    for (int s = sl; s < S; s += 16) v += in[(long)s * E + e];
+  red[sl][el] = v;
+  __syncthreads();
+  for (int step = 8; step > 0; step >>= 1) {
+    if (sl < step) red[sl][el] += red[sl + step][el];
+    __syncthreads();
   }
+  if (sl == 0 && e < E) out[e] = red[0][el];
 }

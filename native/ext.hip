@@ -125,12 +125,13 @@ std::vector<at::Tensor> conv2d_bwd_weight(at::Tensor dy, at::Tensor x, long KH,
     else launch(F{}, F{}, K4{});
   }
   {
-    int blocks = (int)std::min<long>(2048, cdiv(dw_elems, 256));
-    hipLaunchKernelGGL(reduce_splits_k, dim3(blocks), dim3(256), 0, cur_stream(),
-                       dw_ws.data_ptr<float>(), dw.data_ptr<float>(), dw_elems, ms);
+    hipLaunchKernelGGL(reduce_splits_k, dim3((int)cdiv(dw_elems, 16)), dim3(256),
+                       0, cur_stream(), dw_ws.data_ptr<float>(),
+                       dw.data_ptr<float>(), dw_elems, ms);
     if (want_bias)
-      hipLaunchKernelGGL(reduce_splits_k, dim3(1), dim3(256), 0, cur_stream(),
-                         db_ws.data_ptr<float>(), db.data_ptr<float>(), CO, ms);
+      hipLaunchKernelGGL(reduce_splits_k, dim3(cdiv((int)CO, 16)), dim3(256), 0,
+                         cur_stream(), db_ws.data_ptr<float>(),
+                         db.data_ptr<float>(), CO, ms);
   }
   return {dw, db};
 }
@@ -380,6 +381,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_sgd_step", &fused_sgd_step);
   // bf16 NHWC engine
   m.def("mfma_probe", &mfma_probe);
+  m.def("tr16_probe", &tr16_probe);
   m.def("conv2d_fwd_bf16", &conv2d_fwd_bf16);
   m.def("conv2d_dgrad_bf16", &conv2d_dgrad_bf16);
   m.def("conv2d_wgrad_bf16", &conv2d_wgrad_bf16);
