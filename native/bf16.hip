@@ -303,13 +303,17 @@ __global__ __launch_bounds__(256) void conv2d_dgrad_bf16_k(
 // ---------------------------------------------------------------------------
 using lds_bf4_ptr = __attribute__((address_space(3))) bf16x4*;
 
-DEV bf16x8 tr_read_8m(const bf16* lds_img, int lk) {
-  // image: [m][16] bf16 row-major; returns this lane's 8 m-elements at
-  // column (l&15), rows lk*8 .. lk*8+7 (two 4-row transpose reads).
-  bf16x4 r0 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
-      (lds_bf4_ptr)(lds_img + lk * 128));
-  bf16x4 r1 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
-      (lds_bf4_ptr)(lds_img + lk * 128 + 64));
+DEV bf16x8 tr_read_8m(const bf16* lds_img, int l) {
+  // image: [m][16] bf16 row-major. Returns this lane's 8 m-elements at
+  // column (l&15), rows (l>>4)*8 .. +7, via two 4-row transpose reads.
+  // ds_read_b64_tr_b16 semantics (verified by tr16_probe on-device): with
+  // the NATURAL per-lane addressing addr_l = base + lane_in_group*8B, a
+  // 16-lane group collectively reads 64 contiguous bf16 (4 rows x 16) and
+  // lane l receives column (l&15), element j = row j of the 4-row block.
+  int lm = l & 15, lk = l >> 4;
+  const bf16* p0 = lds_img + lk * 128 + lm * 4;
+  bf16x4 r0 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16((lds_bf4_ptr)p0);
+  bf16x4 r1 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16((lds_bf4_ptr)(p0 + 64));
   bf16x8 a;
 #pragma unroll
   for (int j = 0; j < 4; ++j) { a[j] = r0[j]; a[4 + j] = r1[j]; }
@@ -389,14 +393,14 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_bf16_k(
     // ---- MFMA: all waves share the dy frags, each owns KT x-images ------
 #pragma unroll
     for (int mm = 0; mm < MC; mm += 32) {
-      bf16x8 b = tr_read_8m(&lds_dy[mm][0], lk);
+      bf16x8 b = tr_read_8m(&lds_dy[mm][0], l);
       if (WANT_BIAS && wid == 0) {
 #pragma unroll
         for (int j = 0; j < 8; ++j) db_part += to_f32(b[j]);
       }
 #pragma unroll
       for (int kf = 0; kf < KT; ++kf) {
-        bf16x8 a = tr_read_8m(&lds_x[wid * KT + kf][mm][0], lk);
+        bf16x8 a = tr_read_8m(&lds_x[wid * KT + kf][mm][0], l);
         acc[kf] = mfma16x16x32bf(a, b, acc[kf]);
       }
     }
@@ -925,7 +929,7 @@ __global__ void tr16_probe_k(const float* __restrict__ img,
   for (int i = l; i < 64 * 16; i += 64) lds[i / 16][i % 16] = to_bf16(img[i]);
   __syncthreads();
   int lm = l & 15, lk = l >> 4;
-  bf16x8 a = tr_read_8m(&lds[0][0], lk);
+  bf16x8 a = tr_read_8m(&lds[0][0], l);
   (void)lm;
   for (int j = 0; j < 8; ++j) out[l * 8 + j] = to_f32(a[j]);
 }
