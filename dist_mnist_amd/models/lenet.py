@@ -72,8 +72,8 @@ class Net(nn.Module):
             # canonical NCHW (c, h, w) -> perm maps padded k -> canonical ci
             self._fc1_perm = fb.build_flatten_perm(4, 4, 24, 20, x.device)
             self._fc2_perm = fb.build_identity_perm(56, 50, x.device)
-        x = fb.conv2d_bf16(x, self.conv1.weight, self.conv1.bias, cop=16)
-        x = fb.maxpool2x2_relu_bf16(x)
+        # fused stem: conv1 + pool + relu in one kernel (no y1 round-trip)
+        x = fb.conv_pool_stem_bf16(x, self.conv1.weight, self.conv1.bias, cop=16)
         x = fb.conv2d_bf16(x, self.conv2.weight, self.conv2.bias, cop=24)
         x = fb.dropout_bf16(x, p=self.conv2_drop.p, training=self.training,
                             channelwise=True)

@@ -193,3 +193,31 @@ class _LogSoftmaxNllBF16Fn(torch.autograd.Function):
 
 def logsoftmax_nll_bf16(x, target, c=10):
     return _LogSoftmaxNllBF16Fn.apply(x, target, c)
+
+
+class _ConvPoolStemBF16Fn(torch.autograd.Function):
+    """Fused Conv2d(CI<=2, s1 p0) + MaxPool2x2 + ReLU (the LeNet stem).
+    Backward produces only (dw, db) — the stem is the input layer, so no
+    dgrad is ever needed (needs_input_grad[0] is False for loader tensors)."""
+
+    @staticmethod
+    def forward(ctx, x, w, b, cop):
+        p, idx = native.ext().conv_pool_stem_bf16(x.contiguous(), w.contiguous(),
+                                                  b.contiguous(), cop)
+        ctx.save_for_backward(x, p, idx, w)
+        return p
+
+    @staticmethod
+    def backward(ctx, dpool):
+        x, p, idx, w = ctx.saved_tensors
+        if ctx.needs_input_grad[0]:
+            raise RuntimeError("conv_pool_stem has no dgrad (input layer only)")
+        dw, db = native.ext().conv2d_wgrad_pool_bf16(
+            dpool.contiguous(), p, idx, x, w.shape[1], w.shape[0], w.shape[2],
+            w.shape[3])
+        return None, dw, db, None
+
+
+def conv_pool_stem_bf16(x, w, b, cop=None):
+    cop = cop or pad8(w.shape[0])
+    return _ConvPoolStemBF16Fn.apply(x, w, b, cop)

@@ -340,9 +340,24 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_bf16_k(
 
   __shared__ bf16 lds_x[NKF][MC][16];   // [kf][m][16k] sub-images
   __shared__ bf16 lds_dy[MC][16];       // [m][16co]
+  __shared__ int lds_mbase[MC];         // x elem offset of (n, ih0, iw0)
+  __shared__ int lds_mhw[MC];           // biased (ih0, iw0) for bounds
 
   const int KWC = KW * CIp;
   const int OHOW = OH * OW;
+
+  // per-lane k decode for the NKF*2 8-k groups, fixed over the reduction
+  int g_off[NKF * 2];   // (kh*W + kw)*CIp + ci0
+  int g_khw[NKF * 2];   // (kh<<16)|kw
+#pragma unroll
+  for (int g = 0; g < NKF * 2; ++g) {
+    int k = k0 + g * 8;
+    int kcl = k < K ? k : 0;
+    int kh = kcl / KWC, kr = kcl % KWC;
+    int kw = kr / CIp, ci0 = kr % CIp;
+    g_off[g] = (kh * W + kw) * CIp + ci0;
+    g_khw[g] = (kh << 16) | kw;
+  }
 
   f32x4 acc[KT] = {};
   float db_part = 0.f;
@@ -352,29 +367,38 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_bf16_k(
 
   for (int mc = m_begin; mc < m_end; mc += MC) {
     __syncthreads();
+    // ---- per-chunk m decode, once --------------------------------------
+    if (tid < MC) {
+      int m = mc + tid;
+      int mm = m < m_end ? m : 0;
+      int n = mm / OHOW, r = mm % OHOW;
+      int oh = r / OW, ow = r % OW;
+      int ih0 = oh * stride - pad, iw0 = ow * stride - pad;
+      lds_mbase[tid] = (n * H + ih0) * W * CIp + iw0 * CIp;
+      lds_mhw[tid] = ((ih0 + 16384) << 16) | (iw0 + 16384);
+    }
+    __syncthreads();
     // ---- stage x sub-images: entries (kf, m, 8k-half) -------------------
     for (int idx = tid; idx < NKF * MC * 2; idx += 256) {
       int half = idx & 1;
       int rest = idx >> 1;
       int mloc = rest % MC;
       int kf = rest / MC;
+      int g = kf * 2 + half;
       int m = mc + mloc;
-      int k = k0 + kf * 16 + half * 8;
+      int k = k0 + g * 8;
       bf16x8 v = bf8_zero();
       if (m < m_end && k < K) {
-        int n = m / OHOW, r = m % OHOW;
-        int oh = r / OW, ow = r % OW;
-        int ih0 = oh * stride - pad, iw0 = ow * stride - pad;
-        int kh = k / KWC, kr = k % KWC;
-        int kw = kr / CIp, ci0 = kr % CIp;
         bool ok = true;
         if (BOUNDS) {
-          int ih = ih0 + kh, iw = iw0 + kw;
+          int hw = lds_mhw[mloc];
+          int khw = g_khw[g];
+          int ih = (hw >> 16) - 16384 + (khw >> 16);
+          int iw = (hw & 0xffff) - 16384 + (khw & 0xffff);
           ok = ih >= 0 && ih < H && iw >= 0 && iw < W;
         }
         if (ok)
-          v = *reinterpret_cast<const bf16x8*>(
-              &x[((long)(n * H + ih0 + kh) * W + iw0 + kw) * CIp + ci0]);
+          v = *reinterpret_cast<const bf16x8*>(&x[lds_mbase[mloc] + g_off[g]]);
       }
       *reinterpret_cast<bf16x8*>(&lds_x[kf][mloc][half * 8]) = v;
     }
@@ -932,4 +956,220 @@ __global__ void tr16_probe_k(const float* __restrict__ img,
   bf16x8 a = tr_read_8m(&lds[0][0], l);
   (void)lm;
   for (int j = 0; j < 8; ++j) out[l * 8 + j] = to_f32(a[j]);
+}
+
+// ---------------------------------------------------------------------------
+// Fused stem: Conv2d(CI<=2, s=1, p=0) + MaxPool2x2 + ReLU in ONE kernel.
+// For LeNet conv1 (CI=1) the MFMA path pays 8x padding waste and three full
+// activation round-trips (y1 store, pool read, pool store = 2x151MB + 38MB at
+// batch 8192); this kernel writes only the pooled output + argmax, computing
+// the 4 conv pixels per pool cell directly on the VALU (25*CI FMAs each —
+// conv1 is ~1.9 GFLOP at batch 8192, trivial next to the saved bandwidth).
+// Thread = (n, pool cell, 8-co group). Backward = conv2d_wgrad_pool_bf16
+// (no dgrad: the stem is the input layer).
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void conv_pool_stem_bf16_k(
+    const bf16* __restrict__ x, const float* __restrict__ w,
+    const float* __restrict__ bias, bf16* __restrict__ p,
+    unsigned char* __restrict__ idx,
+    int CIp, int CI, int H, int W, int COp, int CO, int KH, int KW,
+    int OH2, int OW2, int N) {
+  // weights in LDS, fp32: [co][CI*KH*KW]
+  extern __shared__ float lds_wf[];
+  const int tid = threadIdx.x;
+  const int Kc = CI * KH * KW;
+  for (int i = tid; i < COp * Kc; i += 256) {
+    int co = i / Kc, k = i % Kc;
+    lds_wf[i] = (co < CO) ? w[co * Kc + k] : 0.f;
+  }
+  __syncthreads();
+
+  long total = (long)N * OH2 * OW2 * (COp / 8);
+  for (long t = (long)blockIdx.x * blockDim.x + tid; t < total;
+       t += (long)gridDim.x * blockDim.x) {
+    int cg = t % (COp / 8);
+    long r = t / (COp / 8);
+    int ow2 = r % OW2;
+    long r2 = r / OW2;
+    int oh2 = r2 % OH2;
+    int n = r2 / OH2;
+
+    // x tile (KH+1) x (KW+1) x CI into registers (<= 6*6*2)
+    float xt[6 * 6 * 2];
+    const int TH = KH + 1, TW = KW + 1;
+    const bf16* xb = x + ((long)(n * H + oh2 * 2) * W + ow2 * 2) * CIp;
+#pragma unroll 2
+    for (int ci = 0; ci < CI; ++ci)
+      for (int ty = 0; ty < TH; ++ty)
+        for (int tx = 0; tx < TW; ++tx)
+          xt[(ci * TH + ty) * TW + tx] = to_f32(xb[(ty * W + tx) * CIp + ci]);
+
+    bf16x8 pout;
+    unsigned char am[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int co = cg * 8 + j;
+      float acc[4];
+      float bval = (co < CO) ? bias[co] : 0.f;
+      acc[0] = acc[1] = acc[2] = acc[3] = bval;
+      if (co < CO) {
+        const float* wr = &lds_wf[co * Kc];
+        for (int ci = 0; ci < CI; ++ci)
+          for (int kh = 0; kh < KH; ++kh)
+            for (int kw = 0; kw < KW; ++kw) {
+              float wv = wr[(ci * KH + kh) * KW + kw];
+              const float* xrow = &xt[(ci * TH + kh) * TW + kw];
+              acc[0] += wv * xrow[0];
+              acc[1] += wv * xrow[1];
+              acc[2] += wv * xrow[TW];
+              acc[3] += wv * xrow[TW + 1];
+            }
+      }
+      float m01 = fmaxf(acc[0], acc[1]), m23 = fmaxf(acc[2], acc[3]);
+      float mx = fmaxf(m01, m23);
+      unsigned char a;
+      if (mx == acc[0]) a = 0; else if (mx == acc[1]) a = 1;
+      else if (mx == acc[2]) a = 2; else a = 3;
+      pout[j] = to_bf16((co < CO) ? fmaxf(mx, 0.f) : 0.f);
+      am[j] = a;
+    }
+    long off = (((long)(n * OH2 + oh2) * OW2 + ow2) * COp) + cg * 8;
+    *reinterpret_cast<bf16x8*>(p + off) = pout;
+    *reinterpret_cast<unsigned long long*>(idx + off) =
+        *reinterpret_cast<unsigned long long*>(am);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// wgrad for the fused stem: identical reduction to conv2d_wgrad_bf16 but the
+// conv-space dy is reconstructed on the fly from (d_pool, pool_y, idx):
+//   dy_conv[n,oh,ow,co] = (idx==quadrant && pool_y>0) ? d_pool : 0.
+// ---------------------------------------------------------------------------
+template <int KT>
+__global__ __launch_bounds__(256) void conv2d_wgrad_pool_bf16_k(
+    const bf16* __restrict__ dpool, const bf16* __restrict__ pool_y,
+    const unsigned char* __restrict__ pidx, const bf16* __restrict__ x,
+    float* __restrict__ dw_ws, float* __restrict__ db_ws,
+    int CIp, int CI, int H, int W, int COp, int CO, int KH, int KW,
+    int OH, int OW, int M, int K, int Kcan, int m_per_split) {
+  constexpr int MC = 64;
+  constexpr int NKF = 4 * KT;
+  const int k0 = blockIdx.x * (NKF * 16);
+  const int co0 = blockIdx.y * 16;
+  const int ms = blockIdx.z;
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6;
+  const int l = tid & 63;
+  const int lm = l & 15;
+  const int lk = l >> 4;
+  const int OH2 = OH / 2, OW2 = OW / 2;
+
+  __shared__ bf16 lds_x[NKF][MC][16];
+  __shared__ bf16 lds_dy[MC][16];
+  __shared__ int lds_mbase[MC];
+
+  const int KWC = KW * CIp;
+  const int OHOW = OH * OW;
+
+  int g_off[NKF * 2];
+#pragma unroll
+  for (int g = 0; g < NKF * 2; ++g) {
+    int k = k0 + g * 8;
+    int kcl = k < K ? k : 0;
+    int kh = kcl / KWC, kr = kcl % KWC;
+    int kw = kr / CIp, ci0 = kr % CIp;
+    g_off[g] = (kh * W + kw) * CIp + ci0;
+  }
+
+  f32x4 acc[KT] = {};
+  float db_part = 0.f;
+
+  int m_begin = ms * m_per_split;
+  int m_end = min(M, m_begin + m_per_split);
+
+  for (int mc = m_begin; mc < m_end; mc += MC) {
+    __syncthreads();
+    if (tid < MC) {
+      int m = mc + tid;
+      int mm = m < m_end ? m : 0;
+      int n = mm / OHOW, r = mm % OHOW;
+      int oh = r / OW, ow = r % OW;
+      lds_mbase[tid] = (n * H + oh) * W * CIp + ow * CIp;  // pad=0, s=1
+    }
+    __syncthreads();
+    for (int idx2 = tid; idx2 < NKF * MC * 2; idx2 += 256) {
+      int half = idx2 & 1;
+      int rest = idx2 >> 1;
+      int mloc = rest % MC;
+      int kf = rest / MC;
+      int g = kf * 2 + half;
+      int m = mc + mloc;
+      int k = k0 + g * 8;
+      bf16x8 v = bf8_zero();
+      if (m < m_end && k < K)
+        v = *reinterpret_cast<const bf16x8*>(&x[lds_mbase[mloc] + g_off[g]]);
+      *reinterpret_cast<bf16x8*>(&lds_x[kf][mloc][half * 8]) = v;
+    }
+    // dy_conv reconstruction, vectorized over 8 co
+    for (int idx2 = tid; idx2 < MC * 2; idx2 += 256) {
+      int half = idx2 & 1;
+      int mloc = idx2 >> 1;
+      int m = mc + mloc;
+      bf16x8 v = bf8_zero();
+      if (m < m_end && co0 + half * 8 < COp) {
+        int n = m / OHOW, r = m % OHOW;
+        int oh = r / OW, ow = r % OW;
+        unsigned char q = (unsigned char)((oh & 1) * 2 + (ow & 1));
+        long off = (((long)(n * OH2 + oh / 2) * OW2 + ow / 2) * COp) + co0 + half * 8;
+        bf16x8 dp = *reinterpret_cast<const bf16x8*>(&dpool[off]);
+        bf16x8 py = *reinterpret_cast<const bf16x8*>(&pool_y[off]);
+        unsigned long long am8 =
+            *reinterpret_cast<const unsigned long long*>(&pidx[off]);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          bool take = (((am8 >> (8 * j)) & 0xff) == q) && (to_f32(py[j]) > 0.f);
+          v[j] = take ? dp[j] : to_bf16(0.f);
+        }
+      }
+      *reinterpret_cast<bf16x8*>(&lds_dy[mloc][half * 8]) = v;
+    }
+    __syncthreads();
+
+#pragma unroll
+    for (int mm = 0; mm < MC; mm += 32) {
+      bf16x8 b = tr_read_8m(&lds_dy[mm][0], l);
+      if (wid == 0) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) db_part += to_f32(b[j]);
+      }
+#pragma unroll
+      for (int kf = 0; kf < KT; ++kf) {
+        bf16x8 a = tr_read_8m(&lds_x[wid * KT + kf][mm][0], l);
+        acc[kf] = mfma16x16x32bf(a, b, acc[kf]);
+      }
+    }
+  }
+
+  {
+    float* slab = dw_ws + (long)ms * (CO * (long)Kcan);
+#pragma unroll
+    for (int kf = 0; kf < KT; ++kf) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        int kk = k0 + (wid * KT + kf) * 16 + lk * 4 + reg;
+        int co = co0 + lm;
+        if (kk < K && co < CO) {
+          int kh = kk / KWC, r = kk % KWC;
+          int kw = r / CIp, ci = r % CIp;
+          if (ci < CI)
+            slab[((co * CI) + ci) * (KH * KW) + kh * KW + kw] = acc[kf][reg];
+        }
+      }
+    }
+  }
+  if (blockIdx.x == 0 && wid == 0) {
+    db_part += __shfl_down(db_part, 16);
+    db_part += __shfl_down(db_part, 32);
+    if (lk == 0 && co0 + lm < CO) db_ws[(long)ms * CO + co0 + lm] = db_part;
+  }
 }

@@ -250,3 +250,29 @@ def test_training_loss_decreases_bf16():
         losses.append(loss.item())
     tail = sum(losses[-10:]) / 10
     assert tail < losses[0] * 0.75, (losses[0], tail, losses[::30])
+
+
+def test_conv_pool_stem_fused():
+    """Fused conv1+pool+relu vs torch oracle, fwd + wgrad/bias grads."""
+    N, CI, H, W, CO = 5, 1, 28, 28, 10
+    x = randn(N, CI, H, W, seed=40)
+    w = randn(CO, CI, 5, 5, seed=41, scale=0.3).requires_grad_(True)
+    b = randn(CO, seed=42).requires_grad_(True)
+    xb = fb.nchw_to_nhwc_padded(x)
+    p = fb.conv_pool_stem_bf16(xb, w, b, cop=16)
+    assert p.shape == (N, 12, 12, 16)
+
+    q = lambda t: t.to(torch.bfloat16).float()
+    wr = q(w.detach()).requires_grad_(True)
+    br = b.detach().clone().requires_grad_(True)
+    ref = F.relu(F.max_pool2d(F.conv2d(q(x), wr, br), 2))
+    torch.testing.assert_close(fb.nhwc_to_nchw(p, CO), ref, rtol=2e-2, atol=3e-2)
+    assert (p[..., CO:].float() == 0).all()
+
+    dy = randn(N, CO, 12, 12, seed=43)
+    p.backward(fb.nchw_to_nhwc_padded(dy, cp=16))
+    ref.backward(q(dy))
+    torch.testing.assert_close(w.grad, wr.grad, rtol=2e-2,
+                               atol=2e-2 * max(1.0, wr.grad.abs().max().item()))
+    torch.testing.assert_close(b.grad, br.grad, rtol=2e-2,
+                               atol=2e-2 * max(1.0, br.grad.abs().max().item()))
