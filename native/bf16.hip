@@ -968,11 +968,12 @@ __global__ void tr16_probe_k(const float* __restrict__ img,
 // Thread = (n, pool cell, 8-co group). Backward = conv2d_wgrad_pool_bf16
 // (no dgrad: the stem is the input layer).
 // ---------------------------------------------------------------------------
+template <int CI, int KH, int KW>
 __global__ __launch_bounds__(256) void conv_pool_stem_bf16_k(
     const bf16* __restrict__ x, const float* __restrict__ w,
     const float* __restrict__ bias, bf16* __restrict__ p,
     unsigned char* __restrict__ idx,
-    int CIp, int CI, int H, int W, int COp, int CO, int KH, int KW,
+    int CIp, int H, int W, int COp, int CO,
     int OH2, int OW2, int N) {
   // weights in LDS, fp32: [co][CI*KH*KW]
   extern __shared__ float lds_wf[];
@@ -994,13 +995,17 @@ __global__ __launch_bounds__(256) void conv_pool_stem_bf16_k(
     int oh2 = r2 % OH2;
     int n = r2 / OH2;
 
-    // x tile (KH+1) x (KW+1) x CI into registers (<= 6*6*2)
-    float xt[6 * 6 * 2];
-    const int TH = KH + 1, TW = KW + 1;
+    // x tile (KH+1) x (KW+1) x CI in registers; all loop bounds are
+    // compile-time so xt stays in VGPRs (guide rule 20: runtime-indexed
+    // arrays spill to scratch)
+    constexpr int TH = KH + 1, TW = KW + 1;
+    float xt[CI * TH * TW];
     const bf16* xb = x + ((long)(n * H + oh2 * 2) * W + ow2 * 2) * CIp;
-#pragma unroll 2
+#pragma unroll
     for (int ci = 0; ci < CI; ++ci)
+#pragma unroll
       for (int ty = 0; ty < TH; ++ty)
+#pragma unroll
         for (int tx = 0; tx < TW; ++tx)
           xt[(ci * TH + ty) * TW + tx] = to_f32(xb[(ty * W + tx) * CIp + ci]);
 
@@ -1014,8 +1019,11 @@ __global__ __launch_bounds__(256) void conv_pool_stem_bf16_k(
       acc[0] = acc[1] = acc[2] = acc[3] = bval;
       if (co < CO) {
         const float* wr = &lds_wf[co * Kc];
+#pragma unroll
         for (int ci = 0; ci < CI; ++ci)
+#pragma unroll
           for (int kh = 0; kh < KH; ++kh)
+#pragma unroll
             for (int kw = 0; kw < KW; ++kw) {
               float wv = wr[(ci * KH + kh) * KW + kw];
               const float* xrow = &xt[(ci * TH + kh) * TW + kw];
