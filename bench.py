@@ -41,6 +41,8 @@ def parse_args():
                    help="compute dtype of the hot path (bf16 = NHWC native engine, "
                         "the BASELINE config-4 headline; fp32 = NCHW engine)")
     p.add_argument("--bucket-mb", type=float, default=25.0)
+    p.add_argument("--no-graphs", action="store_true",
+                   help="disable hipGraph step capture (eager launches)")
     return p.parse_args()
 
 
@@ -74,8 +76,24 @@ def main():
     batches = Prefetcher(stream, device,
                          dtype=torch.bfloat16 if use_bf16 else torch.float32)
 
+    use_graphs = use_cuda and not args.no_graphs
+    graphed = None
+    step_idx = [0]
+    if use_graphs:
+        from dist_mnist_amd.runtime import GraphedStep
+
+        x0, t0 = next(batches)
+        ddp = model if isinstance(model, DistributedDataParallel) else None
+        graphed = GraphedStep(model, opt, x0, t0, ddp=ddp)
+
     def one_step():
         x, t = next(batches)
+        if graphed is not None:
+            slot = step_idx[0] % len(graphed.slots)
+            step_idx[0] += 1
+            graphed.slots[slot][0].copy_(x, non_blocking=True)
+            graphed.slots[slot][1].copy_(t, non_blocking=True)
+            return graphed.step(slot)
         if isinstance(model, DistributedDataParallel):
             model.zero_grad()
         else:

@@ -110,16 +110,25 @@ def maxpool2x2_relu_bf16(x):
     return _MaxPoolReluBF16Fn.apply(x)
 
 
-_counter = [0]
+_ctr_dev: dict = {}
+
+
+def _dropout_ctr(device):
+    """Per-device RNG stream counter living in DEVICE memory so hipGraph
+    replays of a captured step still draw fresh dropout masks (the bump
+    kernel runs inside the capture)."""
+    key = str(device)
+    if key not in _ctr_dev:
+        _ctr_dev[key] = torch.zeros(1, dtype=torch.int64, device=device)
+    return _ctr_dev[key]
 
 
 class _DropoutBF16Fn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, p, channelwise, seed):
-        _counter[0] += 1
         y, mask = native.ext().dropout_bf16_fwd(x.contiguous(), p,
                                                 bool(channelwise), int(seed),
-                                                _counter[0])
+                                                _dropout_ctr(x.device))
         ctx.save_for_backward(mask)
         ctx.meta = (p, channelwise, x.shape)
         return y

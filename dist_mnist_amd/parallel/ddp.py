@@ -137,6 +137,14 @@ class DistributedDataParallel(nn.Module):
             if id(p) not in self._param_bucket and p.grad is not None:
                 p.grad.zero_()
 
+    def disable_comm_hooks(self):
+        """Remove the backward-overlap hooks (used by the hipGraph step:
+        collectives stay OUTSIDE capture; GraphedStep issues one eager
+        all-reduce per bucket after the captured backward)."""
+        for h in self._grad_accum_hooks:
+            h.remove()
+        self._grad_accum_hooks = []
+
     def sync(self):
         """Wait all in-flight bucket all-reduces and divide by world size.
         Reference parity: SUM + /world."""

@@ -1,1 +1,2 @@
+from .graphs import GraphedStep  # noqa: F401
 from .prefetch import Prefetcher  # noqa: F401

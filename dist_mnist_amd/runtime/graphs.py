@@ -1,0 +1,100 @@
+"""hipGraph-captured training step (guide: "HIP streams and graphs instead of
+a tracing compiler"; SURVEY.md §7: at 21k params LeNet is launch-latency-bound
+— the whole fwd+bwd+SGD step captures into one executable graph).
+
+Two static input slots, one captured graph per slot: the Prefetcher's side
+stream fills slot t+1 while the compute stream replays slot t's graph (event
+synced both ways, never a host sync).
+
+Distributed mode splits the step into [fwd+bwd graph] -> eager bucket
+all-reduce (RCCL; kept OUTSIDE capture so multi-GPU replay never depends on
+collective capture support) -> [scale+SGD graph]. LeNet's gradient is one
+87 KB bucket: the collective is pure xGMI latency, so losing backward overlap
+costs nothing at this scale (SURVEY.md §5 comm).
+"""
+
+from __future__ import annotations
+
+import torch
+import torch.distributed as dist
+
+
+class GraphedStep:
+    def __init__(self, model, optimizer, sample_x, sample_t, ddp=None,
+                 n_slots: int = 2, warmup: int = 3):
+        """model: Net in bf16-native or fp32 mode (already on device/train).
+        ddp: DistributedDataParallel wrapper (enabled) or None.
+        sample_x/sample_t: device tensors of the static batch shape."""
+        self.model = model
+        self.opt = optimizer
+        self.ddp = ddp
+        if ddp is not None:
+            ddp.disable_comm_hooks()
+        self.device = sample_x.device
+        self.slots = []
+        for _ in range(n_slots):
+            self.slots.append((sample_x.clone(), sample_t.clone()))
+
+        side = torch.cuda.Stream(device=self.device)
+        # warmup on a side stream (capture requires non-default-stream history)
+        side.wait_stream(torch.cuda.current_stream(self.device))
+        with torch.cuda.stream(side):
+            for _ in range(warmup):
+                self._eager_fwd_bwd(*self.slots[0])
+                self._eager_opt()
+        torch.cuda.current_stream(self.device).wait_stream(side)
+        torch.cuda.synchronize(self.device)
+
+        # capture one fwd+bwd graph per slot
+        self.fb_graphs = []
+        self.losses = []
+        pool = None
+        for x, t in self.slots:
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g, pool=pool):
+                loss = self._eager_fwd_bwd(x, t)
+            pool = g.pool()
+            self.fb_graphs.append(g)
+            self.losses.append(loss)
+
+        # optimizer graph (shared; includes DDP 1/world scaling when needed)
+        self.opt_graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.opt_graph, pool=pool):
+            self._eager_opt(scale=True)
+
+    # -- eager pieces (also used during warmup/capture) ---------------------
+    def _zero_grads(self):
+        if self.ddp is not None:
+            self.ddp.zero_grad()
+        else:
+            for p in self.model.parameters():
+                if p.grad is not None:
+                    p.grad.zero_()
+                else:
+                    p.grad = torch.zeros_like(p)
+
+    def _eager_fwd_bwd(self, x, t):
+        self._zero_grads()
+        _, loss = self.model(x, t)
+        loss.backward()
+        return loss
+
+    def _eager_opt(self, scale: bool = False):
+        if scale and self.ddp is not None and self.ddp.enabled:
+            inv = 1.0 / self.ddp.world_size
+            for b in self.ddp.buckets:
+                b.flat.mul_(inv)
+        self.opt.step()
+
+    # -- per-step ------------------------------------------------------------
+    def step(self, slot: int):
+        """Replay the step for `slot` (whose input buffers must be filled and
+        event-synced by the caller). Returns the loss tensor (device)."""
+        self.fb_graphs[slot].replay()
+        if self.ddp is not None and self.ddp.enabled:
+            # one eager all-reduce per bucket (grads are bucket views filled
+            # by the captured backward)
+            for b in self.ddp.buckets:
+                dist.all_reduce(b.flat, op=dist.ReduceOp.SUM, group=self.ddp.pg)
+        self.opt_graph.replay()
+        return self.losses[slot]

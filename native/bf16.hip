@@ -818,13 +818,18 @@ __global__ void maxpool2x2_relu_bf16_bwd_k(const bf16* __restrict__ dy,
 // of the vector (the 8 lanes are 8 consecutive channels of one pixel).
 // mask: u8 per element (elementwise) or per (n,c) (channelwise).
 // ---------------------------------------------------------------------------
+__global__ void bump_u64_k(unsigned long long* ctr) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) *ctr += 1;
+}
+
 __global__ void dropout_bf16_fwd_k(const bf16* __restrict__ x,
                                    bf16* __restrict__ y,
                                    unsigned char* __restrict__ mask,
                                    long n8, int Cp, long hwc, float p,
                                    int channelwise,
                                    unsigned long long seed,
-                                   unsigned long long ctr) {
+                                   const unsigned long long* __restrict__ ctr_p) {
+  unsigned long long ctr = *ctr_p;
   float inv = 1.f / (1.f - p);
   unsigned int pthresh = (unsigned int)(p * 256.f);
   for (long i8 = (long)blockIdx.x * blockDim.x + threadIdx.x; i8 < n8;
