@@ -125,9 +125,10 @@ class SyntheticBatchStream:
             x = torch.randn(batch_size, 1, 28, 28, generator=g, dtype=torch.float32)
             y = torch.randint(0, 10, (batch_size,), generator=g)
             if layout == "nhwc_bf16":
-                xp = torch.zeros(batch_size, 28, 28, 8, dtype=torch.bfloat16)
-                xp[..., 0] = x[:, 0].to(torch.bfloat16)
-                x = xp
+                # CIp=1: the fused stem kernel reads scalar channels, so the
+                # input needs NO channel padding — 12.8 MB/batch over PCIe
+                # instead of 102 MB (the H2D copy must hide under compute)
+                x = x[:, 0].to(torch.bfloat16).unsqueeze(-1).contiguous()
             else:
                 x = x.to(dtype)
             if pin:

@@ -42,7 +42,8 @@ class Net(nn.Module):
 
     def bf16_native_(self, enable: bool = True):
         """Switch forward to the bf16 NHWC kernel path (GPU only). Inputs
-        must then be NHWC bf16 [N, 28, 28, 8] (see ops.functional_bf16).
+        must then be NHWC bf16 [N, 28, 28, C] (C=1 unpadded for the fused
+        stem, or padded-to-8 for the generic conv path).
         Master weights stay fp32 canonical — checkpoints are unchanged."""
         self._bf16_native = enable
         return self
