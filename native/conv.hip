@@ -409,19 +409,25 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_f32_k(
 // FIXED tree order (column c holds s ≡ c (mod 16), folded c..c+8..).
 __global__ void reduce_splits_k(const float* __restrict__ in,
                                 float* __restrict__ out, long E, int S) {
+  // block (x = 16-wide e-tile, y = z of Z): folds s in {z, z+Z, ...} in a
+  // FIXED order (deterministic) into out[z*E + e]. Z = gridDim.y; the host
+  // helper runs one pass (Z=1) for small S or two passes (Z=16 partials,
+  // then Z=1 over the 16 partials) so large split counts still fill the chip.
   __shared__ float red[16][17];
+  int Z = gridDim.y;
+  int z = blockIdx.y;
   int e0 = blockIdx.x * 16;
-  int el = threadIdx.x & 15;   // e within tile
-  int sl = threadIdx.x >> 4;   // s lane 0..15
+  int el = threadIdx.x & 15;
+  int sl = threadIdx.x >> 4;
   long e = e0 + el;
   float v = 0.f;
   if (e < E)
-    for (int s = sl; s < S; s += 16) v += in[(long)s * E + e];
+    for (int s = z + sl * Z; s < S; s += 16 * Z) v += in[(long)s * E + e];
   red[sl][el] = v;
   __syncthreads();
   for (int step = 8; step > 0; step >>= 1) {
     if (sl < step) red[sl][el] += red[sl + step][el];
     __syncthreads();
   }
-  if (sl == 0 && e < E) out[e] = red[0][el];
+  if (sl == 0 && e < E) out[(long)z * E + e] = red[0][el];
 }

@@ -22,6 +22,24 @@ inline hipStream_t cur_stream() {
   return c10::hip::getCurrentHIPStream().stream();
 }
 
+// deterministic fold of wgrad partial slabs: 1 pass for small S, 2 for large
+inline void launch_reduce_splits(const at::Tensor& ws, at::Tensor& out, long E,
+                                 int S) {
+  if (S > 64) {
+    auto mid = at::empty({16, E}, out.options());
+    hipLaunchKernelGGL(reduce_splits_k, dim3((int)cdiv(E, 16), 16), dim3(256),
+                       0, cur_stream(), ws.data_ptr<float>(),
+                       mid.data_ptr<float>(), E, S);
+    hipLaunchKernelGGL(reduce_splits_k, dim3((int)cdiv(E, 16), 1), dim3(256),
+                       0, cur_stream(), mid.data_ptr<float>(),
+                       out.data_ptr<float>(), E, 16);
+  } else {
+    hipLaunchKernelGGL(reduce_splits_k, dim3((int)cdiv(E, 16), 1), dim3(256),
+                       0, cur_stream(), ws.data_ptr<float>(),
+                       out.data_ptr<float>(), E, S);
+  }
+}
+
 // ---------------------------------------------------------------------------
 at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias, long stride,
                       long pad) {
@@ -125,13 +143,8 @@ std::vector<at::Tensor> conv2d_bwd_weight(at::Tensor dy, at::Tensor x, long KH,
     else launch(F{}, F{}, K4{});
   }
   {
-    hipLaunchKernelGGL(reduce_splits_k, dim3((int)cdiv(dw_elems, 16)), dim3(256),
-                       0, cur_stream(), dw_ws.data_ptr<float>(),
-                       dw.data_ptr<float>(), dw_elems, ms);
-    if (want_bias)
-      hipLaunchKernelGGL(reduce_splits_k, dim3(cdiv((int)CO, 16)), dim3(256), 0,
-                         cur_stream(), db_ws.data_ptr<float>(),
-                         db.data_ptr<float>(), CO, ms);
+    launch_reduce_splits(dw_ws, dw, dw_elems, ms);
+    if (want_bias) launch_reduce_splits(db_ws, db, CO, ms);
   }
   return {dw, db};
 }
