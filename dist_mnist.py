@@ -60,6 +60,9 @@ def parse_args(argv=None):
                    help="DDP gradient bucket size in MiB (xGMI-tuned default)")
     p.add_argument("--ddp", choices=["bucketed", "reference"], default="bucketed",
                    help="bucketed = overlapped engine; reference = per-param blocking allreduce")
+    p.add_argument("--dtype", choices=["fp32", "bf16"], default="fp32",
+                   help="bf16 = native NHWC engine on GPU (fp32 master weights; "
+                        "checkpoints unchanged)")
     p.add_argument("--resume", type=str, default=None)
     p.add_argument("--steps-per-epoch", type=int, default=None,
                    help="cap steps per epoch (synthetic data is unbounded)")
@@ -74,7 +77,10 @@ def run(rank: int, world_size: int, args):
 
     loader, bsz = partition_dataset(rank, world_size, global_batch=args.batch_size,
                                     n=args.dataset_size, seed=args.seed)
+    use_bf16 = args.dtype == "bf16" and use_cuda
     model = Net().to(device)
+    if use_bf16:
+        model.bf16_native_()
     if args.resume:
         load_checkpoint(model, args.resume, device)
 
@@ -91,6 +97,9 @@ def run(rank: int, world_size: int, args):
         for data, target in batches:
             if not use_cuda:
                 data, target = data.to(device), target.to(device)
+            if use_bf16:
+                # native engine input layout: NHWC bf16 [N, 28, 28, 1]
+                data = data[:, 0].to(torch.bfloat16).unsqueeze(-1).contiguous()
             optimizer.zero_grad() if not isinstance(model, DistributedDataParallel) \
                 else model.zero_grad()
             _, loss = model(data, target)
