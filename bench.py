@@ -43,6 +43,9 @@ def parse_args():
     p.add_argument("--bucket-mb", type=float, default=25.0)
     p.add_argument("--no-graphs", action="store_true",
                    help="disable hipGraph step capture (eager launches)")
+    p.add_argument("--model", choices=["lenet", "resnet18"], default="lenet",
+                   help="lenet = BASELINE headline config; resnet18 = config 5 "
+                        "(synthetic 3x224x224, bf16 only)")
     return p.parse_args()
 
 
@@ -59,20 +62,31 @@ def main():
     if use_cuda:
         torch.cuda.set_device(device)
 
-    batch = args.batch or (8192 if use_cuda else 256)
     use_bf16 = args.dtype == "bf16" and use_cuda
     torch.manual_seed(1234)
 
-    model = Net().to(device).train()
-    if use_bf16:
-        model.bf16_native_()
+    if args.model == "resnet18":
+        from dist_mnist_amd.models import ResNet18
+
+        assert use_cuda, "resnet18 config runs on the bf16 GPU engine"
+        use_bf16 = True
+        batch = args.batch or 256
+        model = ResNet18().to(device).train()
+        layout = "rgb224_bf16"
+        n_classes = 1000
+    else:
+        batch = args.batch or (8192 if use_cuda else 256)
+        model = Net().to(device).train()
+        if use_bf16:
+            model.bf16_native_()
+        layout = "nhwc_bf16" if use_bf16 else "nchw_f32"
+        n_classes = 10
     if world > 1:
         model = DistributedDataParallel(model, bucket_bytes=int(args.bucket_mb * 2**20))
     opt = SGD(model.parameters(), lr=0.01, momentum=0.5)
 
     stream = SyntheticBatchStream(batch_size=batch, n_pool=4, seed=1234 + rank,
-                                  pin=use_cuda,
-                                  layout="nhwc_bf16" if use_bf16 else "nchw_f32")
+                                  pin=use_cuda, layout=layout, n_classes=n_classes)
     batches = Prefetcher(stream, device,
                          dtype=torch.bfloat16 if use_bf16 else torch.float32)
 
@@ -134,7 +148,9 @@ def main():
 
     if rank == 0:
         result = {
-            "metric": "images/sec (whole node), LeNet-style CNN on MNIST",
+            "metric": ("images/sec (whole node), ResNet-18 synthetic 3x224x224"
+                       if args.model == "resnet18"
+                       else "images/sec (whole node), LeNet-style CNN on MNIST"),
             "value": images_per_sec,
             "unit": "images/s",
             "n_gpus": n_gpus,
@@ -148,7 +164,8 @@ def main():
             "data": "synthetic",
             "final_loss": float(last_loss.item()) if last_loss is not None else None,
             "config": {
-                "model": "lenet5-style-2conv (SURVEY.md R5)",
+                "model": ("resnet18-synthetic-3x224x224" if args.model == "resnet18"
+                          else "lenet5-style-2conv (SURVEY.md R5)"),
                 "global_batch": world * batch,
                 "per_gpu_batch": batch,
                 "seq_len": None,

@@ -118,12 +118,25 @@ class SyntheticBatchStream:
     """
 
     def __init__(self, batch_size: int, n_pool: int = 4, seed: int = 1234,
-                 dtype=torch.float32, pin: bool = False, layout: str = "nchw_f32"):
+                 dtype=torch.float32, pin: bool = False, layout: str = "nchw_f32",
+                 n_classes: int = 10):
         g = torch.Generator().manual_seed(seed)
         self.pool = []
         for _ in range(n_pool):
+            if layout == "rgb224_bf16":
+                # ResNet-18 config-5 input: NHWC bf16 [N,224,224,8], RGB in 0-2
+                xr = torch.randn(batch_size, 224, 224, 3, generator=g,
+                                 dtype=torch.float32)
+                x = torch.zeros(batch_size, 224, 224, 8, dtype=torch.bfloat16)
+                x[..., :3] = xr.to(torch.bfloat16)
+                y = torch.randint(0, n_classes, (batch_size,), generator=g)
+                if pin:
+                    x = x.pin_memory()
+                    y = y.pin_memory()
+                self.pool.append((x, y))
+                continue
             x = torch.randn(batch_size, 1, 28, 28, generator=g, dtype=torch.float32)
-            y = torch.randint(0, 10, (batch_size,), generator=g)
+            y = torch.randint(0, n_classes, (batch_size,), generator=g)
             if layout == "nhwc_bf16":
                 # CIp=1: the fused stem kernel reads scalar channels, so the
                 # input needs NO channel padding — 12.8 MB/batch over PCIe

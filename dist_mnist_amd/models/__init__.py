@@ -1,1 +1,2 @@
 from .lenet import Net  # noqa: F401
+from .resnet import ResNet18  # noqa: F401

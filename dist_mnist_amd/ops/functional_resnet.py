@@ -1,0 +1,87 @@
+"""ResNet-family bf16 NHWC autograd ops (BASELINE.json config 5): BatchNorm2d
+with fp32 statistics, MaxPool 3x3/s2/p1 (deterministic gather backward),
+global average pool, fused residual add+ReLU, standalone ReLU."""
+
+from __future__ import annotations
+
+import torch
+
+from . import native
+
+
+class _BatchNormBF16Fn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, gamma, beta, running_mean, running_var, training,
+                momentum, eps, relu):
+        y, mean, invstd = native.ext().bn_fwd_bf16(
+            x.contiguous(), gamma.contiguous(), beta.contiguous(),
+            running_mean, running_var, bool(training), momentum, eps, relu)
+        ctx.save_for_backward(x, gamma, mean, invstd, y if relu else None)
+        ctx.relu = relu
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, gamma, mean, invstd, yrelu = ctx.saved_tensors
+        dy = dy.contiguous()
+        if ctx.relu:
+            dy = native.ext().relu_gate_bf16(dy, yrelu)
+        dx, dgamma, dbeta = native.ext().bn_bwd_bf16(dy, x, mean, invstd, gamma)
+        return dx, dgamma, dbeta, None, None, None, None, None, None
+
+
+def batchnorm_bf16(x, gamma, beta, running_mean, running_var, training=True,
+                   momentum=0.1, eps=1e-5, relu=False):
+    return _BatchNormBF16Fn.apply(x, gamma, beta, running_mean, running_var,
+                                  training, momentum, eps, relu)
+
+
+class _MaxPool3x3s2BF16Fn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        y, idx = native.ext().maxpool3x3s2_fwd_bf16(x.contiguous())
+        ctx.save_for_backward(idx)
+        ctx.hw = (x.shape[1], x.shape[2])
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (idx,) = ctx.saved_tensors
+        return native.ext().maxpool3x3s2_bwd_bf16(dy.contiguous(), idx, *ctx.hw)
+
+
+def maxpool3x3s2_bf16(x):
+    return _MaxPool3x3s2BF16Fn.apply(x)
+
+
+class _GapBF16Fn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        ctx.hw = (x.shape[1], x.shape[2])
+        return native.ext().gap_fwd_bf16(x.contiguous())
+
+    @staticmethod
+    def backward(ctx, dy):
+        return native.ext().gap_bwd_bf16(dy.contiguous(), *ctx.hw)
+
+
+def global_avgpool_bf16(x):
+    return _GapBF16Fn.apply(x)
+
+
+class _AddReluBF16Fn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, a, b):
+        out = native.ext().add_bf16(a.contiguous(), b.contiguous(), True)
+        ctx.save_for_backward(out)
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        (out,) = ctx.saved_tensors
+        g = native.ext().relu_gate_bf16(dy.contiguous(), out)
+        return g, g
+
+
+def add_relu_bf16(a, b):
+    return _AddReluBF16Fn.apply(a, b)

@@ -8,6 +8,7 @@
 #include "linear.hip"
 #include "elementwise.hip"
 #include "bf16.hip"
+#include "resnet.hip"
 
 namespace {
 
@@ -372,6 +373,7 @@ void fused_sgd_step(at::Tensor ptrs, at::Tensor chunks, long chunk_elems,
 }
 
 #include "ext_bf16.inc"
+#include "ext_resnet.inc"
 
 }  // namespace
 
@@ -407,6 +409,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("maxpool2x2_relu_bf16_bwd", &maxpool2x2_relu_bf16_bwd);
   m.def("dropout_bf16_fwd", &dropout_bf16_fwd);
   m.def("dropout_bf16_bwd", &dropout_bf16_bwd);
+  m.def("bn_fwd_bf16", &bn_fwd_bf16);
+  m.def("bn_bwd_bf16", &bn_bwd_bf16);
+  m.def("maxpool3x3s2_fwd_bf16", &maxpool3x3s2_fwd_bf16);
+  m.def("maxpool3x3s2_bwd_bf16", &maxpool3x3s2_bwd_bf16);
+  m.def("gap_fwd_bf16", &gap_fwd_bf16);
+  m.def("gap_bwd_bf16", &gap_bwd_bf16);
+  m.def("add_bf16", &add_bf16);
+  m.def("relu_gate_bf16", &relu_gate_bf16);
+  m.def("relu_bf16", &relu_bf16);
   m.def("logsoftmax_nll_bf16_fwd", &logsoftmax_nll_bf16_fwd);
   m.def("logsoftmax_nll_bf16_bwd", &logsoftmax_nll_bf16_bwd);
 }

@@ -1,0 +1,361 @@
+// ResNet-family support kernels, NHWC bf16 (BASELINE.json config 5):
+// BatchNorm2d (training stats, fp32 accumulation), MaxPool 3x3/s2/p1 with
+// gather-style deterministic backward, global average pool, fused
+// residual-add+ReLU, standalone ReLU. Channel axis is fastest (NHWC) so every
+// per-channel op vectorizes as bf16x8 with f32x8-style unpacked math.
+#include "common.h"
+// bf16 types come from bf16.hip (this file is included after it in ext.hip)
+
+// ---------------------------------------------------------------------------
+// BatchNorm2d forward, training mode.
+//  pass 1: per-channel sum / sumsq partials (f32), atomicAdd into stats[C][2]
+//  pass 2: mean/invstd + running-stat update (tiny)
+//  pass 3: normalize y = (x - mean) * invstd * gamma + beta  (+opt ReLU)
+// ---------------------------------------------------------------------------
+__global__ void bn_stats_k(const bf16* __restrict__ x, float* __restrict__ stats,
+                           long npix, int Cp) {
+  // one thread per (pixel-chunk, 8-channel group); block-local reduction in
+  // LDS would add little: atomics on [C][2] from <=2048 blocks are cheap.
+  int cg = blockIdx.y;  // channel group of 8
+  float s[8] = {}, ss[8] = {};
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < npix;
+       i += (long)gridDim.x * blockDim.x) {
+    bf16x8 v = *reinterpret_cast<const bf16x8*>(&x[i * Cp + cg * 8]);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = to_f32(v[j]);
+      s[j] += f;
+      ss[j] += f * f;
+    }
+  }
+  // wave reduce then one atomic per channel per wave
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    for (int off = 32; off > 0; off >>= 1) {
+      s[j] += __shfl_down(s[j], off);
+      ss[j] += __shfl_down(ss[j], off);
+    }
+    if ((threadIdx.x & 63) == 0) {
+      atomicAdd(&stats[(cg * 8 + j) * 2 + 0], s[j]);
+      atomicAdd(&stats[(cg * 8 + j) * 2 + 1], ss[j]);
+    }
+  }
+}
+
+__global__ void bn_finalize_k(const float* __restrict__ stats,
+                              float* __restrict__ mean,
+                              float* __restrict__ invstd,
+                              float* __restrict__ running_mean,
+                              float* __restrict__ running_var,
+                              long npix, int C, float eps, float momentum,
+                              int update_running) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float m = stats[c * 2] / (float)npix;
+  float var = stats[c * 2 + 1] / (float)npix - m * m;
+  var = fmaxf(var, 0.f);
+  mean[c] = m;
+  invstd[c] = rsqrtf(var + eps);
+  if (update_running) {
+    float unbiased = var * (float)npix / (float)max(1L, npix - 1);
+    running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * m;
+    running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
+  }
+}
+
+template <bool RELU>
+__global__ void bn_apply_k(const bf16* __restrict__ x, bf16* __restrict__ y,
+                           const float* __restrict__ mean,
+                           const float* __restrict__ invstd,
+                           const float* __restrict__ gamma,
+                           const float* __restrict__ beta, long npix, int Cp,
+                           int C) {
+  int cg = blockIdx.y;
+  float sc[8], sh[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    int c = cg * 8 + j;
+    if (c < C) {
+      sc[j] = gamma[c] * invstd[c];
+      sh[j] = beta[c] - mean[c] * sc[j];
+    } else {
+      sc[j] = 0.f;
+      sh[j] = 0.f;
+    }
+  }
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < npix;
+       i += (long)gridDim.x * blockDim.x) {
+    bf16x8 v = *reinterpret_cast<const bf16x8*>(&x[i * Cp + cg * 8]);
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = to_f32(v[j]) * sc[j] + sh[j];
+      if (RELU) f = fmaxf(f, 0.f);
+      o[j] = to_bf16(f);
+    }
+    *reinterpret_cast<bf16x8*>(&y[i * Cp + cg * 8]) = o;
+  }
+}
+
+// eval mode: same apply with running stats (host passes mean=running_mean,
+// invstd = rsqrt(running_var+eps) precomputed by bn_finalize-eval path)
+
+// ---------------------------------------------------------------------------
+// BatchNorm backward.
+//  reduce: dbeta[c] = sum dy ; dgamma[c] = sum dy * xhat
+//  apply : dx = gamma*invstd * (dy - dbeta/npix - xhat * dgamma/npix)
+// ---------------------------------------------------------------------------
+__global__ void bn_bwd_reduce_k(const bf16* __restrict__ dy,
+                                const bf16* __restrict__ x,
+                                const float* __restrict__ mean,
+                                const float* __restrict__ invstd,
+                                float* __restrict__ red,  // [C][2] zeroed
+                                long npix, int Cp) {
+  int cg = blockIdx.y;
+  float db[8] = {}, dg[8] = {};
+  float mu[8], is[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    mu[j] = mean[cg * 8 + j];
+    is[j] = invstd[cg * 8 + j];
+  }
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < npix;
+       i += (long)gridDim.x * blockDim.x) {
+    bf16x8 g = *reinterpret_cast<const bf16x8*>(&dy[i * Cp + cg * 8]);
+    bf16x8 v = *reinterpret_cast<const bf16x8*>(&x[i * Cp + cg * 8]);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float gf = to_f32(g[j]);
+      db[j] += gf;
+      dg[j] += gf * (to_f32(v[j]) - mu[j]) * is[j];
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    for (int off = 32; off > 0; off >>= 1) {
+      db[j] += __shfl_down(db[j], off);
+      dg[j] += __shfl_down(dg[j], off);
+    }
+    if ((threadIdx.x & 63) == 0) {
+      atomicAdd(&red[(cg * 8 + j) * 2 + 0], db[j]);
+      atomicAdd(&red[(cg * 8 + j) * 2 + 1], dg[j]);
+    }
+  }
+}
+
+__global__ void bn_bwd_apply_k(const bf16* __restrict__ dy,
+                               const bf16* __restrict__ x,
+                               const float* __restrict__ mean,
+                               const float* __restrict__ invstd,
+                               const float* __restrict__ gamma,
+                               const float* __restrict__ red,
+                               bf16* __restrict__ dx, long npix, int Cp, int C) {
+  int cg = blockIdx.y;
+  float mu[8], is[8], gsc[8], mdb[8], mdg[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    int c = cg * 8 + j;
+    if (c < C) {
+      mu[j] = mean[c];
+      is[j] = invstd[c];
+      gsc[j] = gamma[c] * is[j];
+      mdb[j] = red[c * 2] / (float)npix;
+      mdg[j] = red[c * 2 + 1] / (float)npix;
+    } else {
+      mu[j] = is[j] = gsc[j] = mdb[j] = mdg[j] = 0.f;
+    }
+  }
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < npix;
+       i += (long)gridDim.x * blockDim.x) {
+    bf16x8 g = *reinterpret_cast<const bf16x8*>(&dy[i * Cp + cg * 8]);
+    bf16x8 v = *reinterpret_cast<const bf16x8*>(&x[i * Cp + cg * 8]);
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float xhat = (to_f32(v[j]) - mu[j]) * is[j];
+      float val = gsc[j] * (to_f32(g[j]) - mdb[j] - xhat * mdg[j]);
+      o[j] = to_bf16(val);
+    }
+    *reinterpret_cast<bf16x8*>(&dx[i * Cp + cg * 8]) = o;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// MaxPool kernel=3 stride=2 pad=1 (ResNet stem), NHWC bf16, argmax u8 (0..8).
+// Backward is GATHER-style (each input pixel checks the <=4 windows that can
+// contain it) — deterministic, no atomics despite overlapping windows.
+// ---------------------------------------------------------------------------
+__global__ void maxpool3x3s2_fwd_k(const bf16* __restrict__ x,
+                                   bf16* __restrict__ y,
+                                   unsigned char* __restrict__ idx, int N,
+                                   int H, int W, int Cp, int OH, int OW) {
+  long total = (long)N * OH * OW * (Cp / 8);
+  for (long t = (long)blockIdx.x * blockDim.x + threadIdx.x; t < total;
+       t += (long)gridDim.x * blockDim.x) {
+    int cg = t % (Cp / 8);
+    long r = t / (Cp / 8);
+    int ow = r % OW;
+    long r2 = r / OW;
+    int oh = r2 % OH;
+    int n = r2 / OH;
+    int ih0 = oh * 2 - 1, iw0 = ow * 2 - 1;
+    float best[8];
+    unsigned char arg[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) { best[j] = -3.4e38f; arg[j] = 0; }
+    for (int kh = 0; kh < 3; ++kh) {
+      int ih = ih0 + kh;
+      if (ih < 0 || ih >= H) continue;
+      for (int kw = 0; kw < 3; ++kw) {
+        int iw = iw0 + kw;
+        if (iw < 0 || iw >= W) continue;
+        bf16x8 v = *reinterpret_cast<const bf16x8*>(
+            &x[((long)(n * H + ih) * W + iw) * Cp + cg * 8]);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float f = to_f32(v[j]);
+          if (f > best[j]) { best[j] = f; arg[j] = (unsigned char)(kh * 3 + kw); }
+        }
+      }
+    }
+    long off = (((long)(n * OH + oh) * OW + ow) * Cp) + cg * 8;
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) o[j] = to_bf16(best[j]);
+    *reinterpret_cast<bf16x8*>(&y[off]) = o;
+    unsigned char a8[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) a8[j] = arg[j];
+    *reinterpret_cast<unsigned long long*>(&idx[off]) =
+        *reinterpret_cast<unsigned long long*>(a8);
+  }
+}
+
+__global__ void maxpool3x3s2_bwd_k(const bf16* __restrict__ dy,
+                                   const unsigned char* __restrict__ idx,
+                                   bf16* __restrict__ dx, int N, int H, int W,
+                                   int Cp, int OH, int OW) {
+  long total = (long)N * H * W * (Cp / 8);
+  for (long t = (long)blockIdx.x * blockDim.x + threadIdx.x; t < total;
+       t += (long)gridDim.x * blockDim.x) {
+    int cg = t % (Cp / 8);
+    long r = t / (Cp / 8);
+    int iw = r % W;
+    long r2 = r / W;
+    int ih = r2 % H;
+    int n = r2 / H;
+    float acc[8] = {};
+    // windows (oh,ow) with oh*2-1 <= ih <= oh*2+1 => oh in [(ih-1)/2, (ih+1)/2]
+    for (int oh = max(0, (ih - 1) / 2); oh <= min(OH - 1, (ih + 1) / 2); ++oh) {
+      int kh = ih - (oh * 2 - 1);
+      if (kh < 0 || kh > 2) continue;
+      for (int ow = max(0, (iw - 1) / 2); ow <= min(OW - 1, (iw + 1) / 2); ++ow) {
+        int kw = iw - (ow * 2 - 1);
+        if (kw < 0 || kw > 2) continue;
+        long off = (((long)(n * OH + oh) * OW + ow) * Cp) + cg * 8;
+        unsigned long long a8 =
+            *reinterpret_cast<const unsigned long long*>(&idx[off]);
+        bf16x8 g = *reinterpret_cast<const bf16x8*>(&dy[off]);
+        unsigned char want = (unsigned char)(kh * 3 + kw);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          if (((a8 >> (8 * j)) & 0xff) == want) acc[j] += to_f32(g[j]);
+      }
+    }
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) o[j] = to_bf16(acc[j]);
+    *reinterpret_cast<bf16x8*>(&dx[((long)(n * H + ih) * W + iw) * Cp + cg * 8]) = o;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Global average pool [N,H,W,Cp] -> [N,Cp]; bwd broadcasts dy/(H*W).
+// ---------------------------------------------------------------------------
+__global__ void gap_fwd_k(const bf16* __restrict__ x, bf16* __restrict__ y,
+                          int N, long HW, int Cp) {
+  long total = (long)N * (Cp / 8);
+  for (long t = (long)blockIdx.x * blockDim.x + threadIdx.x; t < total;
+       t += (long)gridDim.x * blockDim.x) {
+    int cg = t % (Cp / 8);
+    int n = t / (Cp / 8);
+    float acc[8] = {};
+    const bf16* base = x + (long)n * HW * Cp + cg * 8;
+    for (long i = 0; i < HW; ++i) {
+      bf16x8 v = *reinterpret_cast<const bf16x8*>(base + i * Cp);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) acc[j] += to_f32(v[j]);
+    }
+    bf16x8 o;
+    float inv = 1.f / (float)HW;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) o[j] = to_bf16(acc[j] * inv);
+    *reinterpret_cast<bf16x8*>(&y[(long)n * Cp + cg * 8]) = o;
+  }
+}
+
+__global__ void gap_bwd_k(const bf16* __restrict__ dy, bf16* __restrict__ dx,
+                          int N, long HW, int Cp) {
+  long total = (long)N * HW * (Cp / 8);
+  float inv = 1.f / (float)HW;
+  for (long t = (long)blockIdx.x * blockDim.x + threadIdx.x; t < total;
+       t += (long)gridDim.x * blockDim.x) {
+    int cg = t % (Cp / 8);
+    long r = t / (Cp / 8);
+    long n = r / HW;
+    bf16x8 g = *reinterpret_cast<const bf16x8*>(&dy[n * Cp + cg * 8]);
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) o[j] = to_bf16(to_f32(g[j]) * inv);
+    *reinterpret_cast<bf16x8*>(&dx[t * 8]) = o;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Residual add (+ReLU) and standalone ReLU, bf16 vec8.
+// add_relu: out = relu(a + b); bwd: da = db = dy * (out > 0).
+// ---------------------------------------------------------------------------
+template <bool RELU>
+__global__ void add_bf16_k(const bf16* __restrict__ a, const bf16* __restrict__ b,
+                           bf16* __restrict__ out, long n8) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n8;
+       i += (long)gridDim.x * blockDim.x) {
+    bf16x8 va = *reinterpret_cast<const bf16x8*>(a + i * 8);
+    bf16x8 vb = *reinterpret_cast<const bf16x8*>(b + i * 8);
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = to_f32(va[j]) + to_f32(vb[j]);
+      if (RELU) f = fmaxf(f, 0.f);
+      o[j] = to_bf16(f);
+    }
+    *reinterpret_cast<bf16x8*>(out + i * 8) = o;
+  }
+}
+
+__global__ void relu_gate_bf16_k(const bf16* __restrict__ dy,
+                                 const bf16* __restrict__ out,
+                                 bf16* __restrict__ dx, long n8) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n8;
+       i += (long)gridDim.x * blockDim.x) {
+    bf16x8 g = *reinterpret_cast<const bf16x8*>(dy + i * 8);
+    bf16x8 v = *reinterpret_cast<const bf16x8*>(out + i * 8);
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      o[j] = (to_f32(v[j]) > 0.f) ? g[j] : to_bf16(0.f);
+    *reinterpret_cast<bf16x8*>(dx + i * 8) = o;
+  }
+}
+
+__global__ void relu_bf16_k(const bf16* __restrict__ x, bf16* __restrict__ y,
+                            long n8) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n8;
+       i += (long)gridDim.x * blockDim.x) {
+    bf16x8 v = *reinterpret_cast<const bf16x8*>(x + i * 8);
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) o[j] = to_bf16(fmaxf(to_f32(v[j]), 0.f));
+    *reinterpret_cast<bf16x8*>(y + i * 8) = o;
+  }
+}
