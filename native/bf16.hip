@@ -65,7 +65,7 @@ __global__ void mfma_probe_16x16x32_k(const float* __restrict__ A,
 // Bias fp32; optional fused ReLU (ResNet); co >= CO stores 0 (keeps channel
 // padding clean for the next layer's A loads).
 // ---------------------------------------------------------------------------
-template <bool HAS_BIAS, bool RELU, bool BOUNDS, int MW>
+template <bool HAS_BIAS, bool RELU, bool BOUNDS, int MW, int NW>
 __global__ __launch_bounds__(256) void conv2d_fwd_bf16_k(
     const bf16* __restrict__ x, const float* __restrict__ w,
     const float* __restrict__ bias, bf16* __restrict__ y,
@@ -74,23 +74,22 @@ __global__ __launch_bounds__(256) void conv2d_fwd_bf16_k(
   constexpr int BK = 32;
   const int BM = 4 * MW * 16;
   const int m0 = blockIdx.x * BM;
-  const int co0 = blockIdx.y * 16;
+  const int co0 = blockIdx.y * (NW * 16);
   const int tid = threadIdx.x;
   const int wid = tid >> 6;
   const int l = tid & 63;
   const int lm = l & 15;
   const int lk = l >> 4;
 
-  // lds_w[co][k-chunk]: row stride 40 bf16 (80B) => conflict-free b128 reads
-  __shared__ bf16 lds_w[16][BK + 8];
-  __shared__ int lds_koff[BK / 8];   // element offset of each 8-k group
-  __shared__ int lds_khw[BK / 8];    // (kh<<16)|kw of each group (uniform in group)
+  // lds_w[co][k-chunk]: row pad 8 => conflict-light b128 reads
+  __shared__ bf16 lds_w[NW * 16][BK + 8];
+  __shared__ int lds_koff[BK / 8];
+  __shared__ int lds_khw[BK / 8];
 
-  const int KWC = KW * CIp;          // k-row length (multiple of 8)
+  const int KWC = KW * CIp;
   const int KHKW = KH * KW;
 
-  // per-(lane, mi) A-side pixel decode (m = ... + lm), fixed across K loop
-  int xbase[MW];   // element offset of x[n][ih0][iw0][0]
+  int xbase[MW];
   int ih0[MW], iw0[MW];
   bool mval[MW];
 #pragma unroll
@@ -105,13 +104,12 @@ __global__ __launch_bounds__(256) void conv2d_fwd_bf16_k(
     xbase[mi] = ((n * H + ih0[mi]) * W + iw0[mi]) * CIp;
   }
 
-  f32x4 acc[MW] = {};
+  f32x4 acc[MW][NW] = {};
 
   for (int k0 = 0; k0 < K; k0 += BK) {
     __syncthreads();
-    // stage weights: lds_w[co][kk] = bf16(w[co0+co][ci][kh][kw]) for k=k0+kk
-    for (int idx = tid; idx < 16 * BK; idx += 256) {
-      int co = idx >> 5;       // 0..15 (idx = co*32 + kk)
+    for (int idx = tid; idx < NW * 16 * BK; idx += 256) {
+      int co = idx >> 5;
       int kk = idx & 31;
       int k = k0 + kk;
       float v = 0.f;
@@ -133,16 +131,18 @@ __global__ __launch_bounds__(256) void conv2d_fwd_bf16_k(
     }
     __syncthreads();
 
-    int kgroups = min(BK, K - k0) / 8;  // K is always a multiple of 8
+    int kgroups = min(BK, K - k0) / 8;
     for (int kg = 0; kg < kgroups; kg += 4) {
       bool gin = lk < (kgroups - kg);
       int gsel = gin ? kg + lk : 0;
-      // B-frag: 8 contiguous k of one co row from LDS (16B read)
-      bf16x8 b = gin ? *reinterpret_cast<const bf16x8*>(&lds_w[lm][gsel * 8])
-                     : bf8_zero();
       int koff = lds_koff[gsel];
       int khw = lds_khw[gsel];
       int kh = khw >> 16, kw = khw & 0xffff;
+      bf16x8 bw[NW];
+#pragma unroll
+      for (int ni = 0; ni < NW; ++ni)
+        bw[ni] = gin ? *reinterpret_cast<const bf16x8*>(&lds_w[ni * 16 + lm][gsel * 8])
+                     : bf8_zero();
 #pragma unroll
       for (int mi = 0; mi < MW; ++mi) {
         bool ok = mval[mi] & gin;
@@ -152,27 +152,30 @@ __global__ __launch_bounds__(256) void conv2d_fwd_bf16_k(
         }
         bf16x8 a = ok ? *reinterpret_cast<const bf16x8*>(&x[xbase[mi] + koff])
                       : bf8_zero();
-        // D[row=m][col=co]: patch supplies rows, weights supply columns.
-        acc[mi] = mfma16x16x32bf(a, b, acc[mi]);
+#pragma unroll
+        for (int ni = 0; ni < NW; ++ni)
+          acc[mi][ni] = mfma16x16x32bf(a, bw[ni], acc[mi][ni]);
       }
     }
   }
 
-  // epilogue: D row = lk*4+reg -> m, col = lm -> co. A 16-lane group writes
-  // 16 consecutive co of one pixel => contiguous 32B NHWC stores.
 #pragma unroll
   for (int mi = 0; mi < MW; ++mi) {
 #pragma unroll
     for (int reg = 0; reg < 4; ++reg) {
       int m = m0 + (wid * MW + mi) * 16 + lk * 4 + reg;
       if (m >= M) continue;
-      int co = co0 + lm;
-      if (co >= COp) continue;
       int n = m / (OH * OW), r = m % (OH * OW);
-      float v = (co < CO) ? acc[mi][reg] : 0.f;
-      if (HAS_BIAS && co < CO) v += bias[co];
-      if (RELU) v = fmaxf(v, 0.f);
-      y[(long)(n * (OH * OW) + r) * COp + co] = to_bf16(v);
+      long yb = (long)(n * (OH * OW) + r) * COp;
+#pragma unroll
+      for (int ni = 0; ni < NW; ++ni) {
+        int co = co0 + ni * 16 + lm;
+        if (co >= COp) continue;
+        float v = (co < CO) ? acc[mi][ni][reg] : 0.f;
+        if (HAS_BIAS && co < CO) v += bias[co];
+        if (RELU) v = fmaxf(v, 0.f);
+        y[yb + co] = to_bf16(v);
+      }
     }
   }
 }
@@ -184,7 +187,7 @@ __global__ __launch_bounds__(256) void conv2d_fwd_bf16_k(
 //   B = weights staged fp32->bf16 into LDS as [ci][k'] rows.
 //   oh = (ih + pad - kh)/stride when divisible & in range, else 0.
 // ---------------------------------------------------------------------------
-template <bool S1, int MW>
+template <bool S1, int MW, int NW>
 __global__ __launch_bounds__(256) void conv2d_dgrad_bf16_k(
     const bf16* __restrict__ dy, const float* __restrict__ w,
     bf16* __restrict__ dx,
@@ -193,15 +196,15 @@ __global__ __launch_bounds__(256) void conv2d_dgrad_bf16_k(
   constexpr int BK = 32;
   const int BM = 4 * MW * 16;
   const int m0 = blockIdx.x * BM;
-  const int ci0 = blockIdx.y * 16;
+  const int ci0 = blockIdx.y * (NW * 16);
   const int tid = threadIdx.x;
   const int wid = tid >> 6;
   const int l = tid & 63;
   const int lm = l & 15;
   const int lk = l >> 4;
 
-  __shared__ bf16 lds_w[16][BK + 8];   // [ci][k'] rows
-  __shared__ int lds_kdec[BK / 8];     // (kh<<20)|(kw<<8)|co0_of_group... packed
+  __shared__ bf16 lds_w[NW * 16][BK + 8];
+  __shared__ int lds_kdec[BK / 8];
 
   const int KWC = KW * COp;
   const int KHKW = KH * KW;
@@ -219,11 +222,11 @@ __global__ __launch_bounds__(256) void conv2d_dgrad_bf16_k(
     iw[mi] = r % W;
   }
 
-  f32x4 acc[MW] = {};
+  f32x4 acc[MW][NW] = {};
 
   for (int k0 = 0; k0 < Kp; k0 += BK) {
     __syncthreads();
-    for (int idx = tid; idx < 16 * BK; idx += 256) {
+    for (int idx = tid; idx < NW * 16 * BK; idx += 256) {
       int ci = idx >> 5;
       int kk = idx & 31;
       int k = k0 + kk;
@@ -249,10 +252,13 @@ __global__ __launch_bounds__(256) void conv2d_dgrad_bf16_k(
     for (int kg = 0; kg < kgroups; kg += 4) {
       bool gin = lk < (kgroups - kg);
       int gsel = gin ? kg + lk : 0;
-      bf16x8 b = gin ? *reinterpret_cast<const bf16x8*>(&lds_w[lm][gsel * 8])
-                     : bf8_zero();
       int kdec = lds_kdec[gsel];
       int kh = kdec >> 20, kw = (kdec >> 10) & 0x3ff, co0g = kdec & 0x3ff;
+      bf16x8 bw[NW];
+#pragma unroll
+      for (int ni = 0; ni < NW; ++ni)
+        bw[ni] = gin ? *reinterpret_cast<const bf16x8*>(&lds_w[ni * 16 + lm][gsel * 8])
+                     : bf8_zero();
 #pragma unroll
       for (int mi = 0; mi < MW; ++mi) {
         bf16x8 a = bf8_zero();
@@ -272,7 +278,9 @@ __global__ __launch_bounds__(256) void conv2d_dgrad_bf16_k(
             }
           }
         }
-        acc[mi] = mfma16x16x32bf(a, b, acc[mi]);
+#pragma unroll
+        for (int ni = 0; ni < NW; ++ni)
+          acc[mi][ni] = mfma16x16x32bf(a, bw[ni], acc[mi][ni]);
       }
     }
   }
@@ -283,9 +291,13 @@ __global__ __launch_bounds__(256) void conv2d_dgrad_bf16_k(
     for (int reg = 0; reg < 4; ++reg) {
       int m = m0 + (wid * MW + mi) * 16 + lk * 4 + reg;
       if (m >= Mp) continue;
-      int ci = ci0 + lm;
-      if (ci >= CIp) continue;
-      dx[(long)m * CIp + ci] = to_bf16(ci < CI ? acc[mi][reg] : 0.f);
+      long xb = (long)m * CIp;
+#pragma unroll
+      for (int ni = 0; ni < NW; ++ni) {
+        int ci = ci0 + ni * 16 + lm;
+        if (ci >= CIp) continue;
+        dx[xb + ci] = to_bf16(ci < CI ? acc[mi][ni][reg] : 0.f);
+      }
     }
   }
 }
@@ -320,7 +332,7 @@ DEV bf16x8 tr_read_8m(const bf16* lds_img, int l) {
   return a;
 }
 
-template <bool BOUNDS, bool WANT_BIAS, int KT>
+template <bool BOUNDS, bool WANT_BIAS, int KT, int NW>
 __global__ __launch_bounds__(256) void conv2d_wgrad_bf16_k(
     const bf16* __restrict__ dy, const bf16* __restrict__ x,
     float* __restrict__ dw_ws, float* __restrict__ db_ws,
@@ -330,7 +342,7 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_bf16_k(
   constexpr int MC = 64;          // m per chunk (shared by all 4 waves)
   constexpr int NKF = 4 * KT;     // 16-row k-frag images per block
   const int k0 = blockIdx.x * (NKF * 16);
-  const int co0 = blockIdx.y * 16;
+  const int co0 = blockIdx.y * (NW * 16);
   const int ms = blockIdx.z;
   const int tid = threadIdx.x;
   const int wid = tid >> 6;
@@ -338,17 +350,16 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_bf16_k(
   const int lm = l & 15;
   const int lk = l >> 4;
 
-  __shared__ bf16 lds_x[NKF][MC][16];   // [kf][m][16k] sub-images
-  __shared__ bf16 lds_dy[MC][16];       // [m][16co]
-  __shared__ int lds_mbase[MC];         // x elem offset of (n, ih0, iw0)
-  __shared__ int lds_mhw[MC];           // biased (ih0, iw0) for bounds
+  __shared__ bf16 lds_x[NKF][MC][16];       // [kf][m][16k] sub-images
+  __shared__ bf16 lds_dy[NW][MC][16];       // [ni][m][16co]
+  __shared__ int lds_mbase[MC];
+  __shared__ int lds_mhw[MC];
 
   const int KWC = KW * CIp;
   const int OHOW = OH * OW;
 
-  // per-lane k decode for the NKF*2 8-k groups, fixed over the reduction
-  int g_off[NKF * 2];   // (kh*W + kw)*CIp + ci0
-  int g_khw[NKF * 2];   // (kh<<16)|kw
+  int g_off[NKF * 2];
+  int g_khw[NKF * 2];
 #pragma unroll
   for (int g = 0; g < NKF * 2; ++g) {
     int k = k0 + g * 8;
@@ -359,15 +370,14 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_bf16_k(
     g_khw[g] = (kh << 16) | kw;
   }
 
-  f32x4 acc[KT] = {};
-  float db_part = 0.f;
+  f32x4 acc[KT][NW] = {};
+  float db_part[NW] = {};
 
   int m_begin = ms * m_per_split;
   int m_end = min(M, m_begin + m_per_split);
 
   for (int mc = m_begin; mc < m_end; mc += MC) {
     __syncthreads();
-    // ---- per-chunk m decode, once --------------------------------------
     if (tid < MC) {
       int m = mc + tid;
       int mm = m < m_end ? m : 0;
@@ -378,7 +388,6 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_bf16_k(
       lds_mhw[tid] = ((ih0 + 16384) << 16) | (iw0 + 16384);
     }
     __syncthreads();
-    // ---- stage x sub-images: entries (kf, m, 8k-half) -------------------
     for (int idx = tid; idx < NKF * MC * 2; idx += 256) {
       int half = idx & 1;
       int rest = idx >> 1;
@@ -393,44 +402,50 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_bf16_k(
         if (BOUNDS) {
           int hw = lds_mhw[mloc];
           int khw = g_khw[g];
-          int ih = (hw >> 16) - 16384 + (khw >> 16);
-          int iw = (hw & 0xffff) - 16384 + (khw & 0xffff);
-          ok = ih >= 0 && ih < H && iw >= 0 && iw < W;
+          int ihh = (hw >> 16) - 16384 + (khw >> 16);
+          int iww = (hw & 0xffff) - 16384 + (khw & 0xffff);
+          ok = ihh >= 0 && ihh < H && iww >= 0 && iww < W;
         }
         if (ok)
           v = *reinterpret_cast<const bf16x8*>(&x[lds_mbase[mloc] + g_off[g]]);
       }
       *reinterpret_cast<bf16x8*>(&lds_x[kf][mloc][half * 8]) = v;
     }
-    // ---- stage dy chunk --------------------------------------------------
-    for (int idx = tid; idx < MC * 2; idx += 256) {
+    for (int idx = tid; idx < NW * MC * 2; idx += 256) {
       int half = idx & 1;
-      int mloc = idx >> 1;
+      int rest = idx >> 1;
+      int mloc = rest % MC;
+      int ni = rest / MC;
       int m = mc + mloc;
       bf16x8 v = bf8_zero();
-      if (m < m_end && co0 + half * 8 < COp)
-        v = *reinterpret_cast<const bf16x8*>(&dy[(long)m * COp + co0 + half * 8]);
-      *reinterpret_cast<bf16x8*>(&lds_dy[mloc][half * 8]) = v;
+      int cb = co0 + ni * 16 + half * 8;
+      if (m < m_end && cb < COp)
+        v = *reinterpret_cast<const bf16x8*>(&dy[(long)m * COp + cb]);
+      *reinterpret_cast<bf16x8*>(&lds_dy[ni][mloc][half * 8]) = v;
     }
     __syncthreads();
 
-    // ---- MFMA: all waves share the dy frags, each owns KT x-images ------
 #pragma unroll
     for (int mm = 0; mm < MC; mm += 32) {
-      bf16x8 b = tr_read_8m(&lds_dy[mm][0], l);
-      if (WANT_BIAS && wid == 0) {
+      bf16x8 b[NW];
 #pragma unroll
-        for (int j = 0; j < 8; ++j) db_part += to_f32(b[j]);
+      for (int ni = 0; ni < NW; ++ni) {
+        b[ni] = tr_read_8m(&lds_dy[ni][mm][0], l);
+        if (WANT_BIAS && wid == 0) {
+#pragma unroll
+          for (int j = 0; j < 8; ++j) db_part[ni] += to_f32(b[ni][j]);
+        }
       }
 #pragma unroll
       for (int kf = 0; kf < KT; ++kf) {
         bf16x8 a = tr_read_8m(&lds_x[wid * KT + kf][mm][0], l);
-        acc[kf] = mfma16x16x32bf(a, b, acc[kf]);
+#pragma unroll
+        for (int ni = 0; ni < NW; ++ni)
+          acc[kf][ni] = mfma16x16x32bf(a, b[ni], acc[kf][ni]);
       }
     }
   }
 
-  // ---- epilogue: waves own disjoint k rows; write partial slab ----------
   {
     float* slab = dw_ws + (long)ms * (CO * (long)Kcan);
 #pragma unroll
@@ -438,20 +453,28 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_bf16_k(
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg) {
         int kk = k0 + (wid * KT + kf) * 16 + lk * 4 + reg;
-        int co = co0 + lm;
-        if (kk < K && co < CO) {
-          int kh = kk / KWC, r = kk % KWC;
-          int kw = r / CIp, ci = r % CIp;
-          if (ci < CI)
-            slab[((co * CI) + ci) * (KH * KW) + kh * KW + kw] = acc[kf][reg];
+        if (kk >= K) continue;
+        int kh = kk / KWC, r = kk % KWC;
+        int kw = r / CIp, ci = r % CIp;
+        if (ci >= CI) continue;
+        long koffc = ((long)ci) * (KH * KW) + kh * KW + kw;
+#pragma unroll
+        for (int ni = 0; ni < NW; ++ni) {
+          int co = co0 + ni * 16 + lm;
+          if (co < CO) slab[(long)co * Kcan + koffc] = acc[kf][ni][reg];
         }
       }
     }
   }
   if (WANT_BIAS && blockIdx.x == 0 && wid == 0) {
-    db_part += __shfl_down(db_part, 16);
-    db_part += __shfl_down(db_part, 32);
-    if (lk == 0 && co0 + lm < CO) db_ws[(long)ms * CO + co0 + lm] = db_part;
+#pragma unroll
+    for (int ni = 0; ni < NW; ++ni) {
+      float v = db_part[ni];
+      v += __shfl_down(v, 16);
+      v += __shfl_down(v, 32);
+      int co = co0 + ni * 16 + lm;
+      if (lk == 0 && co < CO) db_ws[(long)ms * CO + co] = v;
+    }
   }
 }
 

@@ -14,28 +14,45 @@
 // ---------------------------------------------------------------------------
 __global__ void bn_stats_k(const bf16* __restrict__ x, float* __restrict__ stats,
                            long npix, int Cp) {
-  // one thread per (pixel-chunk, 8-channel group); block-local reduction in
-  // LDS would add little: atomics on [C][2] from <=2048 blocks are cheap.
-  int cg = blockIdx.y;  // channel group of 8
+  // thread t owns channel group cg = t % ncg and marches pixels with its
+  // pixel-lane: consecutive threads cover a FULL contiguous pixel row
+  // (coalesced), unlike a per-channel-group grid (16B per 128B line).
+  int ncg = Cp / 8;                  // <= 256 (Cp <= 2048)
+  int cg = threadIdx.x % ncg;
+  int pl = threadIdx.x / ncg;
+  int lanes = blockDim.x / ncg;      // pixel-lanes per block
+  __shared__ float red[256][8];
   float s[8] = {}, ss[8] = {};
-  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < npix;
-       i += (long)gridDim.x * blockDim.x) {
-    bf16x8 v = *reinterpret_cast<const bf16x8*>(&x[i * Cp + cg * 8]);
+  if (pl < lanes) {
+    for (long i = (long)blockIdx.x * lanes + pl; i < npix;
+         i += (long)gridDim.x * lanes) {
+      bf16x8 v = *reinterpret_cast<const bf16x8*>(&x[i * Cp + cg * 8]);
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      float f = to_f32(v[j]);
-      s[j] += f;
-      ss[j] += f * f;
+      for (int j = 0; j < 8; ++j) {
+        float f = to_f32(v[j]);
+        s[j] += f;
+        ss[j] += f * f;
+      }
     }
   }
-  // wave reduce then one atomic per channel per wave
 #pragma unroll
-  for (int j = 0; j < 8; ++j) {
-    for (int off = 32; off > 0; off >>= 1) {
-      s[j] += __shfl_down(s[j], off);
-      ss[j] += __shfl_down(ss[j], off);
-    }
-    if ((threadIdx.x & 63) == 0) {
+  for (int j = 0; j < 8; ++j) red[threadIdx.x][j] = s[j];
+  __syncthreads();
+  if (pl == 0) {
+    for (int l = 1; l < lanes; ++l)
+#pragma unroll
+      for (int j = 0; j < 8; ++j) s[j] += red[cg + l * ncg][j];
+  }
+  __syncthreads();
+#pragma unroll
+  for (int j = 0; j < 8; ++j) red[threadIdx.x][j] = ss[j];
+  __syncthreads();
+  if (pl == 0) {
+    for (int l = 1; l < lanes; ++l)
+#pragma unroll
+      for (int j = 0; j < 8; ++j) ss[j] += red[cg + l * ncg][j];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
       atomicAdd(&stats[(cg * 8 + j) * 2 + 0], s[j]);
       atomicAdd(&stats[(cg * 8 + j) * 2 + 1], ss[j]);
     }
@@ -70,7 +87,11 @@ __global__ void bn_apply_k(const bf16* __restrict__ x, bf16* __restrict__ y,
                            const float* __restrict__ gamma,
                            const float* __restrict__ beta, long npix, int Cp,
                            int C) {
-  int cg = blockIdx.y;
+  int ncg = Cp / 8;
+  int cg = threadIdx.x % ncg;
+  int pl = threadIdx.x / ncg;
+  int lanes = blockDim.x / ncg;
+  if (pl >= lanes) return;
   float sc[8], sh[8];
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
@@ -83,8 +104,8 @@ __global__ void bn_apply_k(const bf16* __restrict__ x, bf16* __restrict__ y,
       sh[j] = 0.f;
     }
   }
-  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < npix;
-       i += (long)gridDim.x * blockDim.x) {
+  for (long i = (long)blockIdx.x * lanes + pl; i < npix;
+       i += (long)gridDim.x * lanes) {
     bf16x8 v = *reinterpret_cast<const bf16x8*>(&x[i * Cp + cg * 8]);
     bf16x8 o;
 #pragma unroll
@@ -109,9 +130,13 @@ __global__ void bn_bwd_reduce_k(const bf16* __restrict__ dy,
                                 const bf16* __restrict__ x,
                                 const float* __restrict__ mean,
                                 const float* __restrict__ invstd,
-                                float* __restrict__ red,  // [C][2] zeroed
+                                float* __restrict__ red_out,  // [C][2] zeroed
                                 long npix, int Cp) {
-  int cg = blockIdx.y;
+  int ncg = Cp / 8;
+  int cg = threadIdx.x % ncg;
+  int pl = threadIdx.x / ncg;
+  int lanes = blockDim.x / ncg;
+  __shared__ float red[256][8];
   float db[8] = {}, dg[8] = {};
   float mu[8], is[8];
 #pragma unroll
@@ -119,26 +144,38 @@ __global__ void bn_bwd_reduce_k(const bf16* __restrict__ dy,
     mu[j] = mean[cg * 8 + j];
     is[j] = invstd[cg * 8 + j];
   }
-  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < npix;
-       i += (long)gridDim.x * blockDim.x) {
-    bf16x8 g = *reinterpret_cast<const bf16x8*>(&dy[i * Cp + cg * 8]);
-    bf16x8 v = *reinterpret_cast<const bf16x8*>(&x[i * Cp + cg * 8]);
+  if (pl < lanes) {
+    for (long i = (long)blockIdx.x * lanes + pl; i < npix;
+         i += (long)gridDim.x * lanes) {
+      bf16x8 g = *reinterpret_cast<const bf16x8*>(&dy[i * Cp + cg * 8]);
+      bf16x8 v = *reinterpret_cast<const bf16x8*>(&x[i * Cp + cg * 8]);
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      float gf = to_f32(g[j]);
-      db[j] += gf;
-      dg[j] += gf * (to_f32(v[j]) - mu[j]) * is[j];
+      for (int j = 0; j < 8; ++j) {
+        float gf = to_f32(g[j]);
+        db[j] += gf;
+        dg[j] += gf * (to_f32(v[j]) - mu[j]) * is[j];
+      }
     }
   }
 #pragma unroll
-  for (int j = 0; j < 8; ++j) {
-    for (int off = 32; off > 0; off >>= 1) {
-      db[j] += __shfl_down(db[j], off);
-      dg[j] += __shfl_down(dg[j], off);
-    }
-    if ((threadIdx.x & 63) == 0) {
-      atomicAdd(&red[(cg * 8 + j) * 2 + 0], db[j]);
-      atomicAdd(&red[(cg * 8 + j) * 2 + 1], dg[j]);
+  for (int j = 0; j < 8; ++j) red[threadIdx.x][j] = db[j];
+  __syncthreads();
+  if (pl == 0)
+    for (int l = 1; l < lanes; ++l)
+#pragma unroll
+      for (int j = 0; j < 8; ++j) db[j] += red[cg + l * ncg][j];
+  __syncthreads();
+#pragma unroll
+  for (int j = 0; j < 8; ++j) red[threadIdx.x][j] = dg[j];
+  __syncthreads();
+  if (pl == 0) {
+    for (int l = 1; l < lanes; ++l)
+#pragma unroll
+      for (int j = 0; j < 8; ++j) dg[j] += red[cg + l * ncg][j];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      atomicAdd(&red_out[(cg * 8 + j) * 2 + 0], db[j]);
+      atomicAdd(&red_out[(cg * 8 + j) * 2 + 1], dg[j]);
     }
   }
 }
@@ -150,7 +187,11 @@ __global__ void bn_bwd_apply_k(const bf16* __restrict__ dy,
                                const float* __restrict__ gamma,
                                const float* __restrict__ red,
                                bf16* __restrict__ dx, long npix, int Cp, int C) {
-  int cg = blockIdx.y;
+  int ncg = Cp / 8;
+  int cg = threadIdx.x % ncg;
+  int pl = threadIdx.x / ncg;
+  int lanes = blockDim.x / ncg;
+  if (pl >= lanes) return;
   float mu[8], is[8], gsc[8], mdb[8], mdg[8];
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
@@ -165,8 +206,8 @@ __global__ void bn_bwd_apply_k(const bf16* __restrict__ dy,
       mu[j] = is[j] = gsc[j] = mdb[j] = mdg[j] = 0.f;
     }
   }
-  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < npix;
-       i += (long)gridDim.x * blockDim.x) {
+  for (long i = (long)blockIdx.x * lanes + pl; i < npix;
+       i += (long)gridDim.x * lanes) {
     bf16x8 g = *reinterpret_cast<const bf16x8*>(&dy[i * Cp + cg * 8]);
     bf16x8 v = *reinterpret_cast<const bf16x8*>(&x[i * Cp + cg * 8]);
     bf16x8 o;
