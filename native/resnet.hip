@@ -51,10 +51,13 @@ __global__ void bn_stats_k(const bf16* __restrict__ x, float* __restrict__ stats
     for (int l = 1; l < lanes; ++l)
 #pragma unroll
       for (int j = 0; j < 8; ++j) ss[j] += red[cg + l * ncg][j];
+    // per-block partial slab (plain stores; deterministic split-reduce
+    // folds the block axis — atomics on [C][2] serialize on 1-2 lines)
+    float* slab = stats + (long)blockIdx.x * (2 * 8 * ncg);
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      atomicAdd(&stats[(cg * 8 + j) * 2 + 0], s[j]);
-      atomicAdd(&stats[(cg * 8 + j) * 2 + 1], ss[j]);
+      slab[(cg * 8 + j) * 2 + 0] = s[j];
+      slab[(cg * 8 + j) * 2 + 1] = ss[j];
     }
   }
 }
@@ -172,10 +175,11 @@ __global__ void bn_bwd_reduce_k(const bf16* __restrict__ dy,
     for (int l = 1; l < lanes; ++l)
 #pragma unroll
       for (int j = 0; j < 8; ++j) dg[j] += red[cg + l * ncg][j];
+    float* slab = red_out + (long)blockIdx.x * (2 * 8 * ncg);
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      atomicAdd(&red_out[(cg * 8 + j) * 2 + 0], db[j]);
-      atomicAdd(&red_out[(cg * 8 + j) * 2 + 1], dg[j]);
+      slab[(cg * 8 + j) * 2 + 0] = db[j];
+      slab[(cg * 8 + j) * 2 + 1] = dg[j];
     }
   }
 }
