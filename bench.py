@@ -43,6 +43,9 @@ def parse_args():
     p.add_argument("--bucket-mb", type=float, default=25.0)
     p.add_argument("--no-graphs", action="store_true",
                    help="disable hipGraph step capture (eager launches)")
+    p.add_argument("--comm", choices=["torch", "native"], default="torch",
+                   help="gradient collective backend for world>1: torch "
+                        "ProcessGroupNCCL, or the framework's C++ RCCL core")
     p.add_argument("--model", choices=["lenet", "resnet18"], default="lenet",
                    help="lenet = BASELINE headline config; resnet18 = config 5 "
                         "(synthetic 3x224x224, bf16 only)")
@@ -82,7 +85,13 @@ def main():
         layout = "nhwc_bf16" if use_bf16 else "nchw_f32"
         n_classes = 10
     if world > 1:
-        model = DistributedDataParallel(model, bucket_bytes=int(args.bucket_mb * 2**20))
+        native_comm = None
+        if args.comm == "native" and use_cuda:
+            from dist_mnist_amd.parallel import RcclComm
+
+            native_comm = RcclComm.bootstrap()
+        model = DistributedDataParallel(model, bucket_bytes=int(args.bucket_mb * 2**20),
+                                        native_comm=native_comm)
     opt = SGD(model.parameters(), lr=0.01, momentum=0.5)
 
     stream = SyntheticBatchStream(batch_size=batch, n_pool=4, seed=1234 + rank,

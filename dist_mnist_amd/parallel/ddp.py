@@ -61,9 +61,13 @@ class DistributedDataParallel(nn.Module):
 
     def __init__(self, module: nn.Module, bucket_bytes: int = DEFAULT_BUCKET_BYTES,
                  process_group=None, broadcast_params: bool = True,
-                 gradient_as_bucket_view: bool = True):
+                 gradient_as_bucket_view: bool = True, native_comm=None):
+        """native_comm: a parallel.rccl_comm.RcclComm — collectives then go
+        through the framework's own C++ RCCL core (stream-ordered, no Work
+        objects) instead of torch.distributed."""
         super().__init__()
         self.module = module
+        self.native_comm = native_comm
         self.pg = process_group
         self.world_size = dist.get_world_size(self.pg) if dist.is_initialized() else 1
         self.enabled = self.world_size > 1
@@ -119,9 +123,13 @@ class DistributedDataParallel(nn.Module):
         def hook(param):
             bucket.pending -= 1
             if bucket.pending == 0:
-                bucket.work = dist.all_reduce(
-                    bucket.flat, op=dist.ReduceOp.SUM, group=self.pg, async_op=True
-                )
+                if self.native_comm is not None:
+                    # stream-ordered ncclAllReduce via the C++ comm core
+                    self.native_comm.all_reduce_sum(bucket.flat)
+                else:
+                    bucket.work = dist.all_reduce(
+                        bucket.flat, op=dist.ReduceOp.SUM, group=self.pg,
+                        async_op=True)
         return hook
 
     # -- public API ---------------------------------------------------------
@@ -152,6 +160,12 @@ class DistributedDataParallel(nn.Module):
             return
         inv = 1.0 / self.world_size
         for b in self.buckets:
+            if self.native_comm is not None:
+                if b.pending != 0:
+                    self.native_comm.all_reduce_sum(b.flat)
+                b.flat.mul_(inv)
+                b.pending = len(b.params)
+                continue
             if b.work is None and b.pending != 0:
                 # grads for some params never materialized (e.g. frozen path):
                 # launch now so ranks do not deadlock.

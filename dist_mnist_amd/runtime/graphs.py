@@ -95,6 +95,9 @@ class GraphedStep:
             # one eager all-reduce per bucket (grads are bucket views filled
             # by the captured backward)
             for b in self.ddp.buckets:
-                dist.all_reduce(b.flat, op=dist.ReduceOp.SUM, group=self.ddp.pg)
+                if self.ddp.native_comm is not None:
+                    self.ddp.native_comm.all_reduce_sum(b.flat)
+                else:
+                    dist.all_reduce(b.flat, op=dist.ReduceOp.SUM, group=self.ddp.pg)
         self.opt_graph.replay()
         return self.losses[slot]

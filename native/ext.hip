@@ -9,6 +9,7 @@
 #include "elementwise.hip"
 #include "bf16.hip"
 #include "resnet.hip"
+#include "comm.hip"
 
 namespace {
 
@@ -374,6 +375,7 @@ void fused_sgd_step(at::Tensor ptrs, at::Tensor chunks, long chunk_elems,
 
 #include "ext_bf16.inc"
 #include "ext_resnet.inc"
+#include "ext_comm.inc"
 
 }  // namespace
 
@@ -409,6 +411,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("maxpool2x2_relu_bf16_bwd", &maxpool2x2_relu_bf16_bwd);
   m.def("dropout_bf16_fwd", &dropout_bf16_fwd);
   m.def("dropout_bf16_bwd", &dropout_bf16_bwd);
+  m.def("comm_unique_id", &comm_unique_id);
+  m.def("comm_init", &comm_init);
+  m.def("comm_all_reduce_sum", &comm_all_reduce_sum);
+  m.def("comm_destroy", &comm_destroy);
+  m.def("comm_rank", &comm_rank);
+  m.def("comm_world", &comm_world);
   m.def("bn_fwd_bf16", &bn_fwd_bf16);
   m.def("bn_bwd_bf16", &bn_bwd_bf16);
   m.def("maxpool3x3s2_fwd_bf16", &maxpool3x3s2_fwd_bf16);

@@ -234,3 +234,33 @@ def test_training_loss_decreases_gpu():
         losses.append(loss.item())
     tail = sum(losses[-10:]) / 10
     assert tail < losses[0] * 0.7, (losses[0], tail, losses[::30])
+
+
+@pytest.mark.timeout(120)
+def test_native_rccl_comm_single_rank(free_port):
+    """C++ RCCL comm core plumbing: bootstrap over a 1-rank gloo group,
+    allreduce on device (identity at world=1), destroy."""
+    import os
+
+    import torch.distributed as dist
+
+    from dist_mnist_amd.parallel import RcclComm
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(free_port)
+    dist.init_process_group("gloo", rank=0, world_size=1)
+    try:
+        comm = RcclComm.bootstrap()
+        t = torch.arange(1024, dtype=torch.float32, device=DEV)
+        ref = t.clone()
+        comm.all_reduce_sum(t)
+        torch.cuda.synchronize()
+        torch.testing.assert_close(t, ref)
+        tb = torch.randn(64, dtype=torch.bfloat16, device=DEV)
+        refb = tb.clone()
+        comm.all_reduce_sum(tb)
+        torch.cuda.synchronize()
+        torch.testing.assert_close(tb, refb)
+        comm.destroy()
+    finally:
+        dist.destroy_process_group()

@@ -46,7 +46,7 @@ def build(verbose: bool = True) -> str:
     cmd += [f"-I{i}" for i in incs]
     cmd += [f"-L{d}" for d in libdirs]
     cmd += ["-ltorch", "-ltorch_hip", "-lc10", "-lc10_hip", "-ltorch_python",
-            "-lamdhip64"]
+            "-lamdhip64", "-lrccl"]
     if verbose:
         print(" ".join(cmd), flush=True)
     subprocess.run(cmd, check=True)
