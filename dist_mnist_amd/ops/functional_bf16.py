@@ -56,13 +56,22 @@ def build_identity_perm(cik: int, c_can: int, device) -> torch.Tensor:
     return perm.to(device)
 
 
+def _use_tiled(w, stride):
+    """hot-path router: the LDS-tiled input-stationary kernel covers square
+    3x3/5x5 stride-1 convs (all ResNet 3x3 s1 convs + LeNet conv2); other
+    shapes use the generic implicit-GEMM gather kernel."""
+    return stride == 1 and w.shape[2] == w.shape[3] and w.shape[2] in (3, 5)
+
+
 class _Conv2dBF16Fn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, b, stride, pad, cop, relu):
-        y = native.ext().conv2d_fwd_bf16(
-            x.contiguous(), w.contiguous(),
-            b.contiguous() if b is not None else torch.Tensor(), cop, stride,
-            pad, relu)
+        xc, wc = x.contiguous(), w.contiguous()
+        bc = b.contiguous() if b is not None else torch.Tensor()
+        if _use_tiled(w, stride):
+            y = native.ext().conv_tiled_s1_bf16(xc, wc, bc, cop, pad, relu, False)
+        else:
+            y = native.ext().conv2d_fwd_bf16(xc, wc, bc, cop, stride, pad, relu)
         ctx.save_for_backward(x, w)
         ctx.meta = (stride, pad, b is not None)
         return y
@@ -75,8 +84,14 @@ class _Conv2dBF16Fn(torch.autograd.Function):
         m = native.ext()
         dx = dw = db = None
         if ctx.needs_input_grad[0]:
-            dx = m.conv2d_dgrad_bf16(dy, w, x.shape[1], x.shape[2], x.shape[3],
-                                     stride, pad)
+            if _use_tiled(w, stride):
+                # s1 dgrad of a KxK conv is a KxK s1 conv with pad' = K-1-pad
+                dx = m.conv_tiled_s1_bf16(dy, w.contiguous(), torch.Tensor(),
+                                          x.shape[3], w.shape[2] - 1 - pad,
+                                          False, True)
+            else:
+                dx = m.conv2d_dgrad_bf16(dy, w, x.shape[1], x.shape[2],
+                                         x.shape[3], stride, pad)
         if ctx.needs_input_grad[1] or (has_bias and ctx.needs_input_grad[2]):
             dw, db = m.conv2d_wgrad_bf16(dy, x.contiguous(), w.shape[1],
                                          w.shape[0], w.shape[2], w.shape[3],

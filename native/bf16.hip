@@ -1209,3 +1209,123 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_pool_bf16_k(
     if (lk == 0 && co0 + lm < CO) db_ws[(long)ms * CO + co0 + lm] = db_part;
   }
 }
+
+// ---------------------------------------------------------------------------
+// LDS-tiled conv, stride 1 (fwd and dgrad), NHWC bf16 — input-stationary.
+// A block owns an 8x8 output-pixel tile of one image and NW*16 output
+// channels; the input region ((8+KH-1)^2 x up-to-64 input channels) is staged
+// into LDS ONCE per ci-chunk, so every im2col re-read of a pixel (KH*KW taps)
+// comes from LDS instead of L2 (the gather engine's bound). Weights stage
+// per-tap ([NW*16][cc]); fp32->bf16 conversion at stage time as usual.
+//  - DGRAD: same kernel with swapped channel roles and flipped taps
+//    (3x3 s1 p1 dgrad IS a 3x3 s1 p1 conv; 5x5 s1 p0 dgrad = 5x5 s1 p4).
+//  - waves: MW=1, wave w owns 16 pixels (2 rows x 8 cols) of the tile.
+// ---------------------------------------------------------------------------
+template <int KH, bool HAS_BIAS, bool RELU, bool DGRAD, int NW>
+__global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
+    const bf16* __restrict__ x, const float* __restrict__ w,
+    const float* __restrict__ bias, bf16* __restrict__ y,
+    int CIp, int CIcan, int H, int W, int COp, int COcan,
+    int OH, int OW, int pad, int N, int tiles_h, int tiles_w) {
+  constexpr int TS = 8;             // output tile side
+  constexpr int R = TS + KH - 1;    // staged region side
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6;
+  const int l = tid & 63;
+  const int lm = l & 15;
+  const int lk = l >> 4;
+
+  // tile decode
+  int tw = blockIdx.x % tiles_w;
+  int rest = blockIdx.x / tiles_w;
+  int th = rest % tiles_h;
+  int n = rest / tiles_h;
+  const int oh0 = th * TS, ow0 = tw * TS;
+  const int co0 = blockIdx.y * (NW * 16);
+
+  // per-pixel stride padded +8 bf16: consecutive pixels land on distinct
+  // bank groups for the 16-pixel b128 A-reads (64-ch stride alone = 2-way)
+  __shared__ bf16 lds_x[R * R * 72 + 8];     // [ry][rx][ccp] (cc <= 64)
+  __shared__ bf16 lds_w[NW * 16][64 + 8];    // per-tap [co][cc]
+
+  // this lane's pixel (A row = lm; all taps offset from it)
+  const int ml = wid * 16 + lm;
+  const int py = ml >> 3, px = ml & 7;
+
+  f32x4 acc[NW] = {};
+
+  const int KHKW = KH * KH;
+  for (int cic = 0; cic < CIp; cic += 64) {
+    int cc = min(64, CIp - cic);
+    int ccp = cc + 8;
+    __syncthreads();
+    // ---- stage x region: rows [oh0-pad, +R) x cols [ow0-pad, +R) ----------
+    for (int idx = tid; idx < R * R * (cc / 8); idx += 256) {
+      int g8 = idx % (cc / 8);
+      int p = idx / (cc / 8);
+      int rx = p % R, ry = p / R;
+      int ih = oh0 - pad + ry, iw = ow0 - pad + rx;
+      bf16x8 v = bf8_zero();
+      if (ih >= 0 && ih < H && iw >= 0 && iw < W)
+        v = *reinterpret_cast<const bf16x8*>(
+            &x[((long)(n * H + ih) * W + iw) * CIp + cic + g8 * 8]);
+      *reinterpret_cast<bf16x8*>(&lds_x[(ry * R + rx) * ccp + g8 * 8]) = v;
+    }
+    for (int tap = 0; tap < KHKW; ++tap) {
+      int kh = tap / KH, kw = tap % KH;
+      __syncthreads();
+      // ---- stage this tap's weights [NW*16 co][cc] ----------------------
+      for (int idx = tid; idx < NW * 16 * cc; idx += 256) {
+        int co = idx / cc;
+        int ci = idx % cc;
+        float v = 0.f;
+        {
+          if (DGRAD) {
+            // output ch = original ci; input ch = original co; flipped tap
+            int o = co0 + co, i = cic + ci;
+            if (o < COcan && i < CIcan)
+              v = w[((i * COcan) + o) * KHKW + (KH - 1 - kh) * KH + (KH - 1 - kw)];
+          } else {
+            int i = cic + ci;
+            if (i < CIcan)
+              v = w[((co0 + co) * CIcan + i) * KHKW + kh * KH + kw];
+          }
+        }
+        lds_w[co][ci] = to_bf16(v);
+      }
+      __syncthreads();
+      // ---- MFMA over this tap's ci range --------------------------------
+      const bf16* xpix = &lds_x[((py + kh) * R + (px + kw)) * ccp];
+      for (int ks = 0; ks < cc; ks += 32) {
+        bool gin = lk * 8 < cc - ks;
+        bf16x8 a = gin ? *reinterpret_cast<const bf16x8*>(&xpix[ks + lk * 8])
+                       : bf8_zero();
+#pragma unroll
+        for (int ni = 0; ni < NW; ++ni) {
+          bf16x8 b = gin ? *reinterpret_cast<const bf16x8*>(
+                               &lds_w[ni * 16 + lm][ks + lk * 8])
+                         : bf8_zero();
+          acc[ni] = mfma16x16x32bf(a, b, acc[ni]);
+        }
+      }
+    }
+  }
+
+  // epilogue: D row = lk*4+reg -> pixel, col = lm -> co
+#pragma unroll
+  for (int reg = 0; reg < 4; ++reg) {
+    int mo = wid * 16 + lk * 4 + reg;
+    int oy = oh0 + (mo >> 3), ox = ow0 + (mo & 7);
+    if (oy >= OH || ox >= OW) continue;
+    long yb = ((long)(n * OH + oy) * OW + ox) * COp;
+#pragma unroll
+    for (int ni = 0; ni < NW; ++ni) {
+      int co = co0 + ni * 16 + lm;
+      if (co >= COp) continue;
+      float v = (co < COcan) ? acc[ni][reg] : 0.f;
+      if (HAS_BIAS && co < COcan) v += bias[co];
+      if (RELU) v = fmaxf(v, 0.f);
+      y[yb + co] = to_bf16(v);
+    }
+  }
+}

@@ -400,6 +400,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_probe", &mfma_probe);
   m.def("tr16_probe", &tr16_probe);
   m.def("conv2d_fwd_bf16", &conv2d_fwd_bf16);
+  m.def("conv_tiled_s1_bf16", &conv_tiled_s1_bf16);
   m.def("conv2d_dgrad_bf16", &conv2d_dgrad_bf16);
   m.def("conv2d_wgrad_bf16", &conv2d_wgrad_bf16);
   m.def("conv_pool_stem_bf16", &conv_pool_stem_bf16);
