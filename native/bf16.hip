@@ -1215,17 +1215,44 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_pool_bf16_k(
 // A block owns an 8x8 output-pixel tile of one image and NW*16 output
 // channels; the input region ((8+KH-1)^2 x up-to-64 input channels) is staged
 // into LDS ONCE per ci-chunk, so every im2col re-read of a pixel (KH*KW taps)
-// comes from LDS instead of L2 (the gather engine's bound). Weights stage
-// per-tap ([NW*16][cc]); fp32->bf16 conversion at stage time as usual.
-//  - DGRAD: same kernel with swapped channel roles and flipped taps
-//    (3x3 s1 p1 dgrad IS a 3x3 s1 p1 conv; 5x5 s1 p0 dgrad = 5x5 s1 p4).
-//  - waves: MW=1, wave w owns 16 pixels (2 rows x 8 cols) of the tile.
+// comes from LDS instead of L2 (the gather engine's bound). Weights come from
+// a per-call bf16 repack w_rp[tap][out_ch][in_ch] (repack_w_k below): B-frags
+// are direct 16-byte loads from L1/L2-resident global memory — no LDS weight
+// stage, no per-tap barriers. DGRAD reuses the same kernel: the repack writes
+// [flipped tap][ci][co], and an s1 KxK dgrad is an s1 KxK conv with
+// pad' = K-1-pad.
 // ---------------------------------------------------------------------------
-template <int KH, bool HAS_BIAS, bool RELU, bool DGRAD, int NW>
+__global__ void repack_w_k(const float* __restrict__ w, bf16* __restrict__ w_rp,
+                           int CO, int CI, int KHKW, int Ap, int Bp,
+                           int dgrad) {
+  // fwd:  w_rp[tap][co][ci] = w[co][ci][tap]        (Ap=COp, Bp=CIp)
+  // dgrad: w_rp[KHKW-1-tap][ci][co] = w[co][ci][tap] (Ap=CIp, Bp=COp)
+  long total = (long)CO * CI * KHKW;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int tap = i % KHKW;
+    long r = i / KHKW;
+    int ci = r % CI;
+    int co = r / CI;
+    float v = w[i];
+    if (dgrad)
+      w_rp[((long)(KHKW - 1 - tap) * Ap + ci) * Bp + co] = to_bf16(v);
+    else
+      w_rp[((long)tap * Ap + co) * Bp + ci] = to_bf16(v);
+  }
+}
+
+__global__ void zero_bf16_k(bf16* __restrict__ p, long n8) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n8;
+       i += (long)gridDim.x * blockDim.x)
+    *reinterpret_cast<bf16x8*>(p + i * 8) = bf8_zero();
+}
+
+template <int KH, bool HAS_BIAS, bool RELU, int NW>
 __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
-    const bf16* __restrict__ x, const float* __restrict__ w,
+    const bf16* __restrict__ x, const bf16* __restrict__ w_rp,
     const float* __restrict__ bias, bf16* __restrict__ y,
-    int CIp, int CIcan, int H, int W, int COp, int COcan,
+    int CIp, int H, int W, int COp, int COcan,
     int OH, int OW, int pad, int N, int tiles_h, int tiles_w) {
   constexpr int TS = 8;             // output tile side
   constexpr int R = TS + KH - 1;    // staged region side
@@ -1235,7 +1262,6 @@ __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
   const int lm = l & 15;
   const int lk = l >> 4;
 
-  // tile decode
   int tw = blockIdx.x % tiles_w;
   int rest = blockIdx.x / tiles_w;
   int th = rest % tiles_h;
@@ -1245,21 +1271,18 @@ __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
 
   // per-pixel stride padded +8 bf16: consecutive pixels land on distinct
   // bank groups for the 16-pixel b128 A-reads (64-ch stride alone = 2-way)
-  __shared__ bf16 lds_x[R * R * 72 + 8];     // [ry][rx][ccp] (cc <= 64)
-  __shared__ bf16 lds_w[NW * 16][64 + 8];    // per-tap [co][cc]
+  __shared__ bf16 lds_x[R * R * 72 + 8];
 
-  // this lane's pixel (A row = lm; all taps offset from it)
   const int ml = wid * 16 + lm;
   const int py = ml >> 3, px = ml & 7;
 
   f32x4 acc[NW] = {};
 
-  const int KHKW = KH * KH;
+  constexpr int KHKW = KH * KH;
   for (int cic = 0; cic < CIp; cic += 64) {
     int cc = min(64, CIp - cic);
     int ccp = cc + 8;
     __syncthreads();
-    // ---- stage x region: rows [oh0-pad, +R) x cols [ow0-pad, +R) ----------
     for (int idx = tid; idx < R * R * (cc / 8); idx += 256) {
       int g8 = idx % (cc / 8);
       int p = idx / (cc / 8);
@@ -1271,31 +1294,12 @@ __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
             &x[((long)(n * H + ih) * W + iw) * CIp + cic + g8 * 8]);
       *reinterpret_cast<bf16x8*>(&lds_x[(ry * R + rx) * ccp + g8 * 8]) = v;
     }
+    __syncthreads();
+#pragma unroll
     for (int tap = 0; tap < KHKW; ++tap) {
       int kh = tap / KH, kw = tap % KH;
-      __syncthreads();
-      // ---- stage this tap's weights [NW*16 co][cc] ----------------------
-      for (int idx = tid; idx < NW * 16 * cc; idx += 256) {
-        int co = idx / cc;
-        int ci = idx % cc;
-        float v = 0.f;
-        {
-          if (DGRAD) {
-            // output ch = original ci; input ch = original co; flipped tap
-            int o = co0 + co, i = cic + ci;
-            if (o < COcan && i < CIcan)
-              v = w[((i * COcan) + o) * KHKW + (KH - 1 - kh) * KH + (KH - 1 - kw)];
-          } else {
-            int i = cic + ci;
-            if (i < CIcan)
-              v = w[((co0 + co) * CIcan + i) * KHKW + kh * KH + kw];
-          }
-        }
-        lds_w[co][ci] = to_bf16(v);
-      }
-      __syncthreads();
-      // ---- MFMA over this tap's ci range --------------------------------
       const bf16* xpix = &lds_x[((py + kh) * R + (px + kw)) * ccp];
+      const bf16* wtap = w_rp + ((long)tap * COp + co0) * CIp + cic;
       for (int ks = 0; ks < cc; ks += 32) {
         bool gin = lk * 8 < cc - ks;
         bf16x8 a = gin ? *reinterpret_cast<const bf16x8*>(&xpix[ks + lk * 8])
@@ -1303,7 +1307,7 @@ __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
 #pragma unroll
         for (int ni = 0; ni < NW; ++ni) {
           bf16x8 b = gin ? *reinterpret_cast<const bf16x8*>(
-                               &lds_w[ni * 16 + lm][ks + lk * 8])
+                               &wtap[(long)(ni * 16 + lm) * CIp + ks + lk * 8])
                          : bf8_zero();
           acc[ni] = mfma16x16x32bf(a, b, acc[ni]);
         }
@@ -1311,7 +1315,6 @@ __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
     }
   }
 
-  // epilogue: D row = lk*4+reg -> pixel, col = lm -> co
 #pragma unroll
   for (int reg = 0; reg < 4; ++reg) {
     int mo = wid * 16 + lk * 4 + reg;
