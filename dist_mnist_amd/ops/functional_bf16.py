@@ -56,11 +56,13 @@ def build_identity_perm(cik: int, c_can: int, device) -> torch.Tensor:
     return perm.to(device)
 
 
-def _use_tiled(w, stride):
-    """hot-path router: the LDS-tiled input-stationary kernel covers square
-    3x3/5x5 stride-1 convs (all ResNet 3x3 s1 convs + LeNet conv2); other
-    shapes use the generic implicit-GEMM gather kernel."""
-    return stride == 1 and w.shape[2] == w.shape[3] and w.shape[2] in (3, 5)
+def _use_tiled(x_cip, w, stride):
+    """hot-path router (measured): the LDS-tiled input-stationary kernel wins
+    for 3x3 stride-1 convs with >=64 input channels (ResNet body); thin-
+    channel or 5x5 shapes stay on the implicit-GEMM gather kernel (its
+    k-groups are already full there and the tile setup doesn't pay)."""
+    return (stride == 1 and w.shape[2] == w.shape[3] and w.shape[2] == 3
+            and x_cip >= 64)
 
 
 class _Conv2dBF16Fn(torch.autograd.Function):
@@ -68,7 +70,7 @@ class _Conv2dBF16Fn(torch.autograd.Function):
     def forward(ctx, x, w, b, stride, pad, cop, relu):
         xc, wc = x.contiguous(), w.contiguous()
         bc = b.contiguous() if b is not None else torch.Tensor()
-        if _use_tiled(w, stride):
+        if _use_tiled(x.shape[3], w, stride):
             y = native.ext().conv_tiled_s1_bf16(xc, wc, bc, cop, pad, relu, False)
         else:
             y = native.ext().conv2d_fwd_bf16(xc, wc, bc, cop, stride, pad, relu)
@@ -84,7 +86,8 @@ class _Conv2dBF16Fn(torch.autograd.Function):
         m = native.ext()
         dx = dw = db = None
         if ctx.needs_input_grad[0]:
-            if _use_tiled(w, stride):
+            # dgrad reads dy (channels = COp of the conv): route on that
+            if _use_tiled(dy.shape[3], w, stride):
                 # s1 dgrad of a KxK conv is a KxK s1 conv with pad' = K-1-pad
                 dx = m.conv_tiled_s1_bf16(dy, w.contiguous(), torch.Tensor(),
                                           x.shape[3], w.shape[2] - 1 - pad,
