@@ -64,10 +64,33 @@ def parse_args(argv=None):
                    help="bf16 = native NHWC engine on GPU (fp32 master weights; "
                         "checkpoints unchanged)")
     p.add_argument("--resume", type=str, default=None)
+    p.add_argument("--eval", action="store_true",
+                   help="per-epoch accuracy on a held-out synthetic split (rank 0)")
+    p.add_argument("--perf-log", action="store_true",
+                   help="per-epoch images/sec + gradient-sync time")
     p.add_argument("--steps-per-epoch", type=int, default=None,
                    help="cap steps per epoch (synthetic data is unbounded)")
     p.add_argument("--dataset-size", type=int, default=60000)
     return p.parse_args(argv)
+
+
+def evaluate(model, device, use_bf16, n=2048, seed=99991):
+    """Held-out accuracy on a fixed synthetic split (extension; the reference
+    lineage's DDP variant carries a test() loop — SURVEY.md §2a)."""
+    from dist_mnist_amd.data import SyntheticMNIST
+
+    ds = SyntheticMNIST(n=n, seed=seed)
+    model.eval()
+    correct = 0
+    with torch.no_grad():
+        for i in range(0, n, 256):
+            x, t = ds.batch(range(i, min(n, i + 256)), device=device)
+            if use_bf16:
+                x = x[:, 0].to(torch.bfloat16).unsqueeze(-1).contiguous()
+            logp = model(x)
+            correct += int((logp.argmax(1) == t).sum())
+    model.train()
+    return correct / n
 
 
 def run(rank: int, world_size: int, args):
@@ -89,10 +112,13 @@ def run(rank: int, world_size: int, args):
     optimizer = SGD(model.parameters(), lr=args.lr, momentum=args.momentum)
 
     num_batches = len(loader.dataset) // bsz if bsz else len(loader)
+    import time as _time
     for epoch in range(args.epochs):
         model.train()
         epoch_loss = 0.0
         steps = 0
+        sync_s = 0.0
+        t_epoch = _time.perf_counter()
         batches = Prefetcher(iter(loader), device) if use_cuda else loader
         for data, target in batches:
             if not use_cuda:
@@ -105,10 +131,12 @@ def run(rank: int, world_size: int, args):
             _, loss = model(data, target)
             loss.backward()
             if world_size > 1:
+                t_s = _time.perf_counter()
                 if isinstance(model, DistributedDataParallel):
                     model.sync()
                 else:
                     average_gradients(model)
+                sync_s += _time.perf_counter() - t_s
             optimizer.step()
             epoch_loss += loss.item()
             steps += 1
@@ -116,6 +144,15 @@ def run(rank: int, world_size: int, args):
                 break
         # the reference's exact per-epoch line (SURVEY.md §5 observability)
         print("Rank ", rank, ", epoch ", epoch, ": ", epoch_loss / max(1, steps))
+        if args.perf_log:
+            dt = _time.perf_counter() - t_epoch
+            print(f"[perf] rank {rank} epoch {epoch}: "
+                  f"{steps * bsz / max(dt, 1e-9):.0f} images/s, "
+                  f"grad-sync {sync_s * 1000:.1f} ms total")
+        if args.eval and rank == 0:
+            base = model.module if isinstance(model, DistributedDataParallel) else model
+            acc = evaluate(base, device, use_bf16)
+            print(f"[eval] epoch {epoch}: held-out accuracy {acc:.4f}")
 
     if args.modelpath and rank == 0:
         save_checkpoint(model, args.modelpath)

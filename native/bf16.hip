@@ -1332,3 +1332,53 @@ __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
     }
   }
 }
+
+// ---------------------------------------------------------------------------
+// Wide-class LogSoftmax+NLL (ResNet: C=1000): one WAVE per row, lanes split
+// the class axis, shfl tree reductions. (The thread-per-row kernel above is
+// for tiny C like LeNet's 10.)
+// ---------------------------------------------------------------------------
+__global__ void logsoftmax_nll_bf16_wide_fwd_k(
+    const bf16* __restrict__ x, const long* __restrict__ target,
+    float* __restrict__ logp, float* __restrict__ loss, int M, int COp, int C,
+    int with_loss) {
+  __shared__ float red[4];
+  int wid = threadIdx.x >> 6;
+  int l = threadIdx.x & 63;
+  float part = 0.f;
+  for (long m = (long)blockIdx.x * 4 + wid; m < M; m += (long)gridDim.x * 4) {
+    const bf16* row = x + m * COp;
+    float mx = -3.4e38f;
+    for (int c = l; c < C; c += 64) mx = fmaxf(mx, to_f32(row[c]));
+    for (int off = 32; off > 0; off >>= 1) mx = fmaxf(mx, __shfl_xor(mx, off));
+    float s = 0.f;
+    for (int c = l; c < C; c += 64) s += __expf(to_f32(row[c]) - mx);
+    for (int off = 32; off > 0; off >>= 1) s += __shfl_xor(s, off);
+    float lse = mx + __logf(s);
+    float* out = logp + m * C;
+    for (int c = l; c < C; c += 64) out[c] = to_f32(row[c]) - lse;
+    if (with_loss && l == 0) part -= to_f32(row[target[m]]) - lse;
+  }
+  if (with_loss) {
+    if (l == 0) red[wid] = part;
+    __syncthreads();
+    if (threadIdx.x == 0)
+      atomicAdd(loss, (red[0] + red[1] + red[2] + red[3]) / (float)M);
+  }
+}
+
+__global__ void logsoftmax_nll_bf16_wide_bwd_k(
+    const float* __restrict__ logp, const long* __restrict__ target,
+    const float* __restrict__ dloss, bf16* __restrict__ dx, int M, int COp,
+    int C) {
+  float scale = dloss[0] / (float)M;
+  long total = (long)M * COp;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int c = i % COp;
+    long m = i / COp;
+    float v = 0.f;
+    if (c < C) v = (__expf(logp[m * C + c]) - (c == target[m] ? 1.f : 0.f)) * scale;
+    dx[i] = to_bf16(v);
+  }
+}
