@@ -62,6 +62,15 @@ class GraphedStep:
         with torch.cuda.graph(self.opt_graph, pool=pool):
             self._eager_opt(scale=True)
 
+        # warmup steps ran WITHOUT gradient sync (hooks disabled, rank-local
+        # data) so ranks drifted slightly; re-broadcast params in place —
+        # the captured graphs hold raw pointers, so an in-place copy is safe.
+        if self.ddp is not None and self.ddp.enabled:
+            with torch.no_grad():
+                for p in self.ddp.module.parameters():
+                    dist.broadcast(p.data, src=0, group=self.ddp.pg)
+            torch.cuda.synchronize(self.device)
+
     # -- eager pieces (also used during warmup/capture) ---------------------
     def _zero_grads(self):
         if self.ddp is not None:
