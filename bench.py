@@ -107,7 +107,14 @@ def main():
 
         x0, t0 = next(batches)
         ddp = model if isinstance(model, DistributedDataParallel) else None
-        graphed = GraphedStep(model, opt, x0, t0, ddp=ddp)
+        try:
+            graphed = GraphedStep(model, opt, x0, t0, ddp=ddp)
+        except Exception as e:  # capture failure must never kill the bench
+            print(f"# hipGraph capture failed ({type(e).__name__}: {e}); "
+                  "falling back to eager launches", flush=True)
+            graphed = None
+            if ddp is not None:
+                ddp._install_hooks()  # restore comm-in-backward overlap
 
     def one_step():
         x, t = next(batches)

@@ -293,3 +293,37 @@ def test_conv_pool_stem_fused():
                                atol=2e-2 * max(1.0, dw_ref.abs().max().item()))
     torch.testing.assert_close(b.grad, db_ref, rtol=2e-2,
                                atol=2e-2 * max(1.0, db_ref.abs().max().item()))
+
+
+def test_graphed_step_replay():
+    """hipGraph-captured step: replays train, inputs swap per slot, dropout
+    masks stay fresh across replays (device-side counter)."""
+    from dist_mnist_amd.models import Net
+    from dist_mnist_amd.ops import SGD
+    from dist_mnist_amd.runtime import GraphedStep
+
+    torch.manual_seed(60)
+    model = Net().to(DEV).train().bf16_native_()
+    opt = SGD(model.parameters(), lr=0.05, momentum=0.5)
+    x0 = torch.zeros(64, 28, 28, 1, dtype=torch.bfloat16, device=DEV)
+    x0[..., 0] = randn(64, 28, 28, seed=61).to(torch.bfloat16)
+    t0 = torch.randint(0, 10, (64,), device=DEV)
+
+    gs = GraphedStep(model, opt, x0, t0)
+    p0 = model.fc2.weight.clone()
+    losses = []
+    for i in range(20):
+        slot = i % 2
+        gs.slots[slot][0].copy_(x0)
+        gs.slots[slot][1].copy_(t0)
+        loss = gs.step(slot)
+        losses.append(float(loss))
+    torch.cuda.synchronize()
+    # params actually moved and loss went down on the fixed batch
+    assert not torch.equal(p0, model.fc2.weight)
+    assert losses[-1] < losses[0], losses
+    # dropout freshness: same input, training mode => losses vary replay to
+    # replay (a frozen mask would give a monotone deterministic sequence;
+    # check successive deltas are not all identical signs/magnitudes)
+    diffs = {round(losses[i + 1] - losses[i], 6) for i in range(10, 19)}
+    assert len(diffs) > 1
