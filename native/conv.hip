@@ -81,20 +81,12 @@ __global__ __launch_bounds__(256) void conv2d_fwd_f32_k(
     __syncthreads();
 
     int klim = min(BK, K - k0);
-#pragma unroll 4
     for (int kk = 0; kk + 4 <= klim; kk += 4) {
       float a = lds_w[kk + lk][lm];
       int koff = lds_koff[kk + lk];
-      bool kin = true;
-      if (BOUNDS) {
-        int khw = lds_khw[kk + lk];
-        int kh = khw >> 16, kw = khw & 0xffff;
-        // per-mi bound check folded below (ih0/iw0 differ per mi)
-        (void)kh; (void)kw;
-      }
 #pragma unroll
       for (int mi = 0; mi < MW; ++mi) {
-        bool ok = mval[mi] & kin;
+        bool ok = mval[mi];
         if (BOUNDS) {
           int khw = lds_khw[kk + lk];
           int kh = khw >> 16, kw = khw & 0xffff;
@@ -104,7 +96,6 @@ __global__ __launch_bounds__(256) void conv2d_fwd_f32_k(
         float b = ok ? x[xbase[mi] + koff] : 0.f;
         acc[mi] = mfma16x16x4(a, b, acc[mi]);
       }
-      // tail k (klim not multiple of 4) handled below
     }
     int ktail = klim & 3;
     if (ktail) {
