@@ -56,13 +56,15 @@ def build_identity_perm(cik: int, c_can: int, device) -> torch.Tensor:
     return perm.to(device)
 
 
-def _use_tiled(x_cip, w, stride):
-    """hot-path router (measured): the LDS-tiled input-stationary kernel wins
-    for 3x3 stride-1 convs with >=64 input channels (ResNet body); thin-
-    channel or 5x5 shapes stay on the implicit-GEMM gather kernel (its
-    k-groups are already full there and the tile setup doesn't pay)."""
-    return (stride == 1 and w.shape[2] == w.shape[3] and w.shape[2] == 3
-            and x_cip >= 64)
+def _use_tiled(x_cip, w, stride, dgrad=False):
+    """hot-path router (tools/microbench.py on MI355X): the LDS-tiled
+    input-stationary kernel wins dgrad for 3x3 s1 convs >=64 in-channels
+    (455-546us vs 545-686 gather) but fwd only from 256 channels up
+    (l4: 530 vs 622; l1: 483 vs 416 — gather wins thin layers). 5x5 and
+    thin-channel shapes stay on the implicit-GEMM gather kernel."""
+    if not (stride == 1 and w.shape[2] == w.shape[3] and w.shape[2] == 3):
+        return False
+    return x_cip >= (64 if dgrad else 256)
 
 
 class _Conv2dBF16Fn(torch.autograd.Function):
@@ -70,7 +72,7 @@ class _Conv2dBF16Fn(torch.autograd.Function):
     def forward(ctx, x, w, b, stride, pad, cop, relu):
         xc, wc = x.contiguous(), w.contiguous()
         bc = b.contiguous() if b is not None else torch.Tensor()
-        if _use_tiled(x.shape[3], w, stride):
+        if _use_tiled(x.shape[3], w, stride, dgrad=False):
             y = native.ext().conv_tiled_s1_bf16(xc, wc, bc, cop, pad, relu, False)
         else:
             y = native.ext().conv2d_fwd_bf16(xc, wc, bc, cop, stride, pad, relu)
@@ -87,7 +89,7 @@ class _Conv2dBF16Fn(torch.autograd.Function):
         dx = dw = db = None
         if ctx.needs_input_grad[0]:
             # dgrad reads dy (channels = COp of the conv): route on that
-            if _use_tiled(dy.shape[3], w, stride):
+            if _use_tiled(dy.shape[3], w, stride, dgrad=True):
                 # s1 dgrad of a KxK conv is a KxK s1 conv with pad' = K-1-pad
                 dx = m.conv_tiled_s1_bf16(dy, w.contiguous(), torch.Tensor(),
                                           x.shape[3], w.shape[2] - 1 - pad,
