@@ -1382,3 +1382,129 @@ __global__ void logsoftmax_nll_bf16_wide_bwd_k(
     dx[i] = to_bf16(v);
   }
 }
+
+// ---------------------------------------------------------------------------
+// Stride-2 dgrad, parity-specialized (ResNet downsample convs). For s=2 a
+// dx pixel (ih, iw) only receives taps with kh === (ih+pad) (mod 2) and
+// kw === (iw+pad) (mod 2). Grouping dx pixels by parity class (blockIdx.z)
+// makes the valid-tap set WAVE-UNIFORM: the kernel loops just that subset
+// (1/2/2/4 taps of 9 for 3x3) with no divisibility predicates, unlike the
+// generic dgrad which scans all KH*KW*COp with per-lane checks.
+// ---------------------------------------------------------------------------
+template <int MW, int NW>
+__global__ __launch_bounds__(256) void conv2d_dgrad_s2_bf16_k(
+    const bf16* __restrict__ dy, const float* __restrict__ w,
+    bf16* __restrict__ dx,
+    int CIp, int CI, int H, int W, int COp, int CO, int KH, int KW,
+    int OH, int OW, int pad, int N) {
+  constexpr int BK = 32;
+  const int BM = 4 * MW * 16;
+  const int cy = blockIdx.z >> 1, cx = blockIdx.z & 1;
+  const int Hc = (H - cy + 1) / 2, Wc = (W - cx + 1) / 2;
+  const long Mc = (long)N * Hc * Wc;   // pixels in this parity class
+  const int m0 = blockIdx.x * BM;
+  const int ci0 = blockIdx.y * (NW * 16);
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6;
+  const int l = tid & 63;
+  const int lm = l & 15;
+  const int lk = l >> 4;
+
+  // valid tap grid for this class
+  const int kh0 = (cy + pad) & 1;
+  const int kw0 = (cx + pad) & 1;
+  const int nkh = (KH - kh0 + 1) / 2;
+  const int nkw = (KW - kw0 + 1) / 2;
+  const int Kp = nkh * nkw * COp;      // reduced k' space
+
+  __shared__ bf16 lds_w[NW * 16][BK + 8];
+  __shared__ int lds_kdec[BK / 8];     // (kh<<20)|(kw<<10)|co0g
+
+  int ih[MW], iw[MW], nn[MW];
+  bool mval[MW];
+#pragma unroll
+  for (int mi = 0; mi < MW; ++mi) {
+    long m = m0 + (wid * MW + mi) * 16 + lm;
+    mval[mi] = m < Mc;
+    long mm = mval[mi] ? m : 0;
+    long per_img = (long)Hc * Wc;
+    nn[mi] = (int)(mm / per_img);
+    int r = (int)(mm % per_img);
+    ih[mi] = cy + 2 * (r / Wc);
+    iw[mi] = cx + 2 * (r % Wc);
+  }
+
+  f32x4 acc[MW][NW] = {};
+
+  for (int k0 = 0; k0 < Kp; k0 += BK) {
+    __syncthreads();
+    for (int idx = tid; idx < NW * 16 * BK; idx += 256) {
+      int ci = idx >> 5;
+      int kk = idx & 31;
+      int k = k0 + kk;
+      float v = 0.f;
+      if (k < Kp && ci0 + ci < CI) {
+        int t = k / COp, co = k % COp;
+        int kh = kh0 + 2 * (t / nkw), kw = kw0 + 2 * (t % nkw);
+        if (co < CO)
+          v = w[((co * CI) + ci0 + ci) * (KH * KW) + kh * KW + kw];
+      }
+      lds_w[ci][kk] = to_bf16(v);
+    }
+    if (tid < BK / 8) {
+      int k = k0 + tid * 8;
+      int kcl = k < Kp ? k : 0;
+      int t = kcl / COp, co0g = kcl % COp;
+      int kh = kh0 + 2 * (t / nkw), kw = kw0 + 2 * (t % nkw);
+      lds_kdec[tid] = (kh << 20) | (kw << 10) | co0g;
+    }
+    __syncthreads();
+
+    int kgroups = min(BK, Kp - k0) / 8;
+    for (int kg = 0; kg < kgroups; kg += 4) {
+      bool gin = lk < (kgroups - kg);
+      int gsel = gin ? kg + lk : 0;
+      int kdec = lds_kdec[gsel];
+      int kh = kdec >> 20, kw = (kdec >> 10) & 0x3ff, co0g = kdec & 0x3ff;
+      bf16x8 bw[NW];
+#pragma unroll
+      for (int ni = 0; ni < NW; ++ni)
+        bw[ni] = gin ? *reinterpret_cast<const bf16x8*>(&lds_w[ni * 16 + lm][gsel * 8])
+                     : bf8_zero();
+#pragma unroll
+      for (int mi = 0; mi < MW; ++mi) {
+        bf16x8 a = bf8_zero();
+        if (mval[mi] & gin) {
+          int oh = (ih[mi] + pad - kh) >> 1;
+          int ow = (iw[mi] + pad - kw) >> 1;
+          if (oh >= 0 && oh < OH && ow >= 0 && ow < OW)
+            a = *reinterpret_cast<const bf16x8*>(
+                &dy[((long)(nn[mi] * OH + oh) * OW + ow) * COp + co0g]);
+        }
+#pragma unroll
+        for (int ni = 0; ni < NW; ++ni)
+          acc[mi][ni] = mfma16x16x32bf(a, bw[ni], acc[mi][ni]);
+      }
+    }
+  }
+
+#pragma unroll
+  for (int mi = 0; mi < MW; ++mi) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      long m = m0 + (wid * MW + mi) * 16 + lk * 4 + reg;
+      if (m >= Mc) continue;
+      long per_img = (long)Hc * Wc;
+      int n = (int)(m / per_img);
+      int r = (int)(m % per_img);
+      int oy = cy + 2 * (r / Wc), ox = cx + 2 * (r % Wc);
+      long xb = ((long)(n * H + oy) * W + ox) * CIp;
+#pragma unroll
+      for (int ni = 0; ni < NW; ++ni) {
+        int ci = ci0 + ni * 16 + lm;
+        if (ci >= CIp) continue;
+        dx[xb + ci] = to_bf16(ci < CI ? acc[mi][ni][reg] : 0.f);
+      }
+    }
+  }
+}
