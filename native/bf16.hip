@@ -1248,14 +1248,16 @@ __global__ void zero_bf16_k(bf16* __restrict__ p, long n8) {
     *reinterpret_cast<bf16x8*>(p + i * 8) = bf8_zero();
 }
 
-template <int KH, bool HAS_BIAS, bool RELU, int NW>
+template <int KH, bool HAS_BIAS, bool RELU, int NW, int MW>
 __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
     const bf16* __restrict__ x, const bf16* __restrict__ w_rp,
     const float* __restrict__ bias, bf16* __restrict__ y,
     int CIp, int H, int W, int COp, int COcan,
     int OH, int OW, int pad, int N, int tiles_h, int tiles_w) {
-  constexpr int TS = 8;             // output tile side
-  constexpr int R = TS + KH - 1;    // staged region side
+  // output tile: 8 rows x (8*MW) cols; wave owns MW 16-pixel fragments at
+  // the same rows, so every B-fragment read feeds MW MFMAs.
+  constexpr int TH = 8, TW = 8 * MW;
+  constexpr int RH = TH + KH - 1, RW = TW + KH - 1;
   const int tid = threadIdx.x;
   const int wid = tid >> 6;
   const int l = tid & 63;
@@ -1266,69 +1268,81 @@ __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
   int rest = blockIdx.x / tiles_w;
   int th = rest % tiles_h;
   int n = rest / tiles_h;
-  const int oh0 = th * TS, ow0 = tw * TS;
+  const int oh0 = th * TH, ow0 = tw * TW;
   const int co0 = blockIdx.y * (NW * 16);
 
-  // per-pixel stride padded +8 bf16: consecutive pixels land on distinct
-  // bank groups for the 16-pixel b128 A-reads (64-ch stride alone = 2-way)
-  __shared__ bf16 lds_x[R * R * 72 + 8];
+  // per-pixel stride padded +8 bf16 (bank spread for the 16-pixel b128 reads)
+  __shared__ bf16 lds_x[RH * RW * 72 + 8];
 
-  const int ml = wid * 16 + lm;
-  const int py = ml >> 3, px = ml & 7;
+  // wave pixel map: fragment mi covers a 2-row x 8-col patch; the block's 4
+  // waves cover the 8 rows. pixel (py, px) for (wave, mi, lm):
+  //   py = wid*2 + (lm>>3), px = mi*8 + (lm&7)
+  const int py = wid * 2 + (lm >> 3);
 
-  f32x4 acc[NW] = {};
+  f32x4 acc[MW][NW] = {};
 
   constexpr int KHKW = KH * KH;
   for (int cic = 0; cic < CIp; cic += 64) {
     int cc = min(64, CIp - cic);
     int ccp = cc + 8;
     __syncthreads();
-    for (int idx = tid; idx < R * R * (cc / 8); idx += 256) {
+    for (int idx = tid; idx < RH * RW * (cc / 8); idx += 256) {
       int g8 = idx % (cc / 8);
       int p = idx / (cc / 8);
-      int rx = p % R, ry = p / R;
+      int rx = p % RW, ry = p / RW;
       int ih = oh0 - pad + ry, iw = ow0 - pad + rx;
       bf16x8 v = bf8_zero();
       if (ih >= 0 && ih < H && iw >= 0 && iw < W)
         v = *reinterpret_cast<const bf16x8*>(
             &x[((long)(n * H + ih) * W + iw) * CIp + cic + g8 * 8]);
-      *reinterpret_cast<bf16x8*>(&lds_x[(ry * R + rx) * ccp + g8 * 8]) = v;
+      *reinterpret_cast<bf16x8*>(&lds_x[(ry * RW + rx) * ccp + g8 * 8]) = v;
     }
     __syncthreads();
 #pragma unroll
     for (int tap = 0; tap < KHKW; ++tap) {
       int kh = tap / KH, kw = tap % KH;
-      const bf16* xpix = &lds_x[((py + kh) * R + (px + kw)) * ccp];
       const bf16* wtap = w_rp + ((long)tap * COp + co0) * CIp + cic;
       for (int ks = 0; ks < cc; ks += 32) {
         bool gin = lk * 8 < cc - ks;
-        bf16x8 a = gin ? *reinterpret_cast<const bf16x8*>(&xpix[ks + lk * 8])
+        bf16x8 bw[NW];
+#pragma unroll
+        for (int ni = 0; ni < NW; ++ni)
+          bw[ni] = gin ? *reinterpret_cast<const bf16x8*>(
+                             &wtap[(long)(ni * 16 + lm) * CIp + ks + lk * 8])
                        : bf8_zero();
 #pragma unroll
-        for (int ni = 0; ni < NW; ++ni) {
-          bf16x8 b = gin ? *reinterpret_cast<const bf16x8*>(
-                               &wtap[(long)(ni * 16 + lm) * CIp + ks + lk * 8])
+        for (int mi = 0; mi < MW; ++mi) {
+          int px = mi * 8 + (lm & 7);
+          const bf16* xpix = &lds_x[((py + kh) * RW + (px + kw)) * ccp];
+          bf16x8 a = gin ? *reinterpret_cast<const bf16x8*>(&xpix[ks + lk * 8])
                          : bf8_zero();
-          acc[ni] = mfma16x16x32bf(a, b, acc[ni]);
+#pragma unroll
+          for (int ni = 0; ni < NW; ++ni)
+            acc[mi][ni] = mfma16x16x32bf(a, bw[ni], acc[mi][ni]);
         }
       }
     }
   }
 
+  // epilogue: D row = lk*4+reg indexes the fragment's 16-pixel space
 #pragma unroll
-  for (int reg = 0; reg < 4; ++reg) {
-    int mo = wid * 16 + lk * 4 + reg;
-    int oy = oh0 + (mo >> 3), ox = ow0 + (mo & 7);
-    if (oy >= OH || ox >= OW) continue;
-    long yb = ((long)(n * OH + oy) * OW + ox) * COp;
+  for (int mi = 0; mi < MW; ++mi) {
 #pragma unroll
-    for (int ni = 0; ni < NW; ++ni) {
-      int co = co0 + ni * 16 + lm;
-      if (co >= COp) continue;
-      float v = (co < COcan) ? acc[ni][reg] : 0.f;
-      if (HAS_BIAS && co < COcan) v += bias[co];
-      if (RELU) v = fmaxf(v, 0.f);
-      y[yb + co] = to_bf16(v);
+    for (int reg = 0; reg < 4; ++reg) {
+      int mo = lk * 4 + reg;                 // 0..15 within fragment
+      int oy = oh0 + wid * 2 + (mo >> 3);
+      int ox = ow0 + mi * 8 + (mo & 7);
+      if (oy >= OH || ox >= OW) continue;
+      long yb = ((long)(n * OH + oy) * OW + ox) * COp;
+#pragma unroll
+      for (int ni = 0; ni < NW; ++ni) {
+        int co = co0 + ni * 16 + lm;
+        if (co >= COp) continue;
+        float v = (co < COcan) ? acc[mi][ni][reg] : 0.f;
+        if (HAS_BIAS && co < COcan) v += bias[co];
+        if (RELU) v = fmaxf(v, 0.f);
+        y[yb + co] = to_bf16(v);
+      }
     }
   }
 }
