@@ -57,14 +57,16 @@ def build_identity_perm(cik: int, c_can: int, device) -> torch.Tensor:
 
 
 def _use_tiled(x_cip, w, stride, dgrad=False):
-    """hot-path router (tools/microbench.py on MI355X): the LDS-tiled
-    input-stationary kernel wins dgrad for 3x3 s1 convs >=64 in-channels
-    (455-546us vs 545-686 gather) but fwd only from 256 channels up
-    (l4: 530 vs 622; l1: 483 vs 416 — gather wins thin layers). 5x5 and
-    thin-channel shapes stay on the implicit-GEMM gather kernel."""
-    if not (stride == 1 and w.shape[2] == w.shape[3] and w.shape[2] == 3):
+    """hot-path router (tools/microbench.py on MI355X, MW=2 kernel): the
+    LDS-tiled input-stationary kernel wins every s1 dgrad (264-530us vs
+    544-686 gather at 64-512ch; 154 vs 172 for LeNet's 5x5) and 3x3 s1
+    forwards from 64 channels (331-530 vs 415-620). Thin-channel forwards
+    (LeNet conv2 5x5: 107 gather vs 122 tiled) stay on the gather kernel."""
+    if not (stride == 1 and w.shape[2] == w.shape[3] and w.shape[2] in (3, 5)):
         return False
-    return x_cip >= (64 if dgrad else 256)
+    if dgrad:
+        return True
+    return w.shape[2] == 3 and x_cip >= 64
 
 
 class _Conv2dBF16Fn(torch.autograd.Function):
