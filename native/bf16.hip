@@ -1248,15 +1248,19 @@ __global__ void zero_bf16_k(bf16* __restrict__ p, long n8) {
     *reinterpret_cast<bf16x8*>(p + i * 8) = bf8_zero();
 }
 
-template <int KH, bool HAS_BIAS, bool RELU, int NW, int MW>
+template <int KH, bool HAS_BIAS, bool RELU, int NW, int MW, bool MWN = false>
 __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
     const bf16* __restrict__ x, const bf16* __restrict__ w_rp,
     const float* __restrict__ bias, bf16* __restrict__ y,
     int CIp, int H, int W, int COp, int COcan,
     int OH, int OW, int pad, int N, int tiles_h, int tiles_w) {
   // output tile: 8 rows x (8*MW) cols; wave owns MW 16-pixel fragments at
-  // the same rows, so every B-fragment read feeds MW MFMAs.
-  constexpr int TH = 8, TW = 8 * MW;
+  // the same rows, so every B-fragment read feeds MW MFMAs. MWN variant
+  // (small maps, e.g. ResNet layer4's 7x7): the MW fragments are the SAME
+  // pixels of MW consecutive IMAGES (region staged per image) — identical
+  // B-reuse without needing a 16-wide output row.
+  constexpr int TH = 8, TW = MWN ? 8 : 8 * MW;
+  constexpr int NIMG = MWN ? MW : 1;
   constexpr int RH = TH + KH - 1, RW = TW + KH - 1;
   const int tid = threadIdx.x;
   const int wid = tid >> 6;
@@ -1267,12 +1271,12 @@ __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
   int tw = blockIdx.x % tiles_w;
   int rest = blockIdx.x / tiles_w;
   int th = rest % tiles_h;
-  int n = rest / tiles_h;
+  int n = (rest / tiles_h) * NIMG;   // first image of this block's group
   const int oh0 = th * TH, ow0 = tw * TW;
   const int co0 = blockIdx.y * (NW * 16);
 
   // per-pixel stride padded +8 bf16 (bank spread for the 16-pixel b128 reads)
-  __shared__ bf16 lds_x[RH * RW * 72 + 8];
+  __shared__ bf16 lds_x[NIMG * RH * RW * 72 + 8];
 
   // wave pixel map: fragment mi covers a 2-row x 8-col patch; the block's 4
   // waves cover the 8 rows. pixel (py, px) for (wave, mi, lm):
@@ -1286,16 +1290,20 @@ __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
     int cc = min(64, CIp - cic);
     int ccp = cc + 8;
     __syncthreads();
-    for (int idx = tid; idx < RH * RW * (cc / 8); idx += 256) {
+    for (int idx = tid; idx < NIMG * RH * RW * (cc / 8); idx += 256) {
       int g8 = idx % (cc / 8);
       int p = idx / (cc / 8);
-      int rx = p % RW, ry = p / RW;
+      int rx = p % RW;
+      int rest2 = p / RW;
+      int ry = rest2 % RH;
+      int img = rest2 / RH;
       int ih = oh0 - pad + ry, iw = ow0 - pad + rx;
       bf16x8 v = bf8_zero();
-      if (ih >= 0 && ih < H && iw >= 0 && iw < W)
+      if (ih >= 0 && ih < H && iw >= 0 && iw < W && n + img < N)
         v = *reinterpret_cast<const bf16x8*>(
-            &x[((long)(n * H + ih) * W + iw) * CIp + cic + g8 * 8]);
-      *reinterpret_cast<bf16x8*>(&lds_x[(ry * RW + rx) * ccp + g8 * 8]) = v;
+            &x[((long)((n + img) * H + ih) * W + iw) * CIp + cic + g8 * 8]);
+      *reinterpret_cast<bf16x8*>(
+          &lds_x[(img * RH * RW + ry * RW + rx) * ccp + g8 * 8]) = v;
     }
     __syncthreads();
 #pragma unroll
@@ -1312,8 +1320,10 @@ __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
                        : bf8_zero();
 #pragma unroll
         for (int mi = 0; mi < MW; ++mi) {
-          int px = mi * 8 + (lm & 7);
-          const bf16* xpix = &lds_x[((py + kh) * RW + (px + kw)) * ccp];
+          int px = (MWN ? 0 : mi * 8) + (lm & 7);
+          int img = MWN ? mi : 0;
+          const bf16* xpix =
+              &lds_x[(img * RH * RW + (py + kh) * RW + (px + kw)) * ccp];
           bf16x8 a = gin ? *reinterpret_cast<const bf16x8*>(&xpix[ks + lk * 8])
                          : bf8_zero();
 #pragma unroll
@@ -1331,9 +1341,10 @@ __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
     for (int reg = 0; reg < 4; ++reg) {
       int mo = lk * 4 + reg;                 // 0..15 within fragment
       int oy = oh0 + wid * 2 + (mo >> 3);
-      int ox = ow0 + mi * 8 + (mo & 7);
-      if (oy >= OH || ox >= OW) continue;
-      long yb = ((long)(n * OH + oy) * OW + ox) * COp;
+      int ox = ow0 + (MWN ? 0 : mi * 8) + (mo & 7);
+      int nimg = n + (MWN ? mi : 0);
+      if (oy >= OH || ox >= OW || nimg >= N) continue;
+      long yb = ((long)(nimg * OH + oy) * OW + ox) * COp;
 #pragma unroll
       for (int ni = 0; ni < NW; ++ni) {
         int co = co0 + ni * 16 + lm;
