@@ -24,9 +24,10 @@ class _BatchNormBF16Fn(torch.autograd.Function):
     def backward(ctx, dy):
         x, gamma, mean, invstd, yrelu = ctx.saved_tensors
         dy = dy.contiguous()
-        if ctx.relu:
-            dy = native.ext().relu_gate_bf16(dy, yrelu)
-        dx, dgamma, dbeta = native.ext().bn_bwd_bf16(dy, x, mean, invstd, gamma)
+        # the relu gate (yact) folds into both bn backward passes
+        dx, dgamma, dbeta = native.ext().bn_bwd_bf16(
+            dy, x, mean, invstd, gamma,
+            yrelu if ctx.relu else torch.Tensor())
         return dx, dgamma, dbeta, None, None, None, None, None, None
 
 

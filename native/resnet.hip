@@ -129,11 +129,13 @@ __global__ void bn_apply_k(const bf16* __restrict__ x, bf16* __restrict__ y,
 //  reduce: dbeta[c] = sum dy ; dgamma[c] = sum dy * xhat
 //  apply : dx = gamma*invstd * (dy - dbeta/npix - xhat * dgamma/npix)
 // ---------------------------------------------------------------------------
+template <bool RELU>
 __global__ void bn_bwd_reduce_k(const bf16* __restrict__ dy,
                                 const bf16* __restrict__ x,
+                                const bf16* __restrict__ yact,
                                 const float* __restrict__ mean,
                                 const float* __restrict__ invstd,
-                                float* __restrict__ red_out,  // [C][2] zeroed
+                                float* __restrict__ red_out,  // [C][2] partial
                                 long npix, int Cp) {
   int ncg = Cp / 8;
   int cg = threadIdx.x % ncg;
@@ -152,9 +154,12 @@ __global__ void bn_bwd_reduce_k(const bf16* __restrict__ dy,
          i += (long)gridDim.x * lanes) {
       bf16x8 g = *reinterpret_cast<const bf16x8*>(&dy[i * Cp + cg * 8]);
       bf16x8 v = *reinterpret_cast<const bf16x8*>(&x[i * Cp + cg * 8]);
+      bf16x8 ya;
+      if (RELU) ya = *reinterpret_cast<const bf16x8*>(&yact[i * Cp + cg * 8]);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         float gf = to_f32(g[j]);
+        if (RELU && !(to_f32(ya[j]) > 0.f)) gf = 0.f;
         db[j] += gf;
         dg[j] += gf * (to_f32(v[j]) - mu[j]) * is[j];
       }
@@ -184,8 +189,10 @@ __global__ void bn_bwd_reduce_k(const bf16* __restrict__ dy,
   }
 }
 
+template <bool RELU>
 __global__ void bn_bwd_apply_k(const bf16* __restrict__ dy,
                                const bf16* __restrict__ x,
+                               const bf16* __restrict__ yact,
                                const float* __restrict__ mean,
                                const float* __restrict__ invstd,
                                const float* __restrict__ gamma,
@@ -214,11 +221,15 @@ __global__ void bn_bwd_apply_k(const bf16* __restrict__ dy,
        i += (long)gridDim.x * lanes) {
     bf16x8 g = *reinterpret_cast<const bf16x8*>(&dy[i * Cp + cg * 8]);
     bf16x8 v = *reinterpret_cast<const bf16x8*>(&x[i * Cp + cg * 8]);
+    bf16x8 ya;
+    if (RELU) ya = *reinterpret_cast<const bf16x8*>(&yact[i * Cp + cg * 8]);
     bf16x8 o;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
+      float gf = to_f32(g[j]);
+      if (RELU && !(to_f32(ya[j]) > 0.f)) gf = 0.f;
       float xhat = (to_f32(v[j]) - mu[j]) * is[j];
-      float val = gsc[j] * (to_f32(g[j]) - mdb[j] - xhat * mdg[j]);
+      float val = gsc[j] * (gf - mdb[j] - xhat * mdg[j]);
       o[j] = to_bf16(val);
     }
     *reinterpret_cast<bf16x8*>(&dx[i * Cp + cg * 8]) = o;

@@ -165,3 +165,32 @@ def test_resnet18_matches_torch_reference():
     assert rel_rms < 0.12, rel_rms
     agree = (logp.argmax(1) == ref_logp.argmax(1)).float().mean().item()
     assert agree >= 0.5, agree
+
+
+def test_batchnorm_relu_bwd_fused_gate():
+    """The ReLU gate folded into bn backward must match torch's
+    relu(batch_norm(x)) gradients."""
+    N, C, H, W = 4, 16, 10, 12
+    x = randn(N, C, H, W, seed=20)
+    gamma = randn(C, seed=21).abs().add(0.5).requires_grad_(True)
+    beta = randn(C, seed=22).requires_grad_(True)
+    rm, rv = torch.zeros(C, device=DEV), torch.ones(C, device=DEV)
+
+    xb = fb.nchw_to_nhwc_padded(x).requires_grad_(True)
+    y = fr.batchnorm_bf16(xb, gamma, beta, rm, rv, training=True, relu=True)
+    dy = randn(N, C, H, W, seed=23)
+    y.backward(fb.nchw_to_nhwc_padded(dy, cp=16))
+
+    q = lambda t: t.to(torch.bfloat16).float()
+    xr = q(x).requires_grad_(True)
+    gr = gamma.detach().clone().requires_grad_(True)
+    br = beta.detach().clone().requires_grad_(True)
+    rm2, rv2 = torch.zeros(C, device=DEV), torch.ones(C, device=DEV)
+    ref = F.relu(F.batch_norm(xr, rm2, rv2, gr, br, training=True,
+                              momentum=0.1, eps=1e-5))
+    ref.backward(q(dy))
+
+    torch.testing.assert_close(fb.nhwc_to_nchw(xb.grad, C), xr.grad,
+                               rtol=5e-2, atol=5e-2)
+    torch.testing.assert_close(gamma.grad, gr.grad, rtol=3e-2, atol=3e-2)
+    torch.testing.assert_close(beta.grad, br.grad, rtol=3e-2, atol=3e-2)
