@@ -1003,94 +1003,76 @@ __global__ __launch_bounds__(256) void conv_pool_stem_bf16_k(
     unsigned char* __restrict__ idx,
     int CIp, int H, int W, int COp, int CO,
     int OH2, int OW2, int N) {
-  // One block per image: the whole (CI=1) input image is tiny (28x28x2B =
-  // 1.6 KB), so it stages into LDS once and the per-cell conv taps become
-  // 2-cycle LDS reads instead of 36 scalar global loads per thread (the
-  // global-load ISSUE rate was the kernel's bound).
-  extern __shared__ float lds_wf[];                   // [COp][CI*KH*KW]
-  __shared__ bf16 lds_img[2048];                      // CI*H*W <= 2048
+  // weights in LDS, fp32: [co][CI*KH*KW]
+  extern __shared__ float lds_wf[];
   const int tid = threadIdx.x;
   const int Kc = CI * KH * KW;
   for (int i = tid; i < COp * Kc; i += 256) {
     int co = i / Kc, k = i % Kc;
     lds_wf[i] = (co < CO) ? w[co * Kc + k] : 0.f;
   }
+  __syncthreads();
 
-  constexpr int TH = KH + 1, TW = KW + 1;
-  const int HWci = H * W * CI;
-  for (int n = blockIdx.x; n < N; n += gridDim.x) {
-    __syncthreads();
-    // stage the image (channel-packed NHWC, CIp stride == CI assumed when
-    // CI==CIp; otherwise gather the real channels)
-    if (CI == CIp) {
-      const unsigned long long* src =
-          reinterpret_cast<const unsigned long long*>(x + (long)n * HWci);
-      unsigned long long* dst = reinterpret_cast<unsigned long long*>(lds_img);
-      for (int i = tid; i < HWci / 4; i += 256) dst[i] = src[i];
-      for (int i = (HWci / 4) * 4 + tid; i < HWci; i += 256)
-        lds_img[i] = x[(long)n * HWci + i];
-    } else {
-      for (int i = tid; i < HWci; i += 256) {
-        int ci = i % CI;
-        int pix = i / CI;
-        lds_img[i] = x[((long)n * H * W + pix) * CIp + ci];
+  long total = (long)N * OH2 * OW2 * (COp / 8);
+  for (long t = (long)blockIdx.x * blockDim.x + tid; t < total;
+       t += (long)gridDim.x * blockDim.x) {
+    int cg = t % (COp / 8);
+    long r = t / (COp / 8);
+    int ow2 = r % OW2;
+    long r2 = r / OW2;
+    int oh2 = r2 % OH2;
+    int n = r2 / OH2;
+
+    // x tile (KH+1) x (KW+1) x CI in registers; all loop bounds are
+    // compile-time so xt stays in VGPRs (guide rule 20: runtime-indexed
+    // arrays spill to scratch)
+    constexpr int TH = KH + 1, TW = KW + 1;
+    float xt[CI * TH * TW];
+    const bf16* xb = x + ((long)(n * H + oh2 * 2) * W + ow2 * 2) * CIp;
+#pragma unroll
+    for (int ci = 0; ci < CI; ++ci)
+#pragma unroll
+      for (int ty = 0; ty < TH; ++ty)
+#pragma unroll
+        for (int tx = 0; tx < TW; ++tx)
+          xt[(ci * TH + ty) * TW + tx] = to_f32(xb[(ty * W + tx) * CIp + ci]);
+
+    bf16x8 pout;
+    unsigned char am[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int co = cg * 8 + j;
+      float acc[4];
+      float bval = (co < CO) ? bias[co] : 0.f;
+      acc[0] = acc[1] = acc[2] = acc[3] = bval;
+      if (co < CO) {
+        const float* wr = &lds_wf[co * Kc];
+#pragma unroll
+        for (int ci = 0; ci < CI; ++ci)
+#pragma unroll
+          for (int kh = 0; kh < KH; ++kh)
+#pragma unroll
+            for (int kw = 0; kw < KW; ++kw) {
+              float wv = wr[(ci * KH + kh) * KW + kw];
+              const float* xrow = &xt[(ci * TH + kh) * TW + kw];
+              acc[0] += wv * xrow[0];
+              acc[1] += wv * xrow[1];
+              acc[2] += wv * xrow[TW];
+              acc[3] += wv * xrow[TW + 1];
+            }
       }
+      float m01 = fmaxf(acc[0], acc[1]), m23 = fmaxf(acc[2], acc[3]);
+      float mx = fmaxf(m01, m23);
+      unsigned char a;
+      if (mx == acc[0]) a = 0; else if (mx == acc[1]) a = 1;
+      else if (mx == acc[2]) a = 2; else a = 3;
+      pout[j] = to_bf16((co < CO) ? fmaxf(mx, 0.f) : 0.f);
+      am[j] = a;
     }
-    __syncthreads();
-
-    int units = OH2 * OW2 * (COp / 8);
-    for (int t = tid; t < units; t += 256) {
-      int cg = t % (COp / 8);
-      int r = t / (COp / 8);
-      int ow2 = r % OW2, oh2 = r / OW2;
-
-      float xt[CI * TH * TW];
-#pragma unroll
-      for (int ci = 0; ci < CI; ++ci)
-#pragma unroll
-        for (int ty = 0; ty < TH; ++ty)
-#pragma unroll
-          for (int tx = 0; tx < TW; ++tx)
-            xt[(ci * TH + ty) * TW + tx] =
-                to_f32(lds_img[((oh2 * 2 + ty) * W + ow2 * 2 + tx) * CI + ci]);
-
-      bf16x8 pout;
-      unsigned char am[8];
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        int co = cg * 8 + j;
-        float acc[4];
-        float bval = (co < CO) ? bias[co] : 0.f;
-        acc[0] = acc[1] = acc[2] = acc[3] = bval;
-        if (co < CO) {
-          const float* wr = &lds_wf[co * Kc];
-#pragma unroll
-          for (int ci = 0; ci < CI; ++ci)
-#pragma unroll
-            for (int kh = 0; kh < KH; ++kh)
-#pragma unroll
-              for (int kw = 0; kw < KW; ++kw) {
-                float wv = wr[(ci * KH + kh) * KW + kw];
-                const float* xrow = &xt[(ci * TH + kh) * TW + kw];
-                acc[0] += wv * xrow[0];
-                acc[1] += wv * xrow[1];
-                acc[2] += wv * xrow[TW];
-                acc[3] += wv * xrow[TW + 1];
-              }
-        }
-        float m01 = fmaxf(acc[0], acc[1]), m23 = fmaxf(acc[2], acc[3]);
-        float mx = fmaxf(m01, m23);
-        unsigned char a;
-        if (mx == acc[0]) a = 0; else if (mx == acc[1]) a = 1;
-        else if (mx == acc[2]) a = 2; else a = 3;
-        pout[j] = to_bf16((co < CO) ? fmaxf(mx, 0.f) : 0.f);
-        am[j] = a;
-      }
-      long off = (((long)(n * OH2 + oh2) * OW2 + ow2) * COp) + cg * 8;
-      *reinterpret_cast<bf16x8*>(p + off) = pout;
-      *reinterpret_cast<unsigned long long*>(idx + off) =
-          *reinterpret_cast<unsigned long long*>(am);
-    }
+    long off = (((long)(n * OH2 + oh2) * OW2 + ow2) * COp) + cg * 8;
+    *reinterpret_cast<bf16x8*>(p + off) = pout;
+    *reinterpret_cast<unsigned long long*>(idx + off) =
+        *reinterpret_cast<unsigned long long*>(am);
   }
 }
 
