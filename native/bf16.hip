@@ -1248,19 +1248,17 @@ __global__ void zero_bf16_k(bf16* __restrict__ p, long n8) {
     *reinterpret_cast<bf16x8*>(p + i * 8) = bf8_zero();
 }
 
-template <int KH, bool HAS_BIAS, bool RELU, int NW, int MW, bool MWN = false>
+template <int KH, bool HAS_BIAS, bool RELU, int NW, int MW, int NIMG = 1>
 __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
     const bf16* __restrict__ x, const bf16* __restrict__ w_rp,
     const float* __restrict__ bias, bf16* __restrict__ y,
     int CIp, int H, int W, int COp, int COcan,
     int OH, int OW, int pad, int N, int tiles_h, int tiles_w) {
-  // output tile: 8 rows x (8*MW) cols; wave owns MW 16-pixel fragments at
-  // the same rows, so every B-fragment read feeds MW MFMAs. MWN variant
-  // (small maps, e.g. ResNet layer4's 7x7): the MW fragments are the SAME
-  // pixels of MW consecutive IMAGES (region staged per image) — identical
-  // B-reuse without needing a 16-wide output row.
-  constexpr int TH = 8, TW = MWN ? 8 : 8 * MW;
-  constexpr int NIMG = MWN ? MW : 1;
+  // output tile: 8 rows x (8*MW/NIMG) cols x NIMG images; a wave owns MW
+  // 16-pixel fragments (split across columns and images), so every
+  // B-fragment read feeds MW MFMAs. NIMG>1 keeps the reuse on small maps
+  // (ResNet layer3/4) where wide column tiles would be mostly padding.
+  constexpr int TH = 8, TW = 8 * (MW / NIMG);
   constexpr int RH = TH + KH - 1, RW = TW + KH - 1;
   const int tid = threadIdx.x;
   const int wid = tid >> 6;
@@ -1320,8 +1318,8 @@ __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
                        : bf8_zero();
 #pragma unroll
         for (int mi = 0; mi < MW; ++mi) {
-          int px = (MWN ? 0 : mi * 8) + (lm & 7);
-          int img = MWN ? mi : 0;
+          int px = (mi % (MW / NIMG)) * 8 + (lm & 7);
+          int img = mi / (MW / NIMG);
           const bf16* xpix =
               &lds_x[(img * RH * RW + (py + kh) * RW + (px + kw)) * ccp];
           bf16x8 a = gin ? *reinterpret_cast<const bf16x8*>(&xpix[ks + lk * 8])
@@ -1341,8 +1339,8 @@ __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
     for (int reg = 0; reg < 4; ++reg) {
       int mo = lk * 4 + reg;                 // 0..15 within fragment
       int oy = oh0 + wid * 2 + (mo >> 3);
-      int ox = ow0 + (MWN ? 0 : mi * 8) + (mo & 7);
-      int nimg = n + (MWN ? mi : 0);
+      int ox = ow0 + (mi % (MW / NIMG)) * 8 + (mo & 7);
+      int nimg = n + mi / (MW / NIMG);
       if (oy >= OH || ox >= OW || nimg >= N) continue;
       long yb = ((long)(nimg * OH + oy) * OW + ox) * COp;
 #pragma unroll
