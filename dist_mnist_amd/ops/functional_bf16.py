@@ -62,11 +62,9 @@ def _use_tiled(x_cip, w, stride, dgrad=False):
     544-686 gather at 64-512ch; 154 vs 172 for LeNet's 5x5) and 3x3 s1
     forwards from 64 channels (331-530 vs 415-620). Thin-channel forwards
     (LeNet conv2 5x5: 107 gather vs 122 tiled) stay on the gather kernel."""
-    if not (stride == 1 and w.shape[2] == w.shape[3] and w.shape[2] in (3, 5)):
-        return False
-    if dgrad:
-        return True
-    return w.shape[2] == 3 and x_cip >= 64
+    if not (stride == 1 and w.shape[2] == w.shape[3] and w.shape[2] == 3):
+        return False  # 5x5 (LeNet conv2): gather wins both directions now
+    return True if dgrad else x_cip >= 64
 
 
 class _Conv2dBF16Fn(torch.autograd.Function):
