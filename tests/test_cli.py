@@ -51,3 +51,28 @@ def test_cli_two_rank_gloo(tmp_path, free_port):
     for rank in range(2):
         assert f"Rank  {rank} , epoch  0" in outs[rank][0] or \
             re.search(rf"Rank\s+{rank}\s*,\s*epoch\s+0", outs[rank][0]), outs[rank][0]
+
+
+@pytest.mark.timeout(300)
+def test_cli_resume(tmp_path, free_port):
+    """--resume loads a prior checkpoint and continues (framework extension
+    on the reference's save-only surface, SURVEY.md §5 checkpoint/resume)."""
+    path = str(tmp_path / "m.pt")
+    env = dict(os.environ, MASTER_ADDR="127.0.0.1", MASTER_PORT=str(free_port),
+               RANK="0", WORLD_SIZE="1")
+    a = subprocess.run(
+        [sys.executable, os.path.join(REPO, "dist_mnist.py"), "--backend", "gloo",
+         "--epochs", "1", "--batch-size", "32", "--dataset-size", "64",
+         "--modelpath", path, "--no-cuda"],
+        env=env, capture_output=True, text=True, timeout=280, cwd=REPO)
+    assert a.returncode == 0, a.stderr[-1500:]
+    env["MASTER_PORT"] = str(free_port + 1)
+    b = subprocess.run(
+        [sys.executable, os.path.join(REPO, "dist_mnist.py"), "--backend", "gloo",
+         "--epochs", "1", "--batch-size", "32", "--dataset-size", "64",
+         "--resume", path, "--modelpath", path + ".2", "--no-cuda"],
+        env=env, capture_output=True, text=True, timeout=280, cwd=REPO)
+    assert b.returncode == 0, b.stderr[-1500:]
+    sd1 = torch.load(path, map_location="cpu", weights_only=True)
+    sd2 = torch.load(path + ".2", map_location="cpu", weights_only=True)
+    assert not torch.equal(sd1["fc2.weight"], sd2["fc2.weight"])  # training continued
