@@ -63,12 +63,17 @@ class GraphedStep:
             self._eager_opt(scale=True)
 
         # warmup steps ran WITHOUT gradient sync (hooks disabled, rank-local
-        # data) so ranks drifted slightly; re-broadcast params in place —
-        # the captured graphs hold raw pointers, so an in-place copy is safe.
+        # data) so ranks drifted; re-broadcast params AND optimizer momentum
+        # in place — the captured graphs hold raw pointers, so in-place
+        # copies are safe and keep ranks bitwise-identical from here on.
         if self.ddp is not None and self.ddp.enabled:
             with torch.no_grad():
                 for p in self.ddp.module.parameters():
                     dist.broadcast(p.data, src=0, group=self.ddp.pg)
+                for st in self.opt.state.values():
+                    buf = st.get("momentum_buffer")
+                    if buf is not None:
+                        dist.broadcast(buf, src=0, group=self.ddp.pg)
             torch.cuda.synchronize(self.device)
 
     # -- eager pieces (also used during warmup/capture) ---------------------
