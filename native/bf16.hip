@@ -1087,7 +1087,12 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_pool_bf16_k(
     const unsigned char* __restrict__ pidx, const bf16* __restrict__ x,
     float* __restrict__ dw_ws, float* __restrict__ db_ws,
     int CIp, int CI, int H, int W, int COp, int CO, int KH, int KW,
-    int OH, int OW, int M, int K, int Kcan, int m_per_split) {
+    int OH, int OW, int M, int K, int Kcan, int m_per_split, long x_elems) {
+  // k-space uses PER-kh-ROW padding: KROW = ceil(KW*CIp/8)*8, K = KH*KROW,
+  // so every 8-k A-fragment stays inside one kh row of the patch even when
+  // KW*CIp is not a multiple of 8 (the stem's CIp can be 1). Pad k entries
+  // read in-row garbage (load base clamped to the tensor) and their dw rows
+  // are simply never written back.
   constexpr int MC = 64;
   constexpr int NKF = 4 * KT;
   const int k0 = blockIdx.x * (NKF * 16);
@@ -1104,7 +1109,8 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_pool_bf16_k(
   __shared__ bf16 lds_dy[MC][16];
   __shared__ int lds_mbase[MC];
 
-  const int KWC = KW * CIp;
+  const int KROWV = KW * CIp;                 // valid entries per kh row
+  const int KROW = ((KROWV + 7) / 8) * 8;     // padded row length
   const int OHOW = OH * OW;
 
   int g_off[NKF * 2];
@@ -1112,9 +1118,8 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_pool_bf16_k(
   for (int g = 0; g < NKF * 2; ++g) {
     int k = k0 + g * 8;
     int kcl = k < K ? k : 0;
-    int kh = kcl / KWC, kr = kcl % KWC;
-    int kw = kr / CIp, ci0 = kr % CIp;
-    g_off[g] = (kh * W + kw) * CIp + ci0;
+    int kh = kcl / KROW, r = kcl % KROW;
+    g_off[g] = kh * W * CIp + r;              // r may run past the real row
   }
 
   f32x4 acc[KT] = {};
@@ -1142,11 +1147,13 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_pool_bf16_k(
       int m = mc + mloc;
       int k = k0 + g * 8;
       bf16x8 v = bf8_zero();
-      if (m < m_end && k < K)
-        v = *reinterpret_cast<const bf16x8*>(&x[lds_mbase[mloc] + g_off[g]]);
+      if (m < m_end && k < K) {
+        long a = (long)lds_mbase[mloc] + g_off[g];
+        if (a + 8 > x_elems) a = x_elems - 8;  // clamp: pad rows are discarded
+        v = *reinterpret_cast<const bf16x8*>(&x[a]);
+      }
       *reinterpret_cast<bf16x8*>(&lds_x[kf][mloc][half * 8]) = v;
     }
-    // dy_conv reconstruction, vectorized over 8 co
     for (int idx2 = tid; idx2 < MC * 2; idx2 += 256) {
       int half = idx2 & 1;
       int mloc = idx2 >> 1;
@@ -1195,10 +1202,12 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_pool_bf16_k(
         int kk = k0 + (wid * KT + kf) * 16 + lk * 4 + reg;
         int co = co0 + lm;
         if (kk < K && co < CO) {
-          int kh = kk / KWC, r = kk % KWC;
-          int kw = r / CIp, ci = r % CIp;
-          if (ci < CI)
-            slab[((co * CI) + ci) * (KH * KW) + kh * KW + kw] = acc[kf][reg];
+          int kh = kk / KROW, r = kk % KROW;
+          if (r < KROWV) {                     // pad k rows never written
+            int kw = r / CIp, ci = r % CIp;
+            if (ci < CI)
+              slab[((co * CI) + ci) * (KH * KW) + kh * KW + kw] = acc[kf][reg];
+          }
         }
       }
     }

@@ -327,3 +327,38 @@ def test_graphed_step_replay():
     # check successive deltas are not all identical signs/magnitudes)
     diffs = {round(losses[i + 1] - losses[i], 6) for i in range(10, 19)}
     assert len(diffs) > 1
+
+
+def test_conv_pool_stem_unpadded_input():
+    """REGRESSION (bench path): the stem takes CIp=1 input ([N,28,28,1] —
+    what the loader actually ships); its wgrad must handle KW*CIp=5 rows
+    that are not 8-aligned (row-padded k-space + clamped loads)."""
+    N, CO = 5, 10
+    x = randn(N, 1, 28, 28, seed=70)
+    w = randn(CO, 1, 5, 5, seed=71, scale=0.3).requires_grad_(True)
+    b = randn(CO, seed=72).requires_grad_(True)
+    xb = x[:, 0].to(torch.bfloat16).unsqueeze(-1).contiguous()  # [N,28,28,1]
+    p = fb.conv_pool_stem_bf16(xb, w, b, cop=16)
+
+    q = lambda t: t.to(torch.bfloat16).float()
+    ref_fwd = F.relu(F.max_pool2d(F.conv2d(q(x), q(w.detach()), b.detach()), 2))
+    torch.testing.assert_close(fb.nhwc_to_nchw(p, CO), ref_fwd, rtol=2e-2, atol=3e-2)
+
+    dy = randn(N, CO, 12, 12, seed=73)
+    p.backward(fb.nchw_to_nhwc_padded(dy, cp=16))
+
+    # oracle follows the kernel's argmax routing (see test_conv_pool_stem_fused)
+    pk, idxk = native.ext().conv_pool_stem_bf16(xb, w.detach(), b.detach(), 16)
+    gate = (pk.float() > 0).to(torch.bfloat16)
+    dyq = (fb.nchw_to_nhwc_padded(dy, cp=16) * gate).float()
+    dy_conv = torch.zeros(N, 24, 24, 16, device=DEV)
+    for qd in range(4):
+        sel = (idxk == qd).float() * dyq
+        dy_conv[:, qd // 2::2, qd % 2::2, :] = sel
+    dy_conv_nchw = dy_conv[..., :CO].permute(0, 3, 1, 2)
+    dw_ref = torch.nn.grad.conv2d_weight(q(x), w.shape, q(dy_conv_nchw))
+    db_ref = q(dy_conv_nchw).sum(dim=(0, 2, 3))
+    torch.testing.assert_close(w.grad, dw_ref, rtol=2e-2,
+                               atol=2e-2 * max(1.0, dw_ref.abs().max().item()))
+    torch.testing.assert_close(b.grad, db_ref, rtol=2e-2,
+                               atol=2e-2 * max(1.0, db_ref.abs().max().item()))
