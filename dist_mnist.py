@@ -111,7 +111,6 @@ def run(rank: int, world_size: int, args):
         model = DistributedDataParallel(model, bucket_bytes=int(args.bucket_mb * 1024 * 1024))
     optimizer = SGD(model.parameters(), lr=args.lr, momentum=args.momentum)
 
-    num_batches = len(loader.dataset) // bsz if bsz else len(loader)
     import time as _time
     for epoch in range(args.epochs):
         model.train()

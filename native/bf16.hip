@@ -1251,12 +1251,6 @@ __global__ void repack_w_k(const float* __restrict__ w, bf16* __restrict__ w_rp,
   }
 }
 
-__global__ void zero_bf16_k(bf16* __restrict__ p, long n8) {
-  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n8;
-       i += (long)gridDim.x * blockDim.x)
-    *reinterpret_cast<bf16x8*>(p + i * 8) = bf8_zero();
-}
-
 template <int KH, bool HAS_BIAS, bool RELU, int NW, int MW, int NIMG = 1>
 __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
     const bf16* __restrict__ x, const bf16* __restrict__ w_rp,
