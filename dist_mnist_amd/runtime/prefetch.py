@@ -54,7 +54,13 @@ class Prefetcher:
         if item is None:
             raise StopIteration
         # compute stream waits for this slot's copy; no host sync
-        torch.cuda.current_stream(self.device).wait_event(self.events[slot])
+        cur = torch.cuda.current_stream(self.device)
+        cur.wait_event(self.events[slot])
+        # the tensors were allocated on the copy stream: mark them as used by
+        # the compute stream so the caching allocator cannot hand their
+        # memory to a later H2D while this stream still reads them
+        item[0].record_stream(cur)
+        item[1].record_stream(cur)
         self.idx += 1
         self._preload()  # kick off the next copy on the side stream
         return item
