@@ -253,7 +253,17 @@ def log_softmax(x):
 # as u8. Channelwise=True zeroes whole (n, c) feature maps (Dropout2d).
 # ---------------------------------------------------------------------------
 
-_dropout_counter = [0]
+_dropout_ctr_dev: dict = {}
+
+
+def _dropout_ctr(device):
+    """Device-side RNG stream counter (shared scheme with the bf16 path):
+    the bump kernel runs on-stream, so hipGraph replays of a captured step
+    still draw fresh masks."""
+    key = str(device)
+    if key not in _dropout_ctr_dev:
+        _dropout_ctr_dev[key] = torch.zeros(1, dtype=torch.int64, device=device)
+    return _dropout_ctr_dev[key]
 
 
 class _DropoutFn(torch.autograd.Function):
@@ -263,8 +273,8 @@ class _DropoutFn(torch.autograd.Function):
             ctx.p = 0.0
             return x
         if x.is_cuda:
-            _dropout_counter[0] += 1
-            y, mask = native.ext().dropout_fwd(_nc(x), p, bool(channelwise), int(seed), _dropout_counter[0])
+            y, mask = native.ext().dropout_fwd(_nc(x), p, bool(channelwise), int(seed),
+                                               _dropout_ctr(x.device))
         else:
             if channelwise:
                 keep = (torch.rand(x.shape[0], x.shape[1], 1, 1, device=x.device) >= p).to(x.dtype)

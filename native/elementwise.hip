@@ -157,10 +157,16 @@ __global__ void logsoftmax_bwd_k(const float* __restrict__ dy,
 // Dropout — counter-based RNG (no state tensors), u8 mask saved for bwd.
 // channelwise (Dropout2d): one decision per (n, c), mask shaped [N*C].
 // ---------------------------------------------------------------------------
+__global__ void bump_u64_f32path_k(unsigned long long* ctr) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) *ctr += 1;
+}
+
 __global__ void dropout_fwd_k(const float* __restrict__ x, float* __restrict__ y,
                               unsigned char* __restrict__ mask, long n,
                               long hw /* 1 for elementwise */, float p,
-                              unsigned long long seed, unsigned long long ctr) {
+                              unsigned long long seed,
+                              const unsigned long long* __restrict__ ctr_p) {
+  unsigned long long ctr = *ctr_p;
   float inv = 1.f / (1.f - p);
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (long)gridDim.x * blockDim.x) {

@@ -329,7 +329,8 @@ at::Tensor logsoftmax_bwd(at::Tensor dy, at::Tensor logp) {
 
 // ---------------------------------------------------------------------------
 std::vector<at::Tensor> dropout_fwd(at::Tensor x, double p, bool channelwise,
-                                    long seed, long ctr) {
+                                    long seed, at::Tensor ctr) {
+  CHECK_I64(ctr);
   CHECK_F32(x);
   long n = x.numel();
   long hw = 1;
@@ -342,10 +343,13 @@ std::vector<at::Tensor> dropout_fwd(at::Tensor x, double p, bool channelwise,
   auto y = at::empty_like(x);
   auto mask = at::empty({mask_n}, x.options().dtype(at::kByte));
   int blocks = (int)std::min<long>(2048, cdiv(n, 256));
+  hipLaunchKernelGGL(bump_u64_f32path_k, dim3(1), dim3(64), 0, cur_stream(),
+                     reinterpret_cast<unsigned long long*>(ctr.data_ptr<long>()));
   hipLaunchKernelGGL(dropout_fwd_k, dim3(blocks), dim3(256), 0, cur_stream(),
                      x.data_ptr<float>(), y.data_ptr<float>(),
                      mask.data_ptr<unsigned char>(), n, hw, (float)p,
-                     (unsigned long long)seed, (unsigned long long)ctr);
+                     (unsigned long long)seed,
+                     reinterpret_cast<const unsigned long long*>(ctr.data_ptr<long>()));
   return {y, mask};
 }
 
