@@ -31,7 +31,9 @@ def _case(rng):
 
 @pytest.mark.timeout(600)
 def test_conv_bf16_shape_fuzz():
-    rng = random.Random(20260913)
+    import os
+
+    rng = random.Random(int(os.environ.get("FUZZ_SEED", 20260913)))
     torch.manual_seed(0)
     for trial in range(24):
         n, ci, h, w, co, kh, stride, pad = _case(rng)
