@@ -98,18 +98,9 @@ class _Conv2dBF16Fn(torch.autograd.Function):
                 dx = m.conv2d_dgrad_bf16(dy, w, x.shape[1], x.shape[2],
                                          x.shape[3], stride, pad)
         if ctx.needs_input_grad[1] or (has_bias and ctx.needs_input_grad[2]):
-            if (not has_bias and stride == 1 and w.shape[2] == 3
-                    and w.shape[3] == 3 and x.shape[3] % 64 == 0
-                    and dy.shape[3] % 32 == 0):
-                # region-stationary wgrad (ResNet body): one x-region stage
-                # per pixel tile instead of a per-(k-tile,co-tile) re-gather
-                dw = m.conv_wgrad_region3_bf16(dy, x.contiguous(), w.shape[1],
-                                               w.shape[0], pad)
-                db = None
-            else:
-                dw, db = m.conv2d_wgrad_bf16(dy, x.contiguous(), w.shape[1],
-                                             w.shape[0], w.shape[2], w.shape[3],
-                                             stride, pad, has_bias)
+            dw, db = m.conv2d_wgrad_bf16(dy, x.contiguous(), w.shape[1],
+                                         w.shape[0], w.shape[2], w.shape[3],
+                                         stride, pad, has_bias)
         if not has_bias:
             db = None
         return dx, dw, db, None, None, None, None

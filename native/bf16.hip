@@ -1534,134 +1534,3 @@ __global__ __launch_bounds__(256) void conv2d_dgrad_s2_bf16_k(
     }
   }
 }
-
-// ---------------------------------------------------------------------------
-// Region-stationary wgrad, stride 1, 3x3 (ResNet body). The generic wgrad
-// re-gathers im2col slices from L2/HBM once per (k-tile x co-tile) — ~9x tap
-// amplification (measured 9% MFMA-busy). Here a block owns the FULL tap grid
-// for one (64-ci chunk, NW*16-co tile): per 8x8 output-pixel tile it stages
-// the x region ([10][10][cc], NHWC vec8, aligned) and the dy tile (tr-layout
-// images) ONCE, then accumulates all 9 taps into registers
-// (acc[9 tap-frags][NW]); partials go to slabs, reduce_splits folds them.
-// A-fragments (8 m at fixed (tap, ci)) are built from the region with
-// per-element LDS reads (the 3x3 shifts misalign vector reads), which still
-// beats the global re-gather by the tap factor.
-// Wave w owns tap-frags rf = w*9..w*9+8; rf -> (tap = rf/4, ci16 = rf%4)
-// over the 36 frags of (9 taps x 64 ci).
-// ---------------------------------------------------------------------------
-template <int NW>
-__global__ __launch_bounds__(256) void conv_wgrad_region3_bf16_k(
-    const bf16* __restrict__ dy, const bf16* __restrict__ x,
-    float* __restrict__ dw_ws, float* __restrict__ db_ws,
-    int CIp, int CI, int H, int W, int COp, int CO,
-    int OH, int OW, int pad, int N, int tiles_h, int tiles_w,
-    int n_splits, int Kcan) {
-  constexpr int TS = 8, R = TS + 2;       // KH = 3
-  const int cic = blockIdx.x * 64;        // ci chunk base
-  const int co0 = blockIdx.y * (NW * 16);
-  const int ms = blockIdx.z;
-  const int tid = threadIdx.x;
-  const int wid = tid >> 6;
-  const int l = tid & 63;
-  const int lm = l & 15;
-  const int lk = l >> 4;
-  const int cc = min(64, CIp - cic);
-
-  __shared__ bf16 lds_x[R * R * 72 + 8];  // [ry][rx][ccp], ccp = cc+8
-  __shared__ bf16 lds_dy[NW][64][16];     // tr-layout per co 16-group
-
-  const int ccp = cc + 8;
-  // this wave's 9 tap-frags: rf = wid*9 + f; tap = rf/4; ci16 = rf%4
-  // (ResNet convs carry no bias, so there is no db path here)
-  f32x4 acc[9][NW] = {};
-
-  const long total_tiles = (long)N * tiles_h * tiles_w;
-  for (long t = ms; t < total_tiles; t += n_splits) {
-    int tw = (int)(t % tiles_w);
-    long r1 = t / tiles_w;
-    int th = (int)(r1 % tiles_h);
-    int n = (int)(r1 / tiles_h);
-    const int oh0 = th * TS, ow0 = tw * TS;
-
-    __syncthreads();
-    // stage x region (aligned vec8 over ci)
-    for (int idx = tid; idx < R * R * (cc / 8); idx += 256) {
-      int g8 = idx % (cc / 8);
-      int p = idx / (cc / 8);
-      int rx = p % R, ry = p / R;
-      int ih = oh0 - pad + ry, iw = ow0 - pad + rx;
-      bf16x8 v = bf8_zero();
-      if (ih >= 0 && ih < H && iw >= 0 && iw < W)
-        v = *reinterpret_cast<const bf16x8*>(
-            &x[((long)(n * H + ih) * W + iw) * CIp + cic + g8 * 8]);
-      *reinterpret_cast<bf16x8*>(&lds_x[(ry * R + rx) * ccp + g8 * 8]) = v;
-    }
-    // stage dy tile into tr-layout images ([m][16co] rows)
-    for (int idx = tid; idx < NW * 64 * 2; idx += 256) {
-      int half = idx & 1;
-      int rest = idx >> 1;
-      int mloc = rest % 64;
-      int ni = rest / 64;
-      int oy = oh0 + (mloc >> 3), ox = ow0 + (mloc & 7);
-      bf16x8 v = bf8_zero();
-      int cb = co0 + ni * 16 + half * 8;
-      if (oy < OH && ox < OW && cb < COp)
-        v = *reinterpret_cast<const bf16x8*>(
-            &dy[((long)(n * OH + oy) * OW + ox) * COp + cb]);
-      *reinterpret_cast<bf16x8*>(&lds_dy[ni][mloc][half * 8]) = v;
-    }
-    __syncthreads();
-
-#pragma unroll
-    for (int msb = 0; msb < 64; msb += 32) {
-      bf16x8 b[NW];
-#pragma unroll
-      for (int ni = 0; ni < NW; ++ni)
-        b[ni] = tr_read_8m(&lds_dy[ni][msb][0], l);
-#pragma unroll
-      for (int f = 0; f < 9; ++f) {
-        int rf = wid * 9 + f;
-        int tap = rf >> 2, ci16 = rf & 3;
-        int kh = tap / 3, kw = tap % 3;
-        int ci = ci16 * 16 + lm;
-        // A-frag: 8 m (one pixel row pair) at fixed (tap, ci)
-        bf16x8 a;
-        bool civ = ci < cc;
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          int m = msb + lk * 8 + j;
-          int py = m >> 3, px = m & 7;
-          a[j] = civ ? lds_x[((py + kh) * R + (px + kw)) * ccp + ci]
-                     : to_bf16(0.f);
-        }
-#pragma unroll
-        for (int ni = 0; ni < NW; ++ni)
-          acc[f][ni] = mfma16x16x32bf(a, b[ni], acc[f][ni]);
-      }
-    }
-  }
-
-  // fold: each wave owns DISTINCT k rows (tap-frags), so no cross-wave fold
-  // is needed for dw; write this split's partial slab directly.
-  {
-    float* slab = dw_ws + (long)ms * (CO * (long)Kcan);
-#pragma unroll
-    for (int f = 0; f < 9; ++f) {
-      int rf = wid * 9 + f;
-      int tap = rf >> 2, ci16 = rf & 3;
-      int kh = tap / 3, kw = tap % 3;
-#pragma unroll
-      for (int reg = 0; reg < 4; ++reg) {
-        int ci = cic + ci16 * 16 + lk * 4 + reg;
-        if (ci >= CI) continue;
-#pragma unroll
-        for (int ni = 0; ni < NW; ++ni) {
-          int co = co0 + ni * 16 + lm;
-          if (co < CO)
-            slab[((long)co * CI + ci) * 9 + kh * 3 + kw] = acc[f][ni][reg];
-        }
-      }
-    }
-  }
-  (void)db_ws;
-}

@@ -409,7 +409,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_wgrad_bf16", &conv2d_wgrad_bf16);
   m.def("conv_pool_stem_bf16", &conv_pool_stem_bf16);
   m.def("conv2d_wgrad_pool_bf16", &conv2d_wgrad_pool_bf16);
-  m.def("conv_wgrad_region3_bf16", &conv_wgrad_region3_bf16);
   m.def("linear_fwd_bf16", &linear_fwd_bf16);
   m.def("linear_dx_bf16", &linear_dx_bf16);
   m.def("linear_dw_bf16", &linear_dw_bf16);

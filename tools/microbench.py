@@ -41,7 +41,7 @@ def main():
         ("rn.l4.3x3", 256, 7, 7, 512, 512, 512, 512, 3, 1, 1),
     ]
     print(f"{'shape':14} {'fwd_gather':>11} {'fwd_tiled':>10} {'dg_gather':>10} "
-          f"{'dg_tiled':>9} {'wgrad':>8} {'wg_reg3':>8}  (us/call)")
+          f"{'dg_tiled':>9} {'wgrad':>8}   (us/call)")
     for name, N, H, W, CIp, CIcan, CO, COp, KH, stride, pad in shapes:
         x = torch.randn(N, H, W, CIp, device=dev).to(torch.bfloat16)
         w = torch.randn(CO, CIcan, KH, KH, device=dev) * 0.1
@@ -58,10 +58,7 @@ def main():
             if stride == 1 else float("nan")
         wg = timeit(lambda: ext.conv2d_wgrad_bf16(dy, x, CIcan, CO, KH, KH,
                                                   stride, pad, True))
-        wr = float("nan")
-        if stride == 1 and KH == 3 and CIp % 64 == 0 and COp % 32 == 0:
-            wr = timeit(lambda: ext.conv_wgrad_region3_bf16(dy, x, CIcan, CO, pad))
-        print(f"{name:14} {fg:11.1f} {ft:10.1f} {dg:10.1f} {dt:9.1f} {wg:8.1f} {wr:8.1f}")
+        print(f"{name:14} {fg:11.1f} {ft:10.1f} {dg:10.1f} {dt:9.1f} {wg:8.1f}")
 
     # stride-2 shapes (gather only for now)
     s2 = [
