@@ -20,7 +20,7 @@
 // rows => BM = 128 output pixels per block; one 16-wide co tile per block.
 // Weights + k-decode staged in LDS per BK=32 chunk.
 // ---------------------------------------------------------------------------
-template <bool HAS_BIAS, bool RELU, bool BOUNDS>
+template <bool HAS_BIAS, bool RELU, bool BOUNDS, int NW>
 __global__ __launch_bounds__(256) void conv2d_fwd_f32_k(
     const float* __restrict__ x, const float* __restrict__ w,
     const float* __restrict__ bias, float* __restrict__ y,
@@ -28,23 +28,22 @@ __global__ __launch_bounds__(256) void conv2d_fwd_f32_k(
     int stride, int pad, int M, int K) {
   constexpr int BM = 128, BK = 32, MW = 2;
   const int m0 = blockIdx.x * BM;
-  const int co0 = blockIdx.y * 16;
+  const int co0 = blockIdx.y * (NW * 16);
   const int tid = threadIdx.x;
   const int wid = tid >> 6;
   const int l = tid & 63;
   const int lm = l & 15;   // B col (m) / A row (co)
   const int lk = l >> 4;   // k slice 0..3
 
-  __shared__ float lds_w[BK][16];
+  __shared__ float lds_w[BK][NW * 16];
   __shared__ int lds_koff[BK];    // ci*H*W + kh*W + kw
   __shared__ int lds_khw[BK];     // (kh<<16)|kw, only used if BOUNDS
 
   const int HW = H * W;
   const int OHOW = OH * OW;
 
-  // per-(lane, mi) output-pixel decode — fixed across the whole K loop
-  int xbase[MW];    // n*CI*HW + ih0*W + iw0
-  int ybase[MW];    // n*CO*OHOW + oh*OW + ow
+  int xbase[MW];
+  int ybase[MW];
   int ih0[MW], iw0[MW];
   bool mval[MW];
 #pragma unroll
@@ -60,13 +59,12 @@ __global__ __launch_bounds__(256) void conv2d_fwd_f32_k(
     ybase[mi] = n * CO * OHOW + r;
   }
 
-  f32x4 acc[MW] = {};
+  f32x4 acc[MW][NW] = {};
 
   for (int k0 = 0; k0 < K; k0 += BK) {
-    // stage W^T tile [BK][16co] + k decode
     __syncthreads();
-    for (int idx = tid; idx < BK * 16; idx += 256) {
-      int kk = idx >> 4, co = idx & 15;
+    for (int idx = tid; idx < BK * NW * 16; idx += 256) {
+      int kk = idx / (NW * 16), co = idx % (NW * 16);
       int k = k0 + kk;
       lds_w[kk][co] = (k < K && co0 + co < CO) ? w[(co0 + co) * K + k] : 0.f;
     }
@@ -82,8 +80,10 @@ __global__ __launch_bounds__(256) void conv2d_fwd_f32_k(
 
     int klim = min(BK, K - k0);
     for (int kk = 0; kk + 4 <= klim; kk += 4) {
-      float a = lds_w[kk + lk][lm];
       int koff = lds_koff[kk + lk];
+      float a[NW];
+#pragma unroll
+      for (int ni = 0; ni < NW; ++ni) a[ni] = lds_w[kk + lk][ni * 16 + lm];
 #pragma unroll
       for (int mi = 0; mi < MW; ++mi) {
         bool ok = mval[mi];
@@ -94,17 +94,23 @@ __global__ __launch_bounds__(256) void conv2d_fwd_f32_k(
           ok = ok && ih >= 0 && ih < H && iw >= 0 && iw < W;
         }
         float b = ok ? x[xbase[mi] + koff] : 0.f;
-        acc[mi] = mfma16x16x4(a, b, acc[mi]);
+#pragma unroll
+        for (int ni = 0; ni < NW; ++ni)
+          acc[mi][ni] = mfma16x16x4(a[ni], b, acc[mi][ni]);
       }
     }
     int ktail = klim & 3;
     if (ktail) {
-      int kk = klim - ktail;  // process remaining 1..3 slices: lanes with lk < ktail
-      float a = (lk < ktail) ? lds_w[kk + lk][lm] : 0.f;
-      int koff = (lk < ktail) ? lds_koff[kk + lk] : 0;
+      int kk = klim - ktail;
+      bool kin = lk < ktail;
+      int koff = kin ? lds_koff[kk + lk] : 0;
+      float a[NW];
+#pragma unroll
+      for (int ni = 0; ni < NW; ++ni)
+        a[ni] = kin ? lds_w[kk + lk][ni * 16 + lm] : 0.f;
 #pragma unroll
       for (int mi = 0; mi < MW; ++mi) {
-        bool ok = mval[mi] && (lk < ktail);
+        bool ok = mval[mi] && kin;
         if (BOUNDS && ok) {
           int khw = lds_khw[kk + lk];
           int kh = khw >> 16, kw = khw & 0xffff;
@@ -112,23 +118,27 @@ __global__ __launch_bounds__(256) void conv2d_fwd_f32_k(
           ok = ih >= 0 && ih < H && iw >= 0 && iw < W;
         }
         float b = ok ? x[xbase[mi] + koff] : 0.f;
-        acc[mi] = mfma16x16x4(a, b, acc[mi]);
+#pragma unroll
+        for (int ni = 0; ni < NW; ++ni)
+          acc[mi][ni] = mfma16x16x4(a[ni], b, acc[mi][ni]);
       }
     }
   }
 
-  // epilogue: D col = lm -> m (coalesced over ow), row = lk*4+reg -> co
 #pragma unroll
   for (int mi = 0; mi < MW; ++mi) {
     if (!mval[mi]) continue;
 #pragma unroll
-    for (int reg = 0; reg < 4; ++reg) {
-      int co = co0 + lk * 4 + reg;
-      if (co >= CO) continue;
-      float v = acc[mi][reg];
-      if (HAS_BIAS) v += bias[co];
-      if (RELU) v = fmaxf(v, 0.f);
-      y[ybase[mi] + co * OHOW] = v;
+    for (int ni = 0; ni < NW; ++ni) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        int co = co0 + ni * 16 + lk * 4 + reg;
+        if (co >= CO) continue;
+        float v = acc[mi][ni][reg];
+        if (HAS_BIAS) v += bias[co];
+        if (RELU) v = fmaxf(v, 0.f);
+        y[ybase[mi] + co * OHOW] = v;
+      }
     }
   }
 }
@@ -138,13 +148,13 @@ __global__ __launch_bounds__(256) void conv2d_fwd_f32_k(
 // A = flipped weights staged as [k'][ci-tile], B = dy gathered with the
 // transposed-conv index map (zero where oh/ow fall outside or off-stride).
 // ---------------------------------------------------------------------------
-template <bool S1>
+template <bool S1, int MW>
 __global__ __launch_bounds__(256) void conv2d_dgrad_f32_k(
     const float* __restrict__ dy, const float* __restrict__ w,
     float* __restrict__ dx,
     int CI, int H, int W, int CO, int KH, int KW, int OH, int OW,
     int stride, int pad, int Mp, int Kp) {
-  constexpr int BM = 128, BK = 32, MW = 2;
+  constexpr int BM = 4 * MW * 16, BK = 32;
   const int m0 = blockIdx.x * BM;
   const int ci0 = blockIdx.y * 16;
   const int tid = threadIdx.x;

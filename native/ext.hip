@@ -56,16 +56,21 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias, long stride,
   int OW = (W + 2 * pad - KW) / stride + 1;
   auto y = at::empty({N, CO, OH, OW}, x.options());
   int M = N * OH * OW, K = CI * KH * KW;
-  dim3 grid(cdiv(M, 128), cdiv(CO, 16));
+  int nw = (CO > 16) ? 2 : 1;   // multi-co tiles amortize the im2col gather
+  dim3 grid(cdiv(M, 128), cdiv(CO, nw * 16));
   bool bounds = pad > 0;
   auto launch = [&](auto hb, auto rl, auto bd) {
-    hipLaunchKernelGGL((conv2d_fwd_f32_k<decltype(hb)::value, decltype(rl)::value,
-                                         decltype(bd)::value>),
-                       grid, dim3(256), 0, cur_stream(), x.data_ptr<float>(),
-                       w.data_ptr<float>(),
-                       has_bias ? bias.data_ptr<float>() : nullptr,
-                       y.data_ptr<float>(), CI, H, W, CO, KH, KW, OH, OW,
-                       (int)stride, (int)pad, M, K);
+    auto go = [&](auto nwc) {
+      hipLaunchKernelGGL((conv2d_fwd_f32_k<decltype(hb)::value, decltype(rl)::value,
+                                           decltype(bd)::value, decltype(nwc)::value>),
+                         grid, dim3(256), 0, cur_stream(), x.data_ptr<float>(),
+                         w.data_ptr<float>(),
+                         has_bias ? bias.data_ptr<float>() : nullptr,
+                         y.data_ptr<float>(), CI, H, W, CO, KH, KW, OH, OW,
+                         (int)stride, (int)pad, M, K);
+    };
+    if (nw == 2) go(std::integral_constant<int, 2>{});
+    else go(std::integral_constant<int, 1>{});
   };
   using T = std::true_type;
   using F = std::false_type;
@@ -84,17 +89,20 @@ at::Tensor conv2d_bwd_input(at::Tensor dy, at::Tensor w, long H, long W,
   int CI = w.size(1), KH = w.size(2), KW = w.size(3);
   auto dx = at::empty({N, CI, H, W}, dy.options());
   int Mp = N * H * W, Kp = CO * KH * KW;
-  dim3 grid(cdiv(Mp, 128), cdiv(CI, 16));
-  if (stride == 1)
-    hipLaunchKernelGGL((conv2d_dgrad_f32_k<true>), grid, dim3(256), 0,
-                       cur_stream(), dy.data_ptr<float>(), w.data_ptr<float>(),
-                       dx.data_ptr<float>(), CI, H, W, CO, KH, KW, OH, OW,
-                       (int)stride, (int)pad, Mp, Kp);
-  else
-    hipLaunchKernelGGL((conv2d_dgrad_f32_k<false>), grid, dim3(256), 0,
-                       cur_stream(), dy.data_ptr<float>(), w.data_ptr<float>(),
-                       dx.data_ptr<float>(), CI, H, W, CO, KH, KW, OH, OW,
-                       (int)stride, (int)pad, Mp, Kp);
+  int mw = (Mp >= 1 << 16) ? 4 : 2;  // amortize LDS weight/kdec reads
+  dim3 grid(cdiv(Mp, mw * 64), cdiv(CI, 16));
+  auto launch = [&](auto s1c, auto mwc) {
+    hipLaunchKernelGGL((conv2d_dgrad_f32_k<decltype(s1c)::value, decltype(mwc)::value>),
+                       grid, dim3(256), 0, cur_stream(), dy.data_ptr<float>(),
+                       w.data_ptr<float>(), dx.data_ptr<float>(), CI, H, W, CO,
+                       KH, KW, OH, OW, (int)stride, (int)pad, Mp, Kp);
+  };
+  using T = std::true_type;
+  using F = std::false_type;
+  using M2 = std::integral_constant<int, 2>;
+  using M4 = std::integral_constant<int, 4>;
+  if (stride == 1) { if (mw == 4) launch(T{}, M4{}); else launch(T{}, M2{}); }
+  else { if (mw == 4) launch(F{}, M4{}); else launch(F{}, M2{}); }
   return dx;
 }
 
