@@ -192,8 +192,8 @@ std::vector<at::Tensor> linear_bwd(at::Tensor dy, at::Tensor x, at::Tensor w,
   CHECK_F32(w);
   int M = x.size(0), CI = x.size(1), CO = w.size(0);
   auto dx = at::empty({M, CI}, x.options());
-  auto dw = at::zeros({CO, CI}, x.options());
-  auto db = at::zeros({want_bias ? CO : 0}, x.options());
+  auto dw = at::empty({CO, CI}, x.options());
+  auto db = at::empty({want_bias ? CO : 0}, x.options());
   {
     dim3 grid(cdiv(M, 64), cdiv(CI, 64));
     if (relu)
@@ -211,12 +211,14 @@ std::vector<at::Tensor> linear_bwd(at::Tensor dy, at::Tensor x, at::Tensor w,
     int m_per_split = cdiv(cdiv(M, ms), 4) * 4;
     ms = cdiv(M, m_per_split);
     dim3 grid(ct, it, ms);
+    auto dw_ws = at::empty({ms, (long)CO * CI}, x.options());
+    auto db_ws = at::empty({want_bias ? ms : 0, want_bias ? CO : 0}, x.options());
     auto launch = [&](auto rl, auto wb) {
       hipLaunchKernelGGL((linear_dw_f32_k<decltype(rl)::value, decltype(wb)::value>),
                          grid, dim3(256), 0, cur_stream(), dy.data_ptr<float>(),
                          x.data_ptr<float>(), yact.data_ptr<float>(),
-                         dw.data_ptr<float>(),
-                         want_bias ? db.data_ptr<float>() : nullptr, M, CI, CO,
+                         dw_ws.data_ptr<float>(),
+                         want_bias ? db_ws.data_ptr<float>() : nullptr, M, CI, CO,
                          m_per_split);
     };
     using T = std::true_type;
@@ -225,6 +227,8 @@ std::vector<at::Tensor> linear_bwd(at::Tensor dy, at::Tensor x, at::Tensor w,
     else if (relu) launch(T{}, F{});
     else if (want_bias) launch(F{}, T{});
     else launch(F{}, F{});
+    launch_reduce_splits(dw_ws, dw, (long)CO * CI, ms);
+    if (want_bias) launch_reduce_splits(db_ws, db, CO, ms);
   }
   return {dx, dw, db};
 }

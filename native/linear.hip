@@ -7,7 +7,7 @@
 //
 // fwd/dx stage both operand chunks through LDS transposed so every global
 // access group is contiguous; dw needs no LDS at all (both gathers are
-// row-major coalesced), split-M + atomicAdd accumulation.
+// row-major coalesced), split-M partial slabs folded deterministically.
 #include "common.h"
 
 // ---------------------------------------------------------------------------
@@ -142,8 +142,8 @@ __global__ __launch_bounds__(256) void linear_dx_f32_k(
 
 // ---------------------------------------------------------------------------
 // dw (+db): D[co][ci]; A = dyr^T gather (16 consecutive co), B = x gather
-// (16 consecutive ci) — both coalesced row-major; reduce over an M split,
-// atomicAdd into zero-initialized dw/db. Waves split the ci tile.
+// (16 consecutive ci) — both coalesced row-major; reduce over an M split into
+// per-split slabs (deterministic fold). Waves split the ci tile.
 // ---------------------------------------------------------------------------
 template <bool RELU, bool WANT_BIAS>
 __global__ __launch_bounds__(256) void linear_dw_f32_k(
@@ -182,15 +182,20 @@ __global__ __launch_bounds__(256) void linear_dw_f32_k(
     if (WANT_BIAS) db_part += a;
   }
 
+  // deterministic partial slabs (dw points at workspace [ms][CO*CI]);
+  // reduce_splits_k folds the split axis in fixed order
+  {
+    float* slab = dw + (long)ms * ((long)CO * CI);
 #pragma unroll
-  for (int reg = 0; reg < 4; ++reg) {
-    int coo = co0 + lk * 4 + reg;
-    int cii = ci0 + wid * 16 + lm;
-    if (coo < CO && cii < CI) atomicAdd(&dw[coo * CI + cii], acc[reg]);
+    for (int reg = 0; reg < 4; ++reg) {
+      int coo = co0 + lk * 4 + reg;
+      int cii = ci0 + wid * 16 + lm;
+      if (coo < CO && cii < CI) slab[(long)coo * CI + cii] = acc[reg];
+    }
   }
   if (WANT_BIAS && blockIdx.y == 0 && wid == 0) {
     db_part += __shfl_down(db_part, 16);
     db_part += __shfl_down(db_part, 32);
-    if (lk == 0 && co0 + lm < CO) atomicAdd(&db[co0 + lm], db_part);
+    if (lk == 0 && co0 + lm < CO) db[(long)ms * CO + co0 + lm] = db_part;
   }
 }
