@@ -67,10 +67,16 @@ __global__ void bn_finalize_k(const float* __restrict__ stats,
                               float* __restrict__ invstd,
                               float* __restrict__ running_mean,
                               float* __restrict__ running_var,
-                              long npix, int C, float eps, float momentum,
-                              int update_running) {
+                              long npix, int C, int Cp, float eps,
+                              float momentum, int update_running) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+  if (c >= Cp) return;
+  if (c >= C) {  // zero the channel pads here (saves an at::zeros launch;
+                 // downstream kernels multiply pads by these values)
+    mean[c] = 0.f;
+    invstd[c] = 0.f;
+    return;
+  }
   float m = stats[c * 2] / (float)npix;
   float var = stats[c * 2 + 1] / (float)npix - m * m;
   var = fmaxf(var, 0.f);
