@@ -104,3 +104,32 @@ def test_loss_decreases_single_process():
         opt.step()
         losses.append(loss.item())
     assert losses[-1] < losses[0] * 0.7, losses[::10]
+
+
+def test_resnet18_state_dict_surface():
+    """ResNet-18 parameter names/shapes interchange with torchvision's
+    resnet18 (torchvision is not installed here; the canonical key set is
+    pinned explicitly)."""
+    from dist_mnist_amd.models import ResNet18
+
+    sd = ResNet18(num_classes=1000).state_dict()
+    # spot-pin the canonical surface
+    expected = {
+        "conv1.weight": (64, 3, 7, 7),
+        "bn1.weight": (64,),
+        "bn1.running_mean": (64,),
+        "layer1.0.conv1.weight": (64, 64, 3, 3),
+        "layer2.0.downsample.0.weight": (128, 64, 1, 1),
+        "layer2.0.downsample.1.running_var": (128,),
+        "layer3.1.conv2.weight": (256, 256, 3, 3),
+        "layer4.1.bn2.bias": (512,),
+        "fc.weight": (1000, 512),
+        "fc.bias": (1000,),
+    }
+    for k, shape in expected.items():
+        assert k in sd, k
+        assert tuple(sd[k].shape) == shape, (k, sd[k].shape)
+    # no extra parameter families beyond the torchvision naming scheme
+    prefixes = ("conv1", "bn1", "layer1", "layer2", "layer3", "layer4", "fc")
+    for k in sd:
+        assert k.startswith(prefixes) or k.endswith("num_batches_tracked"), k
