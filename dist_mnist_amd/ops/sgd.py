@@ -8,6 +8,12 @@ On GPU the whole parameter set updates in ONE kernel launch: a chunk table
 (param/grad/momentum pointers + lengths) is built once at init and the kernel
 grid-strides across the concatenated element range. On CPU it falls back to
 plain torch ops (used by the gloo plumbing config and the no-GPU tests).
+
+step() also RESETS the grads to zero after consuming them (fused into the
+same kernel stores on GPU): the next backward then accumulates onto zeros
+with no per-param zero_() launches, which is what lets the hipGraph step
+(runtime/graphs.py) drop grad-zeroing from the captured region entirely.
+An explicit zero_grad() before backward stays a harmless no-op.
 """
 
 from __future__ import annotations
@@ -87,4 +93,5 @@ class SGD(torch.optim.Optimizer):
                         p.add_(buf, alpha=-group["lr"])
                     else:
                         p.add_(p.grad, alpha=-group["lr"])
+                    p.grad.zero_()  # same consume-and-reset as the GPU kernel
         return loss

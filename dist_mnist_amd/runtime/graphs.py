@@ -45,14 +45,17 @@ class GraphedStep:
         torch.cuda.current_stream(self.device).wait_stream(side)
         torch.cuda.synchronize(self.device)
 
-        # capture one fwd+bwd graph per slot
+        # capture one fwd+bwd graph per slot. No grad-zeroing inside capture:
+        # the fused SGD kernel resets grads to zero as it consumes them
+        # (ops/sgd.py), so each fb replay accumulates onto zeros left by the
+        # previous opt replay.
         self.fb_graphs = []
         self.losses = []
         pool = None
         for x, t in self.slots:
             g = torch.cuda.CUDAGraph()
             with torch.cuda.graph(g, pool=pool):
-                loss = self._eager_fwd_bwd(x, t)
+                loss = self._eager_fwd_bwd(x, t, zero=False)
             pool = g.pool()
             self.fb_graphs.append(g)
             self.losses.append(loss)
@@ -61,6 +64,10 @@ class GraphedStep:
         self.opt_graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.opt_graph, pool=pool):
             self._eager_opt(scale=True)
+
+        # capture records without executing, so grads still hold warmup
+        # values; establish the all-zero invariant once before first replay
+        self._zero_grads()
 
         # warmup steps ran WITHOUT gradient sync (hooks disabled, rank-local
         # data) so ranks drifted; re-broadcast params AND optimizer momentum
@@ -87,8 +94,9 @@ class GraphedStep:
                 else:
                     p.grad = torch.zeros_like(p)
 
-    def _eager_fwd_bwd(self, x, t):
-        self._zero_grads()
+    def _eager_fwd_bwd(self, x, t, zero: bool = True):
+        if zero:
+            self._zero_grads()
         _, loss = self.model(x, t)
         loss.backward()
         return loss

@@ -1236,18 +1236,23 @@ __global__ void repack_w_k(const float* __restrict__ w, bf16* __restrict__ w_rp,
                            int dgrad) {
   // fwd:  w_rp[tap][co][ci] = w[co][ci][tap]        (Ap=COp, Bp=CIp)
   // dgrad: w_rp[KHKW-1-tap][ci][co] = w[co][ci][tap] (Ap=CIp, Bp=COp)
-  long total = (long)CO * CI * KHKW;
+  // Iterates the PADDED destination extent and writes zeros at pad positions
+  // itself (a garbage inf in a pad column would turn 0*inf into NaN inside
+  // valid accumulators) — the destination can be at::empty, no separate fill
+  // launch. Consecutive i => consecutive Bp index, so stores stay coalesced.
+  long total = (long)KHKW * Ap * Bp;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (long)gridDim.x * blockDim.x) {
-    int tap = i % KHKW;
-    long r = i / KHKW;
-    int ci = r % CI;
-    int co = r / CI;
-    float v = w[i];
-    if (dgrad)
-      w_rp[((long)(KHKW - 1 - tap) * Ap + ci) * Bp + co] = to_bf16(v);
-    else
-      w_rp[((long)tap * Ap + co) * Bp + ci] = to_bf16(v);
+    int b = i % Bp;
+    long r = i / Bp;
+    int a = r % Ap;
+    int tap_d = r / Ap;
+    int co = dgrad ? b : a;
+    int ci = dgrad ? a : b;
+    int tap = dgrad ? KHKW - 1 - tap_d : tap_d;
+    bf16 v = to_bf16(0.f);
+    if (co < CO && ci < CI) v = to_bf16(w[((long)co * CI + ci) * KHKW + tap]);
+    w_rp[i] = v;
   }
 }
 

@@ -432,3 +432,20 @@ __global__ void reduce_splits_k(const float* __restrict__ in,
   }
   if (sl == 0 && e < E) out[(long)z * E + e] = red[0][el];
 }
+
+// Single-LAUNCH fold for small element counts (BN statistics slabs,
+// bias-gradient slabs): a wave owns one output element, its 64 lanes stride
+// the split axis, and the cross-lane fold is a fixed xor tree — still a
+// deterministic fixed order for any given (E, S). Replaces the two-pass
+// launch pair where dispatch latency, not bandwidth, is the cost.
+__global__ __launch_bounds__(256) void reduce_splits_wave_k(
+    const float* __restrict__ in, float* __restrict__ out, long E, int S) {
+  long e = (long)blockIdx.x * 4 + (threadIdx.x >> 6);
+  if (e >= E) return;
+  int l = threadIdx.x & 63;
+  float v = 0.f;
+  for (int s = l; s < S; s += 64) v += in[(long)s * E + e];
+#pragma unroll
+  for (int d = 32; d > 0; d >>= 1) v += __shfl_xor(v, d);
+  if (l == 0) out[e] = v;
+}

@@ -191,6 +191,10 @@ __global__ void dropout_bwd_k(const float* __restrict__ dy,
 // Fused multi-tensor SGD+momentum: ONE launch for the whole
 // parameter set. chunks[b] = (tensor_idx, elem_offset); ptrs[t] =
 // (param, grad, momentum, numel). buf = mu*buf + g ; p -= lr*buf.
+// The grad is consumed here and nothing reads it again before the next
+// backward accumulates into it, so the kernel also resets it to zero — this
+// replaces a per-param zero_() launch train (62 fills/step on ResNet-18)
+// with stores fused into the update loop.
 // ---------------------------------------------------------------------------
 __global__ void fused_sgd_k(const long* __restrict__ ptrs,  // [T][4]
                             const int* __restrict__ chunks,  // [NC][2]
@@ -205,6 +209,7 @@ __global__ void fused_sgd_k(const long* __restrict__ ptrs,  // [T][4]
   for (long i = off + threadIdx.x; i < end; i += blockDim.x) {
     float b = momentum * mb[i] + g[i];
     mb[i] = b;
+    g[i] = 0.f;
     p[i] -= lr * b;
   }
 }

@@ -24,10 +24,15 @@ inline hipStream_t cur_stream() {
   return c10::hip::getCurrentHIPStream().stream();
 }
 
-// deterministic fold of wgrad partial slabs: 1 pass for small S, 2 for large
+// deterministic fold of wgrad partial slabs: 1 pass for small S, wave-per-
+// element for small E (one launch), 2 tiled passes otherwise
 inline void launch_reduce_splits(const at::Tensor& ws, at::Tensor& out, long E,
                                  int S) {
-  if (S > 64) {
+  if (S > 64 && E <= 4096) {
+    hipLaunchKernelGGL(reduce_splits_wave_k, dim3((int)cdiv(E, 4)), dim3(256),
+                       0, cur_stream(), ws.data_ptr<float>(),
+                       out.data_ptr<float>(), E, S);
+  } else if (S > 64) {
     auto mid = at::empty({16, E}, out.options());
     hipLaunchKernelGGL(reduce_splits_k, dim3((int)cdiv(E, 16), 16), dim3(256),
                        0, cur_stream(), ws.data_ptr<float>(),

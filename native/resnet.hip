@@ -89,6 +89,81 @@ __global__ void bn_finalize_k(const float* __restrict__ stats,
   }
 }
 
+// Fused slab-fold + finalize: ONE launch replacing reduce_splits x2 +
+// bn_finalize_k (3 dispatches of launch latency per BN forward). A wave owns
+// one channel; its 64 lanes stride the per-block partial slabs reading the
+// (sum, sumsq) pair as one 8-byte load, then fold through a fixed xor tree
+// (deterministic order for a given S) and lane 0 does the finalize math.
+__global__ __launch_bounds__(256) void bn_finalize_fold_k(
+    const float* __restrict__ ws, float* __restrict__ mean,
+    float* __restrict__ invstd, float* __restrict__ running_mean,
+    float* __restrict__ running_var, long npix, int C, int Cp, float eps,
+    float momentum, int update_running, int S) {
+  int c = blockIdx.x * 4 + (threadIdx.x >> 6);
+  if (c >= Cp) return;
+  int l = threadIdx.x & 63;
+  if (c >= C) {  // zero the channel pads (downstream kernels multiply by them)
+    if (l == 0) {
+      mean[c] = 0.f;
+      invstd[c] = 0.f;
+    }
+    return;
+  }
+  long E = (long)Cp * 2;
+  float s = 0.f, ss = 0.f;
+  for (int sp = l; sp < S; sp += 64) {
+    const float* p = ws + (long)sp * E + (long)c * 2;
+    s += p[0];
+    ss += p[1];
+  }
+#pragma unroll
+  for (int d = 32; d > 0; d >>= 1) {
+    s += __shfl_xor(s, d);
+    ss += __shfl_xor(ss, d);
+  }
+  if (l != 0) return;
+  float m = s / (float)npix;
+  float var = fmaxf(ss / (float)npix - m * m, 0.f);
+  mean[c] = m;
+  invstd[c] = rsqrtf(var + eps);
+  if (update_running) {
+    float unbiased = var * (float)npix / (float)max(1L, npix - 1);
+    running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * m;
+    running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
+  }
+}
+
+// Backward twin: folds the (sum_dy, sum_dy_xhat) partial slabs in one launch
+// and ALSO writes dbeta/dgamma densely (their previous extraction from
+// red[Cp][2] cost two strided-copy kernels per BN layer).
+__global__ __launch_bounds__(256) void bn_bwd_fold_k(
+    const float* __restrict__ ws, float* __restrict__ red,
+    float* __restrict__ dbeta, float* __restrict__ dgamma, int C, int Cp,
+    int S) {
+  int c = blockIdx.x * 4 + (threadIdx.x >> 6);
+  if (c >= Cp) return;
+  int l = threadIdx.x & 63;
+  long E = (long)Cp * 2;
+  float s0 = 0.f, s1 = 0.f;
+  for (int sp = l; sp < S; sp += 64) {
+    const float* p = ws + (long)sp * E + (long)c * 2;
+    s0 += p[0];
+    s1 += p[1];
+  }
+#pragma unroll
+  for (int d = 32; d > 0; d >>= 1) {
+    s0 += __shfl_xor(s0, d);
+    s1 += __shfl_xor(s1, d);
+  }
+  if (l != 0) return;
+  red[c * 2] = s0;
+  red[c * 2 + 1] = s1;
+  if (c < C) {
+    dbeta[c] = s0;
+    dgamma[c] = s1;
+  }
+}
+
 template <bool RELU>
 __global__ void bn_apply_k(const bf16* __restrict__ x, bf16* __restrict__ y,
                            const float* __restrict__ mean,
