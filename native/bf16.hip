@@ -1365,6 +1365,135 @@ __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
 }
 
 // ---------------------------------------------------------------------------
+// LDS-tiled stride-2 STEM conv (ResNet conv1: 7x7 s2, CIp==8). The gather
+// engine re-reads each input pixel ~KH*KW/4 times from L2 here (486 us on
+// the b256 stem, the largest single forward kernel); instead a block stages
+// the whole ((TH-1)*2+KH) x ((TW-1)*2+KH) x 8ch input region for an 8x32
+// output tile into LDS once (23 KB) and k-orders the GEMM as (kh | kw*8+ci):
+// every 8-k group is ONE tap's 8 channels = one aligned vec8 LDS read, with
+// the kw axis row-padded to 8 taps (KROW=64) so a 32-k MFMA step spans
+// exactly 4 predicated taps. Weights repack to [kh][co][kw*8+ci]
+// (repack_w_stem_k) making B-fragments contiguous 16-byte loads. MW=4 pixel
+// fragments x NW=4 co-tiles -> 16 MFMAs per 32-k step.
+// ---------------------------------------------------------------------------
+template <int KH, bool HAS_BIAS, bool RELU>
+__global__ __launch_bounds__(256) void conv_stem_s2_bf16_k(
+    const bf16* __restrict__ x, const bf16* __restrict__ w_rp,
+    const float* __restrict__ bias, bf16* __restrict__ y, int H, int W,
+    int COp, int COcan, int OH, int OW, int pad, int N, int tiles_h,
+    int tiles_w) {
+  constexpr int TH = 8, TW = 32, MW = 4, NW = 4;
+  // channels packed to 4 (CI <= 4): k per row = KH*4 <= 32 = ONE MFMA step
+  // per tap row. +1 region pad column: the kw==7 lane half reads it freely
+  // (its B column is repack-zeroed, so the product is 0 either way).
+  constexpr int RH = (TH - 1) * 2 + KH, RW = (TW - 1) * 2 + KH + 1;
+  constexpr int KROW = 32;
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6;
+  const int l = tid & 63;
+  const int lm = l & 15;
+  const int lk = l >> 4;
+
+  int tw = blockIdx.x % tiles_w;
+  int rest = blockIdx.x / tiles_w;
+  int th = rest % tiles_h;
+  int n = rest / tiles_h;
+  const int oh0 = th * TH, ow0 = tw * TW;
+  const int co0 = blockIdx.y * (NW * 16);
+
+  __shared__ bf16 lds_x[RH * RW * 4 + 8];
+
+  // stage the input region ONCE, dropping input channels 4..7 (pure pad):
+  // one 8-byte store per region pixel
+  const int ih0 = oh0 * 2 - pad, iw0 = ow0 * 2 - pad;
+  for (int idx = tid; idx < RH * RW; idx += 256) {
+    int rx = idx % RW, ry = idx / RW;
+    int ih = ih0 + ry, iw = iw0 + rx;
+    bf16x4 v = {};
+    if (ih >= 0 && ih < H && iw >= 0 && iw < W)
+      v = *reinterpret_cast<const bf16x4*>(
+          &x[((long)(n * H + ih) * W + iw) * 8]);
+    *reinterpret_cast<bf16x4*>(&lds_x[idx * 4]) = v;
+  }
+  __syncthreads();
+
+  const int py = wid * 2 + (lm >> 3);  // fragment row within the tile
+  f32x4 acc[MW][NW] = {};
+  // software-pipelined B loads: next row's weights issue before this row's
+  // MFMAs so the 16-MFMA step never stalls on L1/L2 latency
+  const bf16* wbase = w_rp + (long)(co0 + lm) * KROW + lk * 8;
+  bf16x8 bw[NW];
+#pragma unroll
+  for (int ni = 0; ni < NW; ++ni)
+    bw[ni] = *reinterpret_cast<const bf16x8*>(&wbase[(long)ni * 16 * KROW]);
+#pragma unroll
+  for (int kh = 0; kh < KH; ++kh) {
+    bf16x8 bc[NW];
+#pragma unroll
+    for (int ni = 0; ni < NW; ++ni) bc[ni] = bw[ni];
+    if (kh + 1 < KH) {
+      const bf16* wn = wbase + (long)(kh + 1) * COp * KROW;
+#pragma unroll
+      for (int ni = 0; ni < NW; ++ni)
+        bw[ni] = *reinterpret_cast<const bf16x8*>(&wn[(long)ni * 16 * KROW]);
+    }
+#pragma unroll
+    for (int mi = 0; mi < MW; ++mi) {
+      // k_local = lk*8+j -> tap kw = lk*2 + j/4, ci = j&3: the vec8 spans
+      // two adjacent region pixels' 4-channel packs — contiguous, aligned
+      int px = mi * 8 + (lm & 7);
+      bf16x8 a = *reinterpret_cast<const bf16x8*>(
+          &lds_x[((py * 2 + kh) * RW + px * 2 + lk * 2) * 4]);
+#pragma unroll
+      for (int ni = 0; ni < NW; ++ni)
+        acc[mi][ni] = mfma16x16x32bf(a, bc[ni], acc[mi][ni]);
+    }
+  }
+
+#pragma unroll
+  for (int mi = 0; mi < MW; ++mi) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      int mo = lk * 4 + reg;
+      int oy = oh0 + wid * 2 + (mo >> 3);
+      int ox = ow0 + mi * 8 + (mo & 7);
+      if (oy >= OH || ox >= OW) continue;
+      long yb = ((long)(n * OH + oy) * OW + ox) * COp;
+#pragma unroll
+      for (int ni = 0; ni < NW; ++ni) {
+        int co = co0 + ni * 16 + lm;
+        if (co >= COp) continue;
+        float v = (co < COcan) ? acc[mi][ni][reg] : 0.f;
+        if (HAS_BIAS && co < COcan) v += bias[co];
+        if (RELU) v = fmaxf(v, 0.f);
+        y[yb + co] = to_bf16(v);
+      }
+    }
+  }
+}
+
+// w[CO][CI][KH][KW] fp32 -> w_rp[kh][co][kw*4+ci] bf16 (CI <= 4), the kw
+// axis row-padded to KROW=32 k; all pads written as zeros (destination can
+// be at::empty)
+__global__ void repack_w_stem_k(const float* __restrict__ w,
+                                bf16* __restrict__ w_rp, int CO, int CI,
+                                int KH, int KW, int COp) {
+  long total = (long)KH * COp * 32;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int kwci = i & 31;
+    long r = i >> 5;
+    int co = r % COp;
+    int kh = r / COp;
+    int kw = kwci >> 2, ci = kwci & 3;
+    bf16 v = to_bf16(0.f);
+    if (co < CO && ci < CI && kw < KW)
+      v = to_bf16(w[(((long)co * CI + ci) * KH + kh) * KW + kw]);
+    w_rp[i] = v;
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Wide-class LogSoftmax+NLL (ResNet: C=1000): one WAVE per row, lanes split
 // the class axis, shfl tree reductions. (The thread-per-row kernel above is
 // for tiny C like LeNet's 10.)

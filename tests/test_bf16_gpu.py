@@ -37,7 +37,10 @@ def test_mfma_probe_mapping():
         (4, 10, 12, 12, 20, 5, 5, 1, 0),  # conv2
         (2, 3, 18, 18, 7, 3, 3, 1, 1),    # pad
         (2, 8, 16, 16, 16, 3, 3, 2, 1),   # stride 2
-        (2, 3, 32, 32, 8, 7, 7, 2, 3),    # 7x7 s2 stem
+        (2, 3, 32, 32, 8, 7, 7, 2, 3),    # 7x7 s2 (gather engine: COp<64)
+        (2, 3, 37, 37, 64, 7, 7, 2, 3),   # 7x7 s2 stem engine, ragged tiles
+        (4, 5, 41, 37, 128, 7, 7, 2, 3),  # stem engine, 2 co-tiles, non-square
+        (2, 3, 224, 224, 64, 7, 7, 2, 3), # the actual ResNet stem shape
         (131, 1, 28, 28, 10, 5, 5, 1, 0), # tile remainder
     ],
 )
