@@ -1256,7 +1256,8 @@ __global__ void repack_w_k(const float* __restrict__ w, bf16* __restrict__ w_rp,
   }
 }
 
-template <int KH, bool HAS_BIAS, bool RELU, int NW, int MW, int NIMG = 1>
+template <int KH, bool HAS_BIAS, bool RELU, int NW, int MW, int NIMG = 1,
+          int S = 1>
 __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
     const bf16* __restrict__ x, const bf16* __restrict__ w_rp,
     const float* __restrict__ bias, bf16* __restrict__ y,
@@ -1266,8 +1267,12 @@ __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
   // 16-pixel fragments (split across columns and images), so every
   // B-fragment read feeds MW MFMAs. NIMG>1 keeps the reuse on small maps
   // (ResNet layer3/4) where wide column tiles would be mostly padding.
+  // S=2 (ResNet layer2/3 block-0 3x3 convs): the region grows to
+  // (T-1)*2+KH per axis, so the ci chunk halves (CC=32) to stay inside the
+  // 64 KB static-LDS block limit.
   constexpr int TH = 8, TW = 8 * (MW / NIMG);
-  constexpr int RH = TH + KH - 1, RW = TW + KH - 1;
+  constexpr int RH = (TH - 1) * S + KH, RW = (TW - 1) * S + KH;
+  constexpr int CC = (S == 1) ? 64 : 32;
   const int tid = threadIdx.x;
   const int wid = tid >> 6;
   const int l = tid & 63;
@@ -1282,7 +1287,7 @@ __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
   const int co0 = blockIdx.y * (NW * 16);
 
   // per-pixel stride padded +8 bf16 (bank spread for the 16-pixel b128 reads)
-  __shared__ bf16 lds_x[NIMG * RH * RW * 72 + 8];
+  __shared__ bf16 lds_x[NIMG * RH * RW * (CC + 8) + 8];
 
   // wave pixel map: fragment mi covers a 2-row x 8-col patch; the block's 4
   // waves cover the 8 rows. pixel (py, px) for (wave, mi, lm):
@@ -1292,8 +1297,8 @@ __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
   f32x4 acc[MW][NW] = {};
 
   constexpr int KHKW = KH * KH;
-  for (int cic = 0; cic < CIp; cic += 64) {
-    int cc = min(64, CIp - cic);
+  for (int cic = 0; cic < CIp; cic += CC) {
+    int cc = min(CC, CIp - cic);
     int ccp = cc + 8;
     __syncthreads();
     for (int idx = tid; idx < NIMG * RH * RW * (cc / 8); idx += 256) {
@@ -1303,7 +1308,7 @@ __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
       int rest2 = p / RW;
       int ry = rest2 % RH;
       int img = rest2 / RH;
-      int ih = oh0 - pad + ry, iw = ow0 - pad + rx;
+      int ih = oh0 * S - pad + ry, iw = ow0 * S - pad + rx;
       bf16x8 v = bf8_zero();
       if (ih >= 0 && ih < H && iw >= 0 && iw < W && n + img < N)
         v = *reinterpret_cast<const bf16x8*>(
@@ -1329,7 +1334,8 @@ __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
           int px = (mi % (MW / NIMG)) * 8 + (lm & 7);
           int img = mi / (MW / NIMG);
           const bf16* xpix =
-              &lds_x[(img * RH * RW + (py + kh) * RW + (px + kw)) * ccp];
+              &lds_x[(img * RH * RW + (py * S + kh) * RW + (px * S + kw)) *
+                     ccp];
           bf16x8 a = gin ? *reinterpret_cast<const bf16x8*>(&xpix[ks + lk * 8])
                          : bf8_zero();
 #pragma unroll

@@ -41,6 +41,10 @@ def test_mfma_probe_mapping():
         (2, 3, 37, 37, 64, 7, 7, 2, 3),   # 7x7 s2 stem engine, ragged tiles
         (4, 5, 41, 37, 128, 7, 7, 2, 3),  # stem engine, 2 co-tiles, non-square
         (2, 3, 224, 224, 64, 7, 7, 2, 3), # the actual ResNet stem shape
+        (2, 64, 27, 27, 128, 3, 3, 2, 1), # 3x3 s2 tiled engine (layer2 shape)
+        (2, 33, 29, 25, 64, 3, 3, 2, 1),  # s2 tiled, ragged ci chunk, non-sq
+        (2, 256, 14, 14, 512, 3, 3, 2, 1),# s2 tiled two-image (layer4 shape)
+        (4, 64, 15, 15, 64, 3, 3, 2, 1),  # s2 tiled NIMG2, ragged rows
         (131, 1, 28, 28, 10, 5, 5, 1, 0), # tile remainder
     ],
 )
