@@ -1321,7 +1321,11 @@ __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
     for (int tap = 0; tap < KHKW; ++tap) {
       int kh = tap / KH, kw = tap % KH;
       const bf16* wtap = w_rp + ((long)tap * COp + co0) * CIp + cic;
-      for (int ks = 0; ks < cc; ks += 32) {
+      // compile-time trip count (remainder chunks handled by the gin
+      // predicate): the whole tap x ks nest fully unrolls, so the compiler
+      // can hoist and software-pipeline the A/B loads across taps
+#pragma unroll
+      for (int ks = 0; ks < CC; ks += 32) {
         bool gin = lk * 8 < cc - ks;
         bf16x8 bw[NW];
 #pragma unroll
