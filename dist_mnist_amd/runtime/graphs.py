@@ -46,28 +46,55 @@ class GraphedStep:
         torch.cuda.synchronize(self.device)
 
         # capture one fwd+bwd graph per slot. No grad-zeroing inside capture:
-        # the fused SGD kernel resets grads to zero as it consumes them
-        # (ops/sgd.py), so each fb replay accumulates onto zeros left by the
-        # previous opt replay.
+        # - DDP: grads are bucket views, so backward ACCUMULATES; the fused
+        #   SGD kernel resets them to zero as it consumes them (ops/sgd.py).
+        # - single GPU: grads are set to None before each slot's capture, so
+        #   backward ASSIGNS pool-allocated tensors instead (pointer-stable
+        #   across replays of that slot) — no accumulate-add kernels at all
+        #   (62 of them per step on ResNet-18). Each slot then gets its own
+        #   captured optimizer step over its own grad pointers.
         self.fb_graphs = []
         self.losses = []
+        self.opt_graphs = []
         pool = None
+        slot_grads = []
         for x, t in self.slots:
+            if self.ddp is None:
+                for p in self.model.parameters():
+                    p.grad = None
             g = torch.cuda.CUDAGraph()
             with torch.cuda.graph(g, pool=pool):
                 loss = self._eager_fwd_bwd(x, t, zero=False)
             pool = g.pool()
             self.fb_graphs.append(g)
             self.losses.append(loss)
+            slot_grads.append([p.grad for p in self.model.parameters()])
 
-        # optimizer graph (shared; includes DDP 1/world scaling when needed)
-        self.opt_graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.opt_graph, pool=pool):
-            self._eager_opt(scale=True)
-
-        # capture records without executing, so grads still hold warmup
-        # values; establish the all-zero invariant once before first replay
-        self._zero_grads()
+        if self.ddp is None:
+            # one eager step per slot first: it builds/caches the fused-SGD
+            # pointer table for that slot's grads (device-tensor creation is
+            # not allowed inside capture), then the capture records the
+            # table-driven kernel. The extra updates are just more warmup.
+            params = list(self.model.parameters())
+            for grads, _ in zip(slot_grads, self.slots):
+                for p, gr in zip(params, grads):
+                    p.grad = gr
+                self.opt.step()
+                og = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(og, pool=pool):
+                    self._eager_opt(scale=False)
+                self.opt_graphs.append(og)
+        else:
+            # optimizer graph (shared; includes the DDP 1/world scaling)
+            og = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(og, pool=pool):
+                self._eager_opt(scale=True)
+            self.opt_graphs = [og for _ in self.slots]
+            # capture records without executing, so the bucket-view grads
+            # still hold warmup values; establish the all-zero invariant
+            # once before the first replay (assign-mode slots need none:
+            # their backward overwrites)
+            self._zero_grads()
 
         # warmup steps ran WITHOUT gradient sync (hooks disabled, rank-local
         # data) so ranks drifted; re-broadcast params AND optimizer momentum
@@ -121,5 +148,5 @@ class GraphedStep:
                     self.ddp.native_comm.all_reduce_sum(b.flat)
                 else:
                     dist.all_reduce(b.flat, op=dist.ReduceOp.SUM, group=self.ddp.pg)
-        self.opt_graph.replay()
+        self.opt_graphs[slot].replay()
         return self.losses[slot]
