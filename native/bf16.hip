@@ -1375,6 +1375,153 @@ __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
 }
 
 // ---------------------------------------------------------------------------
+// LDS-tiled stride-2 DGRAD (parity classes): for stride 2, dx pixel
+// (2y+p, 2x+q) only receives taps of matching parity, and per class the
+// whole dgrad collapses to a stride-1 2x2 conv over dy:
+//     tap (a,b) in {0,1}^2  <->  kh = p + pad - 2a, kw = q + pad - 2b
+// (invalid kh/kw map to zero weights in the repack — this covers both the
+// 3x3 pad-1 convs, where classes carry 1/2/2/4 live taps, and the 1x1
+// pad-0 downsample, where only class (0,0) tap (0,0) is live). dy is
+// HALF-resolution, so the staged region for an 8-col tile is tiny
+// (9 x (TW+1) x 72 bf16) and every dy element is read from LDS for all
+// four taps. The gather parity engine this replaces was ~11x off its
+// bandwidth bound (profiles/resnet18_pmc_final2_r1.txt: 3% MFMA, scattered
+// per-pixel loads). Epilogue writes dx[2y+p][2x+q] directly.
+// blockIdx.z = class; per-class output dims Hc x Wc derived in-kernel.
+// ---------------------------------------------------------------------------
+template <int NW, int MW, int NIMG>
+__global__ __launch_bounds__(256) void conv_tiled_dgrad_s2_bf16_k(
+    const bf16* __restrict__ dy, const bf16* __restrict__ wc,
+    bf16* __restrict__ dx, int COp, int OH, int OW, int CIp, int CIcan,
+    int H, int W, int N, int tiles_h, int tiles_w, int tap_mask) {
+  constexpr int TH = 8, TW = 8 * (MW / NIMG);
+  constexpr int RH = TH + 1, RW = TW + 1;
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6;
+  const int l = tid & 63;
+  const int lm = l & 15;
+  const int lk = l >> 4;
+
+  int tw = blockIdx.x % tiles_w;
+  int rest = blockIdx.x / tiles_w;
+  int th = rest % tiles_h;
+  int n = (rest / tiles_h) * NIMG;
+  const int oh0 = th * TH, ow0 = tw * TW;
+  const int ci0 = blockIdx.y * (NW * 16);
+  const int p = blockIdx.z >> 1, q = blockIdx.z & 1;
+  const int Hc = (H - p + 1) >> 1, Wc = (W - q + 1) >> 1;
+  const bf16* wcls = wc + (long)blockIdx.z * 4 * CIp * COp;
+
+  __shared__ bf16 lds_y[NIMG * RH * RW * 72 + 8];
+
+  const int py = wid * 2 + (lm >> 3);
+  f32x4 acc[MW][NW] = {};
+
+  // a class with no live taps (1x1 downsample, classes 1-3) skips straight
+  // to the zero-writing epilogue
+  const int k_end = ((tap_mask >> (blockIdx.z * 4)) & 15) ? COp : 0;
+  for (int cic = 0; cic < k_end; cic += 64) {
+    int cc = min(64, COp - cic);
+    int ccp = cc + 8;
+    __syncthreads();
+    for (int idx = tid; idx < NIMG * RH * RW * (cc / 8); idx += 256) {
+      int g8 = idx % (cc / 8);
+      int pp = idx / (cc / 8);
+      int rx = pp % RW;
+      int rest2 = pp / RW;
+      int ry = rest2 % RH;
+      int img = rest2 / RH;
+      int ih = oh0 + ry, iw = ow0 + rx;
+      bf16x8 v = bf8_zero();
+      if (ih < OH && iw < OW && n + img < N)
+        v = *reinterpret_cast<const bf16x8*>(
+            &dy[((long)((n + img) * OH + ih) * OW + iw) * COp + cic +
+                g8 * 8]);
+      *reinterpret_cast<bf16x8*>(
+          &lds_y[(img * RH * RW + ry * RW + rx) * ccp + g8 * 8]) = v;
+    }
+    __syncthreads();
+#pragma unroll
+    for (int tap = 0; tap < 4; ++tap) {
+      // live-tap mask (host bit cls*4+tap): classes carry 1/2/2/4 live
+      // taps for 3x3 pad-1 and a single live tap for 1x1 — skipping the
+      // zero-weight ones is ~44% of the MFMA work (block-uniform branch)
+      if (!((tap_mask >> (blockIdx.z * 4 + tap)) & 1)) continue;
+      int a = tap >> 1, b = tap & 1;
+      const bf16* wtap = wcls + ((long)tap * CIp + ci0) * COp + cic;
+#pragma unroll
+      for (int ks = 0; ks < 64; ks += 32) {
+        bool gin = lk * 8 < cc - ks;
+        bf16x8 bw[NW];
+#pragma unroll
+        for (int ni = 0; ni < NW; ++ni)
+          bw[ni] = gin ? *reinterpret_cast<const bf16x8*>(
+                             &wtap[(long)(ni * 16 + lm) * COp + ks + lk * 8])
+                       : bf8_zero();
+#pragma unroll
+        for (int mi = 0; mi < MW; ++mi) {
+          int px = (mi % (MW / NIMG)) * 8 + (lm & 7);
+          int img = mi / (MW / NIMG);
+          const bf16* ypix =
+              &lds_y[(img * RH * RW + (py + a) * RW + (px + b)) * ccp];
+          bf16x8 av = gin ? *reinterpret_cast<const bf16x8*>(&ypix[ks + lk * 8])
+                          : bf8_zero();
+#pragma unroll
+          for (int ni = 0; ni < NW; ++ni)
+            acc[mi][ni] = mfma16x16x32bf(av, bw[ni], acc[mi][ni]);
+        }
+      }
+    }
+  }
+
+#pragma unroll
+  for (int mi = 0; mi < MW; ++mi) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      int mo = lk * 4 + reg;
+      int oy = oh0 + wid * 2 + (mo >> 3);
+      int ox = ow0 + (mi % (MW / NIMG)) * 8 + (mo & 7);
+      int nimg = n + mi / (MW / NIMG);
+      if (oy >= Hc || ox >= Wc || nimg >= N) continue;
+      long xb = ((long)(nimg * H + 2 * oy + p) * W + 2 * ox + q) * CIp;
+#pragma unroll
+      for (int ni = 0; ni < NW; ++ni) {
+        int ci = ci0 + ni * 16 + lm;
+        if (ci >= CIp) continue;
+        dx[xb + ci] = to_bf16(ci < CIcan ? acc[mi][ni][reg] : 0.f);
+      }
+    }
+  }
+}
+
+// w[CO][CI][KH][KW] fp32 -> wc[class][a*2+b][ci][co] bf16 with
+// kh = p + pad - 2a, kw = q + pad - 2b (invalid taps and channel pads
+// written as zeros; destination can be at::empty)
+__global__ void repack_w_dgrad_s2_k(const float* __restrict__ w,
+                                    bf16* __restrict__ wc, int CO, int CI,
+                                    int KH, int KW, int CIp, int COp,
+                                    int pad) {
+  long percls = (long)4 * CIp * COp;
+  long total = 4 * percls;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int co = i % COp;
+    long r = i / COp;
+    int ci = r % CIp;
+    long r2 = r / CIp;
+    int tap = r2 & 3;
+    int cls = r2 >> 2;
+    int pcl = cls >> 1, qcl = cls & 1;
+    int kh = pcl + pad - 2 * (tap >> 1);
+    int kw = qcl + pad - 2 * (tap & 1);
+    bf16 v = to_bf16(0.f);
+    if (kh >= 0 && kh < KH && kw >= 0 && kw < KW && co < CO && ci < CI)
+      v = to_bf16(w[(((long)co * CI + ci) * KH + kh) * KW + kw]);
+    wc[i] = v;
+  }
+}
+
+// ---------------------------------------------------------------------------
 // LDS-tiled stride-2 STEM conv (ResNet conv1: 7x7 s2, CIp==8). The gather
 // engine re-reads each input pixel ~KH*KW/4 times from L2 here (486 us on
 // the b256 stem, the largest single forward kernel); instead a block stages

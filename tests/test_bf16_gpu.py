@@ -72,6 +72,9 @@ def test_conv2d_bf16_fwd(N, CI, H, W, CO, KH, KW, stride, pad):
         (4, 10, 12, 12, 20, 5, 5, 1, 0),
         (2, 3, 18, 18, 7, 3, 3, 1, 1),
         (2, 8, 16, 16, 16, 3, 3, 2, 1),
+        (2, 64, 27, 27, 128, 3, 3, 2, 1),  # tiled s2 dgrad (3x3 pad1)
+        (2, 64, 15, 15, 128, 3, 3, 2, 1),  # tiled s2 dgrad, odd H (ragged classes)
+        (2, 64, 14, 14, 128, 1, 1, 2, 0),  # tiled s2 dgrad (1x1 downsample)
     ],
 )
 def test_conv2d_bf16_bwd(N, CI, H, W, CO, KH, KW, stride, pad):
