@@ -482,6 +482,168 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_bf16_k(
 constexpr int WGRAD_BF16_KTILE = 4 * 2 * 16;  // KT=2 -> 128
 
 // ---------------------------------------------------------------------------
+// Region-stationary wgrad for the s1 3x3 pad-1 shapes (ResNet residual
+// convs). The generic engine above re-gathers the im2col form of x from
+// global per k-tile (~9x the raw bytes, x2 per extra co-tile: ~925 MB/call
+// on layer2, L2-thrashed across splits — measured 7% MFMA-busy). Here a
+// block owns [64 co x 32 ci x all 9 taps] and marches m in 256-pixel
+// chunks:
+//   - the x ROWS for a chunk stage once (raw bytes, not im2col): region
+//     [rows+halo][W+2 zero-padded cols][32 ci], vec8-aligned since the
+//     ci stride is 64 B;
+//   - dy [256][64] stages once per chunk and each wave hoists its 8
+//     B-fragments into VGPRs immediately (dy is TAP-INVARIANT — the
+//     insight that makes the 9-tap reuse free);
+//   - per tap, [256][16]x2 A-images build LDS->LDS from the region with
+//     per-m row predicates (vec8 reads at ci-stride, vec8 writes), into
+//     the same 32 KB the dy planes used (double-buffered across taps);
+//   - fragments then come out via the same tr16 recipe as the generic
+//     engine: 144 MFMAs per wave-chunk against 152 transpose reads.
+// Global traffic per call: x once per co-tile + dy once per ci-tile
+// (~310 MB on layer2 vs ~925) and the MFMA:gather ratio flips the kernel
+// from L2-latency-bound to matrix-core-bound.
+// ---------------------------------------------------------------------------
+constexpr int WGRS_PX = 464;  // region capacity: max (255/OW+4)*(OW+2), OW<=56
+
+__global__ __launch_bounds__(256) void conv2d_wgrad_rs_bf16_k(
+    const bf16* __restrict__ dy, const bf16* __restrict__ x,
+    float* __restrict__ dw_ws, float* __restrict__ db_ws, int CIp, int CI,
+    int COp, int CO, int OH, int OW, int M, int Kcan, int m_per_split) {
+  const int cic0 = blockIdx.x * 32;       // ci tile
+  const int co0 = blockIdx.y * 64;        // co tile (4 waves x 16)
+  const int ms = blockIdx.z;
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6;
+  const int l = tid & 63;
+  const int lm = l & 15;
+  const int W2 = OW + 2;
+
+  __shared__ bf16 lds_reg[WGRS_PX * 32];      // [px][32ci]
+  __shared__ bf16 lds_img[4][256][16];        // dy planes, then A-image slots
+  __shared__ int lds_mm[256];                 // packed (px_index<<10 | oy+1)
+
+  const int m_begin = ms * m_per_split;
+  const int m_end = min(M, m_begin + m_per_split);
+
+  f32x4 acc[2][9] = {};   // [ci-frag][tap]
+  float db_part = 0.f;    // bias grad: colsum of this wave's dy fragments
+
+  for (int mc = m_begin; mc < m_end; mc += 256) {
+    const int g0 = mc / OW;               // first global row of the chunk
+    const int mlast = min(mc + 255, m_end - 1);
+    const int rows = mlast / OW - g0 + 3;  // incl. +-1 halo
+    __syncthreads();
+    // per-m map: region pixel index + output row (for tap predicates)
+    if (tid < 256) {
+      int m = mc + tid;
+      int mm = m < m_end ? m : mc;
+      int g = mm / OW, ox = mm % OW;
+      int px = (g - g0 + 1) * W2 + ox + 1;
+      int oy = g % OH;
+      lds_mm[tid] = (px << 10) | (m < m_end ? oy + 1 : 0);  // 0 => dead m
+    }
+    // x region rows [g0-1, g0+rows-2], zero-padded cols and out-of-image rows
+    for (int idx = tid; idx < rows * W2 * 4; idx += 256) {
+      int g8 = idx & 3;
+      int p = idx >> 2;
+      int c = p % W2;
+      int r = p / W2;
+      int g = g0 - 1 + r;
+      int iw = c - 1;
+      bf16x8 v = bf8_zero();
+      if (g >= 0 && g < (M / OW) && iw >= 0 && iw < OW)
+        v = *reinterpret_cast<const bf16x8*>(
+            &x[((long)g * OW + iw) * CIp + cic0 + g8 * 8]);
+      *reinterpret_cast<bf16x8*>(&lds_reg[(p << 5) + g8 * 8]) = v;
+    }
+    // dy planes [ni][256][16co]
+    for (int idx = tid; idx < 4 * 256 * 2; idx += 256) {
+      int half = idx & 1;
+      int rest = idx >> 1;
+      int mloc = rest & 255;
+      int ni = rest >> 8;
+      int m = mc + mloc;
+      bf16x8 v = bf8_zero();
+      int cb = co0 + ni * 16 + half * 8;
+      if (m < m_end && cb < COp)
+        v = *reinterpret_cast<const bf16x8*>(&dy[(long)m * COp + cb]);
+      *reinterpret_cast<bf16x8*>(&lds_img[ni][mloc][half * 8]) = v;
+    }
+    __syncthreads();
+
+    // hoist this wave's 8 dy B-fragments before the planes are reused
+    bf16x8 bfr[8];
+#pragma unroll
+    for (int s = 0; s < 8; ++s) bfr[s] = tr_read_8m(&lds_img[wid][s * 32][0], l);
+    if (db_ws != nullptr && blockIdx.x == 0) {
+#pragma unroll
+      for (int s = 0; s < 8; ++s)
+#pragma unroll
+        for (int j = 0; j < 8; ++j) db_part += to_f32(bfr[s][j]);
+    }
+    __syncthreads();
+
+    // A-image build for one tap: planes (buf*2, buf*2+1) = ci halves
+    auto build = [&](int tap, int buf) {
+      int kh = tap / 3 - 1, kw = tap % 3 - 1;  // tap offsets (pad 1)
+      for (int idx = tid; idx < 256 * 4; idx += 256) {
+        int g8 = idx & 3;
+        int mloc = idx >> 2;
+        int mm = lds_mm[mloc];
+        int oy = (mm & 1023) - 1;
+        bf16x8 v = bf8_zero();
+        if (oy >= 0 && oy + kh >= 0 && oy + kh < OH) {
+          int px = (mm >> 10) + kh * W2 + kw;
+          v = *reinterpret_cast<const bf16x8*>(&lds_reg[(px << 5) + g8 * 8]);
+        }
+        *reinterpret_cast<bf16x8*>(
+            &lds_img[buf * 2 + (g8 >> 1)][mloc][(g8 & 1) * 8]) = v;
+      }
+    };
+
+    build(0, 0);
+    __syncthreads();
+#pragma unroll
+    for (int tap = 0; tap < 9; ++tap) {
+      int buf = tap & 1;
+      if (tap + 1 < 9) build(tap + 1, buf ^ 1);  // overlap with compute
+#pragma unroll
+      for (int f = 0; f < 2; ++f) {
+#pragma unroll
+        for (int s = 0; s < 8; ++s) {
+          bf16x8 a = tr_read_8m(&lds_img[buf * 2 + f][s * 32][0], l);
+          acc[f][tap] = mfma16x16x32bf(a, bfr[s], acc[f][tap]);
+        }
+      }
+      __syncthreads();
+    }
+  }
+
+  // partial slab: D row (lk*4+reg) = ci-local, col (lm) = co
+  float* slab = dw_ws + (long)ms * (CO * (long)Kcan);
+  const int lk = l >> 4;
+#pragma unroll
+  for (int f = 0; f < 2; ++f) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      int ci = cic0 + f * 16 + lk * 4 + reg;
+      if (ci >= CI) continue;
+      int co = co0 + wid * 16 + lm;
+      if (co >= CO) continue;
+#pragma unroll
+      for (int tap = 0; tap < 9; ++tap)
+        slab[(long)co * Kcan + (long)ci * 9 + tap] = acc[f][tap][reg];
+    }
+  }
+  if (db_ws != nullptr && blockIdx.x == 0) {
+    db_part += __shfl_down(db_part, 16);
+    db_part += __shfl_down(db_part, 32);
+    int co = co0 + wid * 16 + lm;
+    if (lk == 0 && co < CO) db_ws[(long)ms * CO + co] = db_part;
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Linear forward, bf16: y[M][COp] = x[M][CIk] . w^T + b (optional ReLU).
 // w is fp32 canonical [CO][CIcan]; perm[k] maps the (padded, possibly
 // NHWC-flattened) k index to the canonical column (-1 => zero pad), so the
