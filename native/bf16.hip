@@ -505,6 +505,7 @@ constexpr int WGRAD_BF16_KTILE = 4 * 2 * 16;  // KT=2 -> 128
 // ---------------------------------------------------------------------------
 constexpr int WGRS_PX = 464;  // region capacity: max (255/OW+4)*(OW+2), OW<=56
 
+template <int MC>
 __global__ __launch_bounds__(256) void conv2d_wgrad_rs_bf16_k(
     const bf16* __restrict__ dy, const bf16* __restrict__ x,
     float* __restrict__ dw_ws, float* __restrict__ db_ws, int CIp, int CI,
@@ -519,8 +520,12 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_rs_bf16_k(
   const int W2 = OW + 2;
 
   __shared__ bf16 lds_reg[WGRS_PX * 32];      // [px][32ci]
-  __shared__ bf16 lds_img[4][256][16];        // dy planes, then A-image slots
-  __shared__ int lds_mm[256];                 // packed (px_index<<10 | oy+1)
+  // dy planes then A-image slots; +16-elem stagger per plane so the build's
+  // paired-plane vec8 writes land on different bank columns
+  __shared__ bf16 lds_img_raw[4 * MC * 16 + 3 * 16];
+  __shared__ int lds_mm[MC];                  // packed (px_index<<10 | oy+1)
+  auto plane = [&](int p) { return &lds_img_raw[p * (MC * 16 + 16)]; };
+  constexpr int MSTEPS = MC / 32;
 
   const int m_begin = ms * m_per_split;
   const int m_end = min(M, m_begin + m_per_split);
@@ -528,13 +533,13 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_rs_bf16_k(
   f32x4 acc[2][9] = {};   // [ci-frag][tap]
   float db_part = 0.f;    // bias grad: colsum of this wave's dy fragments
 
-  for (int mc = m_begin; mc < m_end; mc += 256) {
+  for (int mc = m_begin; mc < m_end; mc += MC) {
     const int g0 = mc / OW;               // first global row of the chunk
-    const int mlast = min(mc + 255, m_end - 1);
+    const int mlast = min(mc + MC - 1, m_end - 1);
     const int rows = mlast / OW - g0 + 3;  // incl. +-1 halo
     __syncthreads();
     // per-m map: region pixel index + output row (for tap predicates)
-    if (tid < 256) {
+    if (tid < MC) {
       int m = mc + tid;
       int mm = m < m_end ? m : mc;
       int g = mm / OW, ox = mm % OW;
@@ -557,27 +562,28 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_rs_bf16_k(
       *reinterpret_cast<bf16x8*>(&lds_reg[(p << 5) + g8 * 8]) = v;
     }
     // dy planes [ni][256][16co]
-    for (int idx = tid; idx < 4 * 256 * 2; idx += 256) {
+    for (int idx = tid; idx < 4 * MC * 2; idx += 256) {
       int half = idx & 1;
       int rest = idx >> 1;
-      int mloc = rest & 255;
-      int ni = rest >> 8;
+      int mloc = rest % MC;
+      int ni = rest / MC;
       int m = mc + mloc;
       bf16x8 v = bf8_zero();
       int cb = co0 + ni * 16 + half * 8;
       if (m < m_end && cb < COp)
         v = *reinterpret_cast<const bf16x8*>(&dy[(long)m * COp + cb]);
-      *reinterpret_cast<bf16x8*>(&lds_img[ni][mloc][half * 8]) = v;
+      *reinterpret_cast<bf16x8*>(&plane(ni)[mloc * 16 + half * 8]) = v;
     }
     __syncthreads();
 
     // hoist this wave's 8 dy B-fragments before the planes are reused
-    bf16x8 bfr[8];
+    bf16x8 bfr[MSTEPS];
 #pragma unroll
-    for (int s = 0; s < 8; ++s) bfr[s] = tr_read_8m(&lds_img[wid][s * 32][0], l);
+    for (int s = 0; s < MSTEPS; ++s)
+      bfr[s] = tr_read_8m(plane(wid) + s * 32 * 16, l);
     if (db_ws != nullptr && blockIdx.x == 0) {
 #pragma unroll
-      for (int s = 0; s < 8; ++s)
+      for (int s = 0; s < MSTEPS; ++s)
 #pragma unroll
         for (int j = 0; j < 8; ++j) db_part += to_f32(bfr[s][j]);
     }
@@ -586,7 +592,7 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_rs_bf16_k(
     // A-image build for one tap: planes (buf*2, buf*2+1) = ci halves
     auto build = [&](int tap, int buf) {
       int kh = tap / 3 - 1, kw = tap % 3 - 1;  // tap offsets (pad 1)
-      for (int idx = tid; idx < 256 * 4; idx += 256) {
+      for (int idx = tid; idx < MC * 4; idx += 256) {
         int g8 = idx & 3;
         int mloc = idx >> 2;
         int mm = lds_mm[mloc];
@@ -597,7 +603,7 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_rs_bf16_k(
           v = *reinterpret_cast<const bf16x8*>(&lds_reg[(px << 5) + g8 * 8]);
         }
         *reinterpret_cast<bf16x8*>(
-            &lds_img[buf * 2 + (g8 >> 1)][mloc][(g8 & 1) * 8]) = v;
+            &plane(buf * 2 + (g8 >> 1))[mloc * 16 + (g8 & 1) * 8]) = v;
       }
     };
 
@@ -610,8 +616,8 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_rs_bf16_k(
 #pragma unroll
       for (int f = 0; f < 2; ++f) {
 #pragma unroll
-        for (int s = 0; s < 8; ++s) {
-          bf16x8 a = tr_read_8m(&lds_img[buf * 2 + f][s * 32][0], l);
+        for (int s = 0; s < MSTEPS; ++s) {
+          bf16x8 a = tr_read_8m(plane(buf * 2 + f) + s * 32 * 16, l);
           acc[f][tap] = mfma16x16x32bf(a, bfr[s], acc[f][tap]);
         }
       }
