@@ -503,9 +503,11 @@ constexpr int WGRAD_BF16_KTILE = 4 * 2 * 16;  // KT=2 -> 128
 // (~310 MB on layer2 vs ~925) and the MFMA:gather ratio flips the kernel
 // from L2-latency-bound to matrix-core-bound.
 // ---------------------------------------------------------------------------
-constexpr int WGRS_PX = 464;  // region capacity: max (255/OW+4)*(OW+2), OW<=56
-
-template <int MC>
+// S=2 extension (the layer2/3/4 block-0 3x3 s2 convs): requires
+// H==2*OH, W==2*OW so the input global row for output row g is the LINEAR
+// 2g+kh-1 — the same contiguous-row staging then works with a 2x-wide
+// region and the tap predicate S*oy+kh in [0, S*OH).
+template <int MC, int S = 1>
 __global__ __launch_bounds__(256) void conv2d_wgrad_rs_bf16_k(
     const bf16* __restrict__ dy, const bf16* __restrict__ x,
     float* __restrict__ dw_ws, float* __restrict__ db_ws, int CIp, int CI,
@@ -517,9 +519,12 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_rs_bf16_k(
   const int wid = tid >> 6;
   const int l = tid & 63;
   const int lm = l & 15;
-  const int W2 = OW + 2;
+  const int Wi = S * OW;   // input width (host guarantees W == S*OW)
+  const int W2 = Wi + 2;
+  // region capacity: worst-case rows*(Wi+2) over the routed OW range
+  constexpr int PX = (S == 2) ? 528 : (MC <= 64 ? 296 : MC <= 128 ? 352 : 464);
 
-  __shared__ bf16 lds_reg[WGRS_PX * 32];      // [px][32ci]
+  __shared__ bf16 lds_reg[PX * 32];           // [px][32ci]
   // dy planes then A-image slots; +16-elem stagger per plane so the build's
   // paired-plane vec8 writes land on different bank columns
   __shared__ bf16 lds_img_raw[4 * MC * 16 + 3 * 16];
@@ -534,16 +539,16 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_rs_bf16_k(
   float db_part = 0.f;    // bias grad: colsum of this wave's dy fragments
 
   for (int mc = m_begin; mc < m_end; mc += MC) {
-    const int g0 = mc / OW;               // first global row of the chunk
+    const int g0 = mc / OW;               // first OUTPUT global row of chunk
     const int mlast = min(mc + MC - 1, m_end - 1);
-    const int rows = mlast / OW - g0 + 3;  // incl. +-1 halo
+    const int rows = S * (mlast / OW - g0) + 3;  // input rows incl. +-1 halo
     __syncthreads();
     // per-m map: region pixel index + output row (for tap predicates)
     if (tid < MC) {
       int m = mc + tid;
       int mm = m < m_end ? m : mc;
       int g = mm / OW, ox = mm % OW;
-      int px = (g - g0 + 1) * W2 + ox + 1;
+      int px = (S * (g - g0) + 1) * W2 + S * ox + 1;
       int oy = g % OH;
       lds_mm[tid] = (px << 10) | (m < m_end ? oy + 1 : 0);  // 0 => dead m
     }
@@ -553,12 +558,12 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_rs_bf16_k(
       int p = idx >> 2;
       int c = p % W2;
       int r = p / W2;
-      int g = g0 - 1 + r;
+      int gi = S * g0 - 1 + r;             // input global row
       int iw = c - 1;
       bf16x8 v = bf8_zero();
-      if (g >= 0 && g < (M / OW) && iw >= 0 && iw < OW)
+      if (gi >= 0 && gi < S * (M / OW) && iw >= 0 && iw < Wi)
         v = *reinterpret_cast<const bf16x8*>(
-            &x[((long)g * OW + iw) * CIp + cic0 + g8 * 8]);
+            &x[((long)gi * Wi + iw) * CIp + cic0 + g8 * 8]);
       *reinterpret_cast<bf16x8*>(&lds_reg[(p << 5) + g8 * 8]) = v;
     }
     // dy planes [ni][256][16co]
@@ -598,7 +603,7 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_rs_bf16_k(
         int mm = lds_mm[mloc];
         int oy = (mm & 1023) - 1;
         bf16x8 v = bf8_zero();
-        if (oy >= 0 && oy + kh >= 0 && oy + kh < OH) {
+        if (oy >= 0 && S * oy + kh >= 0 && S * oy + kh < S * OH) {
           int px = (mm >> 10) + kh * W2 + kw;
           v = *reinterpret_cast<const bf16x8*>(&lds_reg[(px << 5) + g8 * 8]);
         }
