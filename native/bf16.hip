@@ -1548,6 +1548,168 @@ __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
 }
 
 // ---------------------------------------------------------------------------
+// Region-stationary STEM wgrad (7x7 s2, <=4 real input channels, pad 3,
+// H==2*OH): the generic engine's im2col gather reads M*K*2B ~ 2.5 GB here
+// (49x the raw input) — the single most expensive kernel of the ResNet
+// step at 918 us. Channels pack to 4 (as in the stem fwd) so one tap ROW
+// is k=(kw*4+ci) padded to 32 = two 16-k fragments; a block owns all
+// 7x32 k and 64 co, marching m in 96-pixel chunks of ONE output row
+// (OW >= MC guarantees a chunk spans <= 2 rows -> 9 input rows + pad
+// region, 17 KB). dy fragments hoist to VGPRs once per chunk (tap-row
+// invariant), per-kh A-images build LDS->LDS into the vacated dy planes
+// (b64 pairs: the 8-byte pixel stride leaves vec8 alignment only at even
+// columns). Same tr16 fragment recipe and partial-slab fold as the other
+// wgrad engines.
+// ---------------------------------------------------------------------------
+template <int MC>
+__global__ __launch_bounds__(256) void conv2d_wgrad_stem_bf16_k(
+    const bf16* __restrict__ dy, const bf16* __restrict__ x,
+    float* __restrict__ dw_ws, float* __restrict__ db_ws, int CI, int COp,
+    int CO, int OH, int OW, int M, int Kcan, int m_per_split) {
+  const int co0 = blockIdx.y * 64;
+  const int ms = blockIdx.z;
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6;
+  const int l = tid & 63;
+  const int lm = l & 15;
+  const int Wi = 2 * OW;
+  const int W2 = Wi + 6;  // 3 pad cols each side
+  constexpr int PXROWS = 9, PXCOLS = 262;  // OW <= 128
+
+  __shared__ bf16 lds_reg[PXROWS * PXCOLS * 4];  // [row][col][4ci]
+  __shared__ bf16 lds_img_raw[4 * MC * 16 + 3 * 16];
+  __shared__ int lds_mm[MC];  // packed (px<<8 | oy+1)
+  auto plane = [&](int p) { return &lds_img_raw[p * (MC * 16 + 16)]; };
+  constexpr int MSTEPS = MC / 32;
+
+  const int m_begin = ms * m_per_split;
+  const int m_end = min(M, m_begin + m_per_split);
+
+  f32x4 acc[2][7] = {};  // [kf][kh]
+  float db_part = 0.f;
+
+  for (int mc = m_begin; mc < m_end; mc += MC) {
+    const int g0 = mc / OW;
+    const int mlast = min(mc + MC - 1, m_end - 1);
+    const int rows = 2 * (mlast / OW - g0) + 7;
+    __syncthreads();
+    if (tid < MC) {
+      int m = mc + tid;
+      int mm = m < m_end ? m : mc;
+      int g = mm / OW, ox = mm % OW;
+      int px = (2 * (g - g0)) * PXCOLS + 2 * ox;  // kh/kw offsets added later
+      int oy = g % OH;
+      lds_mm[tid] = (px << 8) | (m < m_end ? oy + 1 : 0);
+    }
+    // x region rows gi in [2*g0-3, ...]: 4-packed channels, b64 stores
+    for (int idx = tid; idx < rows * W2; idx += 256) {
+      int c = idx % W2;
+      int r = idx / W2;
+      int gi = 2 * g0 - 3 + r;
+      int iw = c - 3;
+      bf16x4 v = {};
+      if (gi >= 0 && gi < 2 * (M / OW) && iw >= 0 && iw < Wi)
+        v = *reinterpret_cast<const bf16x4*>(&x[((long)gi * Wi + iw) * 8]);
+      *reinterpret_cast<bf16x4*>(&lds_reg[(r * PXCOLS + c) * 4]) = v;
+    }
+    // dy planes [ni][MC][16co]
+    for (int idx = tid; idx < 4 * MC * 2; idx += 256) {
+      int half = idx & 1;
+      int rest = idx >> 1;
+      int mloc = rest % MC;
+      int ni = rest / MC;
+      int m = mc + mloc;
+      bf16x8 v = bf8_zero();
+      int cb = co0 + ni * 16 + half * 8;
+      if (m < m_end && cb < COp)
+        v = *reinterpret_cast<const bf16x8*>(&dy[(long)m * COp + cb]);
+      *reinterpret_cast<bf16x8*>(&plane(ni)[mloc * 16 + half * 8]) = v;
+    }
+    __syncthreads();
+
+    bf16x8 bfr[MSTEPS];
+#pragma unroll
+    for (int s = 0; s < MSTEPS; ++s)
+      bfr[s] = tr_read_8m(plane(wid) + s * 32 * 16, l);
+    if (db_ws != nullptr) {
+#pragma unroll
+      for (int s = 0; s < MSTEPS; ++s)
+#pragma unroll
+        for (int j = 0; j < 8; ++j) db_part += to_f32(bfr[s][j]);
+    }
+    __syncthreads();
+
+    // A-image build for tap row kh: entry (m, kw*4+ci); two b64 region
+    // reads (adjacent columns) make one 16-byte image write
+    auto build = [&](int kh, int buf) {
+      for (int idx = tid; idx < MC * 4; idx += 256) {
+        int g8 = idx & 3;          // (plane f = g8>>1, half = g8&1)
+        int mloc = idx >> 2;
+        int mm = lds_mm[mloc];
+        int oy = (mm & 255) - 1;
+        bf16x4 v0 = {}, v1 = {};
+        int iy = 2 * oy + kh - 3;
+        if (oy >= 0 && iy >= 0 && iy < 2 * OH) {
+          // region row = 2*(g-g0)+kh, col = 2*ox + kw with kw = g8*2 (+1)
+          int base = (mm >> 8) + kh * PXCOLS + (g8 << 1);
+          v0 = *reinterpret_cast<const bf16x4*>(&lds_reg[base * 4]);
+          v1 = *reinterpret_cast<const bf16x4*>(&lds_reg[(base + 1) * 4]);
+        }
+        bf16x8 v;
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          v[j] = v0[j];
+          v[4 + j] = v1[j];
+        }
+        *reinterpret_cast<bf16x8*>(
+            &plane((g8 >> 1) * 2 + buf)[mloc * 16 + (g8 & 1) * 8]) = v;
+      }
+    };
+
+    build(0, 0);
+    __syncthreads();
+#pragma unroll
+    for (int kh = 0; kh < 7; ++kh) {
+      int buf = kh & 1;
+      if (kh + 1 < 7) build(kh + 1, buf ^ 1);
+#pragma unroll
+      for (int f = 0; f < 2; ++f) {
+#pragma unroll
+        for (int s = 0; s < MSTEPS; ++s) {
+          bf16x8 a = tr_read_8m(plane(f * 2 + buf) + s * 32 * 16, l);
+          acc[f][kh] = mfma16x16x32bf(a, bfr[s], acc[f][kh]);
+        }
+      }
+      __syncthreads();
+    }
+  }
+
+  // slab: D row (lk*4+reg) = k-local within the kf frag, col (lm) = co
+  float* slab = dw_ws + (long)ms * (CO * (long)Kcan);
+  const int lk = l >> 4;
+#pragma unroll
+  for (int f = 0; f < 2; ++f) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      int kidx = f * 16 + lk * 4 + reg;  // kw*4+ci
+      int kw = kidx >> 2, ci = kidx & 3;
+      if (kw >= 7 || ci >= CI) continue;
+      int co = co0 + wid * 16 + lm;
+      if (co >= CO) continue;
+#pragma unroll
+      for (int kh = 0; kh < 7; ++kh)
+        slab[(long)co * Kcan + (long)ci * 49 + kh * 7 + kw] = acc[f][kh][reg];
+    }
+  }
+  if (db_ws != nullptr) {
+    db_part += __shfl_down(db_part, 16);
+    db_part += __shfl_down(db_part, 32);
+    int co = co0 + wid * 16 + lm;
+    if (lk == 0 && co < CO) db_ws[(long)ms * CO + co] = db_part;
+  }
+}
+
+// ---------------------------------------------------------------------------
 // LDS-tiled stride-2 DGRAD (parity classes): for stride 2, dx pixel
 // (2y+p, 2x+q) only receives taps of matching parity, and per class the
 // whole dgrad collapses to a stride-1 2x2 conv over dy:
