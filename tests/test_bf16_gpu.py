@@ -73,6 +73,7 @@ def test_conv2d_bf16_fwd(N, CI, H, W, CO, KH, KW, stride, pad):
         (2, 3, 18, 18, 7, 3, 3, 1, 1),
         (2, 8, 16, 16, 16, 3, 3, 2, 1),
         (2, 64, 27, 27, 128, 3, 3, 2, 1),  # tiled s2 dgrad (3x3 pad1)
+        (2, 64, 28, 28, 128, 3, 3, 2, 1),  # H==2*OH: region wgrad S=2 path
         (2, 64, 15, 15, 128, 3, 3, 2, 1),  # tiled s2 dgrad, odd H (ragged classes)
         (2, 64, 14, 14, 128, 1, 1, 2, 0),  # tiled s2 dgrad (1x1 downsample)
         (2, 3, 192, 192, 64, 7, 7, 2, 3),  # stem region wgrad (7x7 s2)
