@@ -1,8 +1,10 @@
 #!/usr/bin/env python3
 """Per-op A/B microbenchmark on the GPU box (guide §5.4: within-probe timing).
 
-Times the bf16 conv engines (gather vs tiled) on the model shapes so routing
-decisions are measured, not guessed.
+Times the bf16 conv engines on the model shapes so routing decisions are
+measured, not guessed. Columns call the HOST dispatchers, so "fwd/dgrad/
+wgrad" always measure the currently-routed engine; the explicit
+conv_tiled_s1 columns force the tiled engine for comparison.
 
     gpurun -- 'python tools/microbench.py > gpurun_out/micro.txt'
 """
@@ -60,7 +62,10 @@ def main():
                                                   stride, pad, True))
         print(f"{name:14} {fg:11.1f} {ft:10.1f} {dg:10.1f} {dt:9.1f} {wg:8.1f}")
 
-    # stride-2 shapes (gather only for now)
+    # stride-2 shapes: the host dispatchers route these to the stem /
+    # S=2 tiled / parity-class engines automatically, so the numbers below
+    # are the ROUTED engine. A/B the region-wgrad chunk sizes with
+    # DMA_WGRS_MC=32|64|128|256 and DMA_WGST_MC=32|64|96 in the env.
     s2 = [
         ("rn.stem7x7s2", 64, 224, 224, 8, 3, 64, 64, 7, 2, 3),
         ("rn.l2.ds3x3s2", 256, 56, 56, 64, 64, 128, 128, 3, 2, 1),
