@@ -79,6 +79,11 @@ class GraphedStep:
             for grads, _ in zip(slot_grads, self.slots):
                 for p, gr in zip(params, grads):
                     p.grad = gr
+                    # capture RECORDED the producing kernels without running
+                    # them, so the pool tensors hold uninitialized memory —
+                    # zero them so this table-building step is a benign
+                    # momentum-decay update, not a garbage-gradient one
+                    gr.zero_()
                 self.opt.step()
                 og = torch.cuda.CUDAGraph()
                 with torch.cuda.graph(og, pool=pool):
