@@ -113,8 +113,17 @@ def main():
             print(f"# hipGraph capture failed ({type(e).__name__}: {e}); "
                   "falling back to eager launches", flush=True)
             graphed = None
-            if ddp is not None:
-                ddp._install_hooks()  # restore comm-in-backward overlap
+        if world > 1:
+            # all ranks must take the SAME step path: a mixed graphed/eager
+            # fleet could issue the bucket collectives in different orders
+            ok = torch.tensor([1.0 if graphed is not None else 0.0],
+                              device=device if dist.get_backend() == "nccl"
+                              else "cpu")
+            dist.all_reduce(ok, op=dist.ReduceOp.MIN)
+            if ok.item() < 1.0:
+                graphed = None
+        if graphed is None and ddp is not None:
+            ddp._install_hooks()  # restore comm-in-backward overlap
 
     def one_step():
         x, t = next(batches)

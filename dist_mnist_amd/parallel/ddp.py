@@ -114,6 +114,8 @@ class DistributedDataParallel(nn.Module):
     def _install_hooks(self):
         if not self.enabled:
             return
+        if self._grad_accum_hooks:  # idempotent: never double-register
+            return
         for b in self.buckets:
             for p in b.params:
                 h = p.register_post_accumulate_grad_hook(self._make_hook(b))
