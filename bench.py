@@ -104,24 +104,21 @@ def main():
     step_idx = [0]
     if use_graphs:
         from dist_mnist_amd.runtime import GraphedStep
+        from dist_mnist_amd.runtime.graphs import resync_after_fallback
 
         x0, t0 = next(batches)
         ddp = model if isinstance(model, DistributedDataParallel) else None
         try:
+            # GraphedStep raises CaptureFailed on ALL ranks after a collective
+            # agreement (before any constructor collective), so the fleet
+            # takes one consistent step path.
             graphed = GraphedStep(model, opt, x0, t0, ddp=ddp)
         except Exception as e:  # capture failure must never kill the bench
             print(f"# hipGraph capture failed ({type(e).__name__}: {e}); "
                   "falling back to eager launches", flush=True)
             graphed = None
-        if world > 1:
-            # all ranks must take the SAME step path: a mixed graphed/eager
-            # fleet could issue the bucket collectives in different orders
-            ok = torch.tensor([1.0 if graphed is not None else 0.0],
-                              device=device if dist.get_backend() == "nccl"
-                              else "cpu")
-            dist.all_reduce(ok, op=dist.ReduceOp.MIN)
-            if ok.item() < 1.0:
-                graphed = None
+            # warmup drifted params rank-locally; resync before eager training
+            resync_after_fallback(opt, ddp)
         if graphed is None and ddp is not None:
             ddp._install_hooks()  # restore comm-in-backward overlap
 

@@ -102,7 +102,13 @@ def partition_dataset(rank: int, world_size: int, global_batch: int = GLOBAL_BAT
     bsz = int(global_batch / float(world_size))
     partition_sizes = [1.0 / world_size for _ in range(world_size)]
     partition = DataPartitioner(dataset, partition_sizes, seed=seed).use(rank)
-    loader = torch.utils.data.DataLoader(partition, batch_size=bsz, shuffle=True)
+    # The reference leaves the per-epoch DataLoader shuffle unseeded (a
+    # parity wart); here it is seeded per rank so run-to-run training order
+    # is deterministic, matching the repo's determinism story everywhere
+    # else. The shard split above is already seed-1234 reference behavior.
+    gen = torch.Generator().manual_seed(seed * 1000003 + rank)
+    loader = torch.utils.data.DataLoader(partition, batch_size=bsz, shuffle=True,
+                                         generator=gen)
     return loader, bsz
 
 

@@ -27,28 +27,32 @@ class SGD(torch.optim.Optimizer):
     def __init__(self, params, lr=0.01, momentum=0.0):
         defaults = dict(lr=lr, momentum=momentum)
         super().__init__(params, defaults)
-        self._table = None  # lazily built device pointer table
+        # Pointer tables keyed by (param_group_index, tuple-of-grad-pointers).
+        # Entries are NEVER evicted: hipGraph-captured optimizer steps
+        # (runtime/graphs.py) record the raw device addresses of a slot's
+        # (ptrs_t, chunks_t) tensors, so freeing a superseded table would
+        # leave an earlier slot's captured graph replaying over freed memory.
+        # A training run only ever has a handful of distinct grad-pointer
+        # sets (one per graph slot), so the cache stays tiny.
+        self._tables: dict = {}
 
     CHUNK = 4096  # elements per block in the fused kernel
 
-    def _build_table(self, group):
+    def _build_table(self, params):
         """Device tables for the one-launch fused step:
         ptrs[t] = (param_ptr, grad_ptr, momentum_ptr, numel);
         chunks[b] = (tensor_idx, elem_offset) — one block per chunk.
 
-        Pointers are stable for the life of the training run: grads are bucket
-        views installed once by the DDP engine (parallel/ddp.py) or allocated
-        here on first step; momentum buffers are allocated here.
+        `params` is exactly the filtered list step() launches over, so the
+        cache key (grad pointers of that list) and the table rows always
+        describe the same tensors.
         """
         ptrs = []
         chunks = []
-        params = [p for p in group["params"] if p.requires_grad]
         for ti, p in enumerate(params):
             st = self.state[p]
             if "momentum_buffer" not in st:
                 st["momentum_buffer"] = torch.zeros_like(p, memory_format=torch.contiguous_format)
-            if p.grad is None:
-                p.grad = torch.zeros_like(p, memory_format=torch.contiguous_format)
             assert p.is_contiguous() and p.grad.is_contiguous()
             ptrs.append(
                 (p.data_ptr(), p.grad.data_ptr(), st["momentum_buffer"].data_ptr(), p.numel())
@@ -58,7 +62,7 @@ class SGD(torch.optim.Optimizer):
         dev = params[0].device
         ptrs_t = torch.tensor(ptrs, dtype=torch.int64, device=dev)
         chunks_t = torch.tensor(chunks, dtype=torch.int32, device=dev)
-        return ptrs_t, chunks_t, [p.grad.data_ptr() for p in params]
+        return ptrs_t, chunks_t
 
     @torch.no_grad()
     def step(self, closure=None):
@@ -66,18 +70,17 @@ class SGD(torch.optim.Optimizer):
         if closure is not None:
             with torch.enable_grad():
                 loss = closure()
-        for group in self.param_groups:
+        for gi, group in enumerate(self.param_groups):
             params = [p for p in group["params"] if p.requires_grad and p.grad is not None]
             if not params:
                 continue
             if params[0].is_cuda:
-                cached = self._table
-                if (
-                    cached is None
-                    or cached[2] != [p.grad.data_ptr() for p in params if p.grad is not None]
-                ):
-                    self._table = self._build_table(group)
-                ptrs_t, chunks_t, _ = self._table
+                key = (gi, tuple(p.grad.data_ptr() for p in params))
+                table = self._tables.get(key)
+                if table is None:
+                    table = self._build_table(params)
+                    self._tables[key] = table
+                ptrs_t, chunks_t = table
                 native.ext().fused_sgd_step(
                     ptrs_t, chunks_t, self.CHUNK, group["lr"], group["momentum"]
                 )

@@ -19,6 +19,32 @@ import torch
 import torch.distributed as dist
 
 
+class CaptureFailed(RuntimeError):
+    """hipGraph capture failed — raised on ALL ranks after a collective
+    agreement, so a mixed graphed/eager fleet can never form (the eager and
+    graphed paths issue bucket collectives in different orders)."""
+
+
+def resync_after_fallback(opt, ddp):
+    """Restore rank-identical state after a capture failure at world>1.
+
+    GraphedStep's warmup steps run without gradient sync, so ranks drift;
+    on the graphed path the constructor re-broadcasts params+momentum, but
+    on the eager fallback the caller must resync instead. Different ranks
+    may have failed at different points (momentum buffers may exist on some
+    ranks only), so the momentum state is CLEARED everywhere — zeros are
+    trivially rank-identical — rather than broadcast, and the pointer-table
+    cache with it (its tables reference the freed buffers)."""
+    if ddp is None or not ddp.enabled:
+        return
+    with torch.no_grad():
+        for p in ddp.module.parameters():
+            dist.broadcast(p.data, src=0, group=ddp.pg)
+    opt.state.clear()
+    if hasattr(opt, "_tables"):
+        opt._tables.clear()
+
+
 class GraphedStep:
     def __init__(self, model, optimizer, sample_x, sample_t, ddp=None,
                  n_slots: int = 2, warmup: int = 3):
@@ -35,6 +61,42 @@ class GraphedStep:
         for _ in range(n_slots):
             self.slots.append((sample_x.clone(), sample_t.clone()))
 
+        # The whole warmup+capture region issues NO collectives (comm hooks
+        # are disabled above), so a rank that fails anywhere in it can still
+        # reach the agreement point below with a matching collective
+        # sequence. Constructor collectives (param/momentum re-broadcast)
+        # happen only AFTER all ranks agree capture succeeded.
+        err = None
+        try:
+            self._capture(warmup)
+        except Exception as e:  # noqa: BLE001 — agreement must still run
+            err = e
+        if ddp is not None and ddp.enabled:
+            dev = (self.device if dist.get_backend(ddp.pg) == "nccl" else "cpu")
+            ok = torch.tensor([0.0 if err is not None else 1.0], device=dev)
+            dist.all_reduce(ok, op=dist.ReduceOp.MIN, group=ddp.pg)
+            if ok.item() < 1.0:
+                raise CaptureFailed(
+                    f"capture failed on at least one rank "
+                    f"(this rank: {err!r})") from err
+        elif err is not None:
+            raise CaptureFailed(str(err)) from err
+
+        # warmup steps ran WITHOUT gradient sync (hooks disabled, rank-local
+        # data) so ranks drifted; re-broadcast params AND optimizer momentum
+        # in place — the captured graphs hold raw pointers, so in-place
+        # copies are safe and keep ranks bitwise-identical from here on.
+        if self.ddp is not None and self.ddp.enabled:
+            with torch.no_grad():
+                for p in self.ddp.module.parameters():
+                    dist.broadcast(p.data, src=0, group=self.ddp.pg)
+                for st in self.opt.state.values():
+                    buf = st.get("momentum_buffer")
+                    if buf is not None:
+                        dist.broadcast(buf, src=0, group=self.ddp.pg)
+            torch.cuda.synchronize(self.device)
+
+    def _capture(self, warmup: int):
         side = torch.cuda.Stream(device=self.device)
         # warmup on a side stream (capture requires non-default-stream history)
         side.wait_stream(torch.cuda.current_stream(self.device))
@@ -100,20 +162,6 @@ class GraphedStep:
             # once before the first replay (assign-mode slots need none:
             # their backward overwrites)
             self._zero_grads()
-
-        # warmup steps ran WITHOUT gradient sync (hooks disabled, rank-local
-        # data) so ranks drifted; re-broadcast params AND optimizer momentum
-        # in place — the captured graphs hold raw pointers, so in-place
-        # copies are safe and keep ranks bitwise-identical from here on.
-        if self.ddp is not None and self.ddp.enabled:
-            with torch.no_grad():
-                for p in self.ddp.module.parameters():
-                    dist.broadcast(p.data, src=0, group=self.ddp.pg)
-                for st in self.opt.state.values():
-                    buf = st.get("momentum_buffer")
-                    if buf is not None:
-                        dist.broadcast(buf, src=0, group=self.ddp.pg)
-            torch.cuda.synchronize(self.device)
 
     # -- eager pieces (also used during warmup/capture) ---------------------
     def _zero_grads(self):

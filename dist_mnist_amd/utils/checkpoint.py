@@ -13,7 +13,14 @@ import torch
 
 def save_checkpoint(model, path: str):
     module = model.module if hasattr(model, "module") else model
-    sd = {k: v.detach().to("cpu", torch.float32) for k, v in module.state_dict().items()}
+    # fp32 cast applies to floating tensors only: integer buffers (e.g.
+    # ResNet BatchNorm num_batches_tracked, int64) keep their dtype so the
+    # file stays interchangeable with stock torch state_dicts.
+    sd = {
+        k: (v.detach().to("cpu", torch.float32) if v.is_floating_point()
+            else v.detach().to("cpu"))
+        for k, v in module.state_dict().items()
+    }
     torch.save(sd, path)
 
 

@@ -434,8 +434,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dropout_bf16_fwd", &dropout_bf16_fwd);
   m.def("dropout_bf16_bwd", &dropout_bf16_bwd);
   m.def("comm_unique_id", &comm_unique_id);
+  m.def("comm_id_size", &comm_id_size);
   m.def("comm_init", &comm_init);
+  m.def("comm_set_timeout", &comm_set_timeout);
+  m.def("comm_aborted", &comm_aborted);
   m.def("comm_all_reduce_sum", &comm_all_reduce_sum);
+  m.def("comm_abort", &comm_abort);
   m.def("comm_destroy", &comm_destroy);
   m.def("comm_rank", &comm_rank);
   m.def("comm_world", &comm_world);

@@ -199,6 +199,24 @@ def test_dropout2d_bf16_channelwise():
     assert 0.4 < live.float().mean().item() < 0.6
 
 
+def test_dropout2d_bf16_backward_matches_mask():
+    """Backward cross-check vs the torch fp32 semantics: dx = dy * mask / (1-p)
+    with the SAME per-(sample,channel) mask the forward drew."""
+    x = torch.randn(32, 8, 8, 16, device=DEV).to(torch.bfloat16).requires_grad_(True)
+    y = fb.dropout_bf16(x, p=0.5, training=True, channelwise=True)
+    live = (y.float() != 0).any(dim=1, keepdim=True).any(dim=2, keepdim=True)
+    dy = torch.randn_like(y.float()).to(torch.bfloat16)
+    y.backward(dy)
+    # torch reference of the same op given the drawn mask
+    ref = dy.float() * live.float() * 2.0
+    torch.testing.assert_close(x.grad.float(), ref.to(torch.bfloat16).float(),
+                               rtol=1e-2, atol=1e-3)
+    # forward consistency: y == x * mask * 2 for the same mask
+    torch.testing.assert_close(
+        y.float(), (x.detach().float() * live.float() * 2.0)
+        .to(torch.bfloat16).float(), rtol=1e-2, atol=1e-3)
+
+
 def test_net_bf16_matches_fp32_reference():
     from dist_mnist_amd.models import Net
 

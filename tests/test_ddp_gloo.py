@@ -119,6 +119,45 @@ def _run_world(fn, world, port):
     assert all(s == "ok" for s in results.values()), results
 
 
+def _worker_resync_after_fallback(rank, world, port, q):
+    """resync_after_fallback (runtime/graphs.py) restores rank-identical
+    params + cleared momentum after a capture-failure eager fallback, even
+    when ranks drifted differently and hold DIFFERENT optimizer state."""
+    from dist_mnist_amd.ops import SGD
+    from dist_mnist_amd.parallel import DistributedDataParallel
+    from dist_mnist_amd.runtime.graphs import resync_after_fallback
+
+    _init(rank, world, port)
+    try:
+        torch.manual_seed(300)
+        ddp = DistributedDataParallel(Net().eval(), broadcast_params=False)
+        opt = SGD(ddp.module.parameters(), lr=0.05, momentum=0.5)
+
+        # drift: rank-local unsynced steps, rank 1 takes one MORE step so
+        # even the optimizer-state structure differs across ranks
+        g = torch.Generator().manual_seed(700 + rank)
+        x = torch.randn(8, 1, 28, 28, generator=g)
+        t = torch.randint(0, 10, (8,), generator=g)
+        for _ in range(1 + rank):
+            ddp.zero_grad()
+            _, loss = ddp(x, t)
+            loss.backward()  # NO sync: ranks drift
+            opt.step()
+
+        resync_after_fallback(opt, ddp)
+        assert len(opt.state) == 0  # momentum cleared everywhere
+
+        flat = torch.cat([p.detach().flatten() for p in ddp.module.parameters()])
+        gathered = [torch.empty_like(flat) for _ in range(world)]
+        dist.all_gather(gathered, flat)
+        torch.testing.assert_close(gathered[0], gathered[1], rtol=0, atol=0)
+        q.put((rank, "ok"))
+    except Exception as e:  # pragma: no cover
+        q.put((rank, f"FAIL: {type(e).__name__}: {e}"))
+    finally:
+        dist.destroy_process_group()
+
+
 @pytest.mark.timeout(180)
 def test_bucketed_matches_reference_semantics(free_port):
     _run_world(_worker_bucketed_vs_reference, 2, free_port)
@@ -127,3 +166,8 @@ def test_bucketed_matches_reference_semantics(free_port):
 @pytest.mark.timeout(180)
 def test_dp_training_equals_large_batch(free_port):
     _run_world(_worker_dp_equals_large_batch, 2, free_port)
+
+
+@pytest.mark.timeout(180)
+def test_resync_after_fallback_restores_rank_identity(free_port):
+    _run_world(_worker_resync_after_fallback, 2, free_port)

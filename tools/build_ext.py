@@ -18,7 +18,7 @@ import sysconfig
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
-def build(verbose: bool = True) -> str:
+def build(verbose: bool = True, asan: bool = False) -> str:
     import torch
     import torch.utils.cpp_extension as ce
 
@@ -42,6 +42,14 @@ def build(verbose: bool = True) -> str:
         "-fno-gpu-rdc",
         "-Wno-unused-result",
     ]
+    if asan:
+        # Host-side AddressSanitizer (SURVEY.md §5 race/sanitizer plan):
+        # instruments the C++ binding/host code only (-Xarch_host keeps the
+        # gfx950 device pass clean). Run python under the matching runtime:
+        #   LD_PRELOAD=$(hipcc -print-file-name=libclang_rt.asan-x86_64.so) \
+        #   ASAN_OPTIONS=detect_leaks=0 python -m pytest ...
+        # (leak detection off: torch/HIP runtime allocations dwarf ours).
+        cmd += ["-Xarch_host", "-fsanitize=address", "-shared-libsan", "-g"]
     cmd += list(getattr(ce, "COMMON_HIP_FLAGS", []))
     cmd += [f"-I{i}" for i in incs]
     cmd += [f"-L{d}" for d in libdirs]
@@ -54,6 +62,6 @@ def build(verbose: bool = True) -> str:
 
 
 if __name__ == "__main__":
-    build()
+    build(asan="--asan" in sys.argv)
     print("built dist_mnist_amd/_C.so")
     sys.exit(0)
