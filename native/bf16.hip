@@ -303,6 +303,171 @@ __global__ __launch_bounds__(256) void conv2d_dgrad_bf16_k(
 }
 
 // ---------------------------------------------------------------------------
+// Small-K gather conv (fwd and dgrad), NHWC bf16 — whole-weight LDS residency.
+// For LeNet-class shapes (K <= ~768, one or two 16-col tiles) the generic
+// gather engines above are dominated by their per-BK=32 chunk structure:
+// at K=400 that is 13 weight restages (each with a full k->(kh,kw,ci)
+// integer-division decode per element) and 26 __syncthreads per block for
+// only ~4 MFMAs of work per barrier pair — measured 3.2-3.5% MFMA-busy
+// (profiles/lenet_bf16_pmc_counters.txt).  Here the ENTIRE repacked weight
+// image [NW*16][K] plus the per-8k decode tables stage ONCE (one decode per
+// element total), then the k loop runs with NO barriers and NO LDS writes:
+// per k-group each lane issues one predicated 16-B x gather (L1-resident at
+// these image sizes: a block's rows span ~2 images = ~9 KB) and MW*NW MFMAs.
+// DGRAD=true consumes dy as A with the transposed/flipped weight decode of
+// conv2d_dgrad_bf16_k (stride-1 only; s2 stays on the parity engines).
+// ---------------------------------------------------------------------------
+template <bool DGRAD, bool HAS_BIAS, bool RELU, bool BOUNDS, int MW, int NW>
+__global__ __launch_bounds__(256) void conv_smallk_bf16_k(
+    const bf16* __restrict__ x, const float* __restrict__ w,
+    const float* __restrict__ bias, bf16* __restrict__ y,
+    int CIp, int CI, int H, int W, int COp, int CO, int KH, int KW,
+    int OH, int OW, int pad, int M, int K) {
+  // dynamic LDS: [NW*16][K+8] bf16 weight image + K/8-entry decode tables
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int Krow = K + 8;                       // K is a multiple of 8
+  bf16* lds_w = reinterpret_cast<bf16*>(smem);
+  int* lds_dec = reinterpret_cast<int*>(smem + ((NW * 16 * Krow * 2 + 15) & ~15));
+  int* lds_off = lds_dec + (K >> 3);            // fwd only: x offset per group
+
+  const int BM = 4 * MW * 16;
+  const int m0 = blockIdx.x * BM;
+  const int n0 = blockIdx.y * (NW * 16);        // co tile (fwd) / ci tile (dgrad)
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6;
+  const int l = tid & 63;
+  const int lm = l & 15;
+  const int lk = l >> 4;
+
+  const int KWC = KW * (DGRAD ? COp : CIp);
+  const int KHKW = KH * KW;
+
+  // ---- one-time stage: whole weight image + decode tables ----------------
+  for (int idx = tid; idx < NW * 16 * K; idx += 256) {
+    int row = idx / K;                          // co (fwd) / ci (dgrad)
+    int k = idx % K;
+    int kh = k / KWC, r = k % KWC;
+    float v = 0.f;
+    if (DGRAD) {
+      int kw = r / COp, co = r % COp;
+      if (co < CO && n0 + row < CI)
+        v = w[((co * CI) + n0 + row) * KHKW + kh * KW + kw];
+    } else {
+      int kw = r / CIp, ci = r % CIp;
+      if (ci < CI && n0 + row < CO)
+        v = w[((n0 + row) * CI + ci) * KHKW + kh * KW + kw];
+    }
+    lds_w[row * Krow + k] = to_bf16(v);
+  }
+  for (int g = tid; g < (K >> 3); g += 256) {
+    int k = g * 8;
+    int kh = k / KWC, r = k % KWC;
+    if (DGRAD) {
+      int kw = r / COp, c0 = r % COp;
+      lds_dec[g] = (kh << 20) | (kw << 10) | c0;
+    } else {
+      int kw = r / CIp, ci0 = r % CIp;
+      lds_dec[g] = (kh << 16) | kw;
+      lds_off[g] = (kh * W + kw) * CIp + ci0;
+    }
+  }
+  __syncthreads();
+
+  // ---- per-lane m decode --------------------------------------------------
+  int xbase[MW], ih0[MW], iw0[MW], nn[MW];
+  bool mval[MW];
+#pragma unroll
+  for (int mi = 0; mi < MW; ++mi) {
+    int m = m0 + (wid * MW + mi) * 16 + lm;
+    mval[mi] = m < M;
+    int mm = mval[mi] ? m : 0;
+    if (DGRAD) {
+      nn[mi] = mm / (H * W);
+      int r = mm % (H * W);
+      ih0[mi] = r / W;                          // ih
+      iw0[mi] = r % W;                          // iw
+    } else {
+      int n = mm / (OH * OW), r = mm % (OH * OW);
+      int oh = r / OW, ow = r % OW;
+      ih0[mi] = oh - pad;
+      iw0[mi] = ow - pad;
+      xbase[mi] = ((n * H + ih0[mi]) * W + iw0[mi]) * CIp;
+    }
+  }
+
+  f32x4 acc[MW][NW] = {};
+
+  // ---- barrier-free k loop ------------------------------------------------
+  const int ngroups = K >> 3;
+  for (int kg = 0; kg < ngroups; kg += 4) {
+    bool gin = lk < (ngroups - kg);
+    int gsel = gin ? kg + lk : 0;
+    int kdec = lds_dec[gsel];
+    bf16x8 bw[NW];
+#pragma unroll
+    for (int ni = 0; ni < NW; ++ni)
+      bw[ni] = gin ? *reinterpret_cast<const bf16x8*>(
+                         &lds_w[(ni * 16 + lm) * Krow + gsel * 8])
+                   : bf8_zero();
+    if (DGRAD) {
+      int kh = kdec >> 20, kw = (kdec >> 10) & 0x3ff, c0g = kdec & 0x3ff;
+#pragma unroll
+      for (int mi = 0; mi < MW; ++mi) {
+        bf16x8 a = bf8_zero();
+        if (mval[mi] & gin) {
+          int tH = ih0[mi] + pad - kh;
+          int tW = iw0[mi] + pad - kw;
+          if (tH >= 0 && tH < OH && tW >= 0 && tW < OW)
+            a = *reinterpret_cast<const bf16x8*>(
+                &x[((long)(nn[mi] * OH + tH) * OW + tW) * COp + c0g]);
+        }
+#pragma unroll
+        for (int ni = 0; ni < NW; ++ni)
+          acc[mi][ni] = mfma16x16x32bf(a, bw[ni], acc[mi][ni]);
+      }
+    } else {
+      int koff = lds_off[gsel];
+      int kh = kdec >> 16, kw = kdec & 0xffff;
+#pragma unroll
+      for (int mi = 0; mi < MW; ++mi) {
+        bool ok = mval[mi] & gin;
+        if (BOUNDS) {
+          int ih = ih0[mi] + kh, iw = iw0[mi] + kw;
+          ok = ok && ih >= 0 && ih < H && iw >= 0 && iw < W;
+        }
+        bf16x8 a = ok ? *reinterpret_cast<const bf16x8*>(&x[xbase[mi] + koff])
+                      : bf8_zero();
+#pragma unroll
+        for (int ni = 0; ni < NW; ++ni)
+          acc[mi][ni] = mfma16x16x32bf(a, bw[ni], acc[mi][ni]);
+      }
+    }
+  }
+
+  // ---- epilogue -----------------------------------------------------------
+  const int NC = DGRAD ? CIp : COp;             // padded col extent
+  const int NCv = DGRAD ? CI : CO;              // valid cols
+#pragma unroll
+  for (int mi = 0; mi < MW; ++mi) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      int m = m0 + (wid * MW + mi) * 16 + lk * 4 + reg;
+      if (m >= M) continue;
+      long yb = (long)m * NC;
+#pragma unroll
+      for (int ni = 0; ni < NW; ++ni) {
+        int c = n0 + ni * 16 + lm;
+        if (c >= NC) continue;
+        float v = (c < NCv) ? acc[mi][ni][reg] : 0.f;
+        if (HAS_BIAS && c < NCv) v += bias[c];
+        if (RELU) v = fmaxf(v, 0.f);
+        y[yb + c] = to_bf16(v);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Conv2d backward-weight (wgrad) + bias, NHWC bf16 -> fp32 canonical dw/db.
 // Reduction over m. Waves split the k-tile (k-tile = 4*KT*16); the dy chunk
 // and the im2col(x) chunk are staged NON-transposed as [m][16] sub-images
@@ -332,14 +497,14 @@ DEV bf16x8 tr_read_8m(const bf16* lds_img, int l) {
   return a;
 }
 
-template <bool BOUNDS, bool WANT_BIAS, int KT, int NW>
+template <bool BOUNDS, bool WANT_BIAS, int KT, int NW, int MC = 64>
 __global__ __launch_bounds__(256) void conv2d_wgrad_bf16_k(
     const bf16* __restrict__ dy, const bf16* __restrict__ x,
     float* __restrict__ dw_ws, float* __restrict__ db_ws,
     int CIp, int CI, int H, int W, int COp, int CO, int KH, int KW,
     int OH, int OW, int stride, int pad, int M, int K, int Kcan,
     int m_per_split) {
-  constexpr int MC = 64;          // m per chunk (shared by all 4 waves)
+  // MC = m per chunk (shared by all 4 waves)
   constexpr int NKF = 4 * KT;     // 16-row k-frag images per block
   const int k0 = blockIdx.x * (NKF * 16);
   const int co0 = blockIdx.y * (NW * 16);
@@ -1186,13 +1351,15 @@ __global__ __launch_bounds__(256) void conv_pool_stem_bf16_k(
   }
   __syncthreads();
 
-  long total = (long)N * OH2 * OW2 * (COp / 8);
+  // thread = (n, pool cell); loops over ALL co groups so the x tile loads
+  // once (round 1 re-loaded it per 8-co group), with bf16x2 vector row
+  // reads when CIp==1 (guide G13: scalar bf16 loads cost ~2x)
+  using bf16x2 = __attribute__((ext_vector_type(2))) __bf16;
+  long total = (long)N * OH2 * OW2;
   for (long t = (long)blockIdx.x * blockDim.x + tid; t < total;
        t += (long)gridDim.x * blockDim.x) {
-    int cg = t % (COp / 8);
-    long r = t / (COp / 8);
-    int ow2 = r % OW2;
-    long r2 = r / OW2;
+    int ow2 = t % OW2;
+    long r2 = t / OW2;
     int oh2 = r2 % OH2;
     int n = r2 / OH2;
 
@@ -1202,50 +1369,67 @@ __global__ __launch_bounds__(256) void conv_pool_stem_bf16_k(
     constexpr int TH = KH + 1, TW = KW + 1;
     float xt[CI * TH * TW];
     const bf16* xb = x + ((long)(n * H + oh2 * 2) * W + ow2 * 2) * CIp;
-#pragma unroll
-    for (int ci = 0; ci < CI; ++ci)
+    if (CI == 1 && TW % 2 == 0 && CIp == 1) {
+      // CIp==1 (runtime!) row is contiguous; base is 4-B aligned (even
+      // pixel index). Padded inputs (CIp=8, from nchw_to_nhwc_padded) take
+      // the strided scalar branch below.
 #pragma unroll
       for (int ty = 0; ty < TH; ++ty)
 #pragma unroll
-        for (int tx = 0; tx < TW; ++tx)
-          xt[(ci * TH + ty) * TW + tx] = to_f32(xb[(ty * W + tx) * CIp + ci]);
-
-    bf16x8 pout;
-    unsigned char am[8];
+        for (int tx = 0; tx < TW; tx += 2) {
+          bf16x2 v = *reinterpret_cast<const bf16x2*>(&xb[ty * W + tx]);
+          xt[ty * TW + tx] = to_f32(v[0]);
+          xt[ty * TW + tx + 1] = to_f32(v[1]);
+        }
+    } else {
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      int co = cg * 8 + j;
-      float acc[4];
-      float bval = (co < CO) ? bias[co] : 0.f;
-      acc[0] = acc[1] = acc[2] = acc[3] = bval;
-      if (co < CO) {
-        const float* wr = &lds_wf[co * Kc];
+      for (int ci = 0; ci < CI; ++ci)
 #pragma unroll
-        for (int ci = 0; ci < CI; ++ci)
+        for (int ty = 0; ty < TH; ++ty)
 #pragma unroll
-          for (int kh = 0; kh < KH; ++kh)
-#pragma unroll
-            for (int kw = 0; kw < KW; ++kw) {
-              float wv = wr[(ci * KH + kh) * KW + kw];
-              const float* xrow = &xt[(ci * TH + kh) * TW + kw];
-              acc[0] += wv * xrow[0];
-              acc[1] += wv * xrow[1];
-              acc[2] += wv * xrow[TW];
-              acc[3] += wv * xrow[TW + 1];
-            }
-      }
-      float m01 = fmaxf(acc[0], acc[1]), m23 = fmaxf(acc[2], acc[3]);
-      float mx = fmaxf(m01, m23);
-      unsigned char a;
-      if (mx == acc[0]) a = 0; else if (mx == acc[1]) a = 1;
-      else if (mx == acc[2]) a = 2; else a = 3;
-      pout[j] = to_bf16((co < CO) ? fmaxf(mx, 0.f) : 0.f);
-      am[j] = a;
+          for (int tx = 0; tx < TW; ++tx)
+            xt[(ci * TH + ty) * TW + tx] = to_f32(xb[(ty * W + tx) * CIp + ci]);
     }
-    long off = (((long)(n * OH2 + oh2) * OW2 + ow2) * COp) + cg * 8;
-    *reinterpret_cast<bf16x8*>(p + off) = pout;
-    *reinterpret_cast<unsigned long long*>(idx + off) =
-        *reinterpret_cast<unsigned long long*>(am);
+
+    long off0 = (((long)(n * OH2 + oh2) * OW2 + ow2) * COp);
+    for (int cg = 0; cg < COp / 8; ++cg) {
+      bf16x8 pout;
+      unsigned char am[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int co = cg * 8 + j;
+        float acc[4];
+        float bval = (co < CO) ? bias[co] : 0.f;
+        acc[0] = acc[1] = acc[2] = acc[3] = bval;
+        if (co < CO) {
+          const float* wr = &lds_wf[co * Kc];
+#pragma unroll
+          for (int ci = 0; ci < CI; ++ci)
+#pragma unroll
+            for (int kh = 0; kh < KH; ++kh)
+#pragma unroll
+              for (int kw = 0; kw < KW; ++kw) {
+                float wv = wr[(ci * KH + kh) * KW + kw];
+                const float* xrow = &xt[(ci * TH + kh) * TW + kw];
+                acc[0] += wv * xrow[0];
+                acc[1] += wv * xrow[1];
+                acc[2] += wv * xrow[TW];
+                acc[3] += wv * xrow[TW + 1];
+              }
+        }
+        float m01 = fmaxf(acc[0], acc[1]), m23 = fmaxf(acc[2], acc[3]);
+        float mx = fmaxf(m01, m23);
+        unsigned char a;
+        if (mx == acc[0]) a = 0; else if (mx == acc[1]) a = 1;
+        else if (mx == acc[2]) a = 2; else a = 3;
+        pout[j] = to_bf16((co < CO) ? fmaxf(mx, 0.f) : 0.f);
+        am[j] = a;
+      }
+      long off = off0 + cg * 8;
+      *reinterpret_cast<bf16x8*>(p + off) = pout;
+      *reinterpret_cast<unsigned long long*>(idx + off) =
+          *reinterpret_cast<unsigned long long*>(am);
+    }
   }
 }
 
@@ -1254,7 +1438,7 @@ __global__ __launch_bounds__(256) void conv_pool_stem_bf16_k(
 // conv-space dy is reconstructed on the fly from (d_pool, pool_y, idx):
 //   dy_conv[n,oh,ow,co] = (idx==quadrant && pool_y>0) ? d_pool : 0.
 // ---------------------------------------------------------------------------
-template <int KT>
+template <int KT, int MC = 64>
 __global__ __launch_bounds__(256) void conv2d_wgrad_pool_bf16_k(
     const bf16* __restrict__ dpool, const bf16* __restrict__ pool_y,
     const unsigned char* __restrict__ pidx, const bf16* __restrict__ x,
@@ -1266,7 +1450,8 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_pool_bf16_k(
   // KW*CIp is not a multiple of 8 (the stem's CIp can be 1). Pad k entries
   // read in-row garbage (load base clamped to the tensor) and their dw rows
   // are simply never written back.
-  constexpr int MC = 64;
+  // KT sized so 4*KT*16 >= K with minimal waste (LeNet stem: K=40 -> KT=1;
+  // round-1 ran KT=2 = 128 k-slots, 69% dead MFMA/staging).
   constexpr int NKF = 4 * KT;
   const int k0 = blockIdx.x * (NKF * 16);
   const int co0 = blockIdx.y * 16;
@@ -1281,6 +1466,7 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_pool_bf16_k(
   __shared__ bf16 lds_x[NKF][MC][16];
   __shared__ bf16 lds_dy[MC][16];
   __shared__ int lds_mbase[MC];
+  __shared__ int lds_mpool[MC];   // pooled-offset | quadrant (COp mult of 8)
 
   const int KROWV = KW * CIp;                 // valid entries per kh row
   const int KROW = ((KROWV + 7) / 8) * 8;     // padded row length
@@ -1303,12 +1489,15 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_pool_bf16_k(
 
   for (int mc = m_begin; mc < m_end; mc += MC) {
     __syncthreads();
-    if (tid < MC) {
-      int m = mc + tid;
+    for (int mi = tid; mi < MC; mi += 256) {
+      int m = mc + mi;
       int mm = m < m_end ? m : 0;
       int n = mm / OHOW, r = mm % OHOW;
       int oh = r / OW, ow = r % OW;
-      lds_mbase[tid] = (n * H + oh) * W * CIp + ow * CIp;  // pad=0, s=1
+      lds_mbase[mi] = (n * H + oh) * W * CIp + ow * CIp;  // pad=0, s=1
+      int q = (oh & 1) * 2 + (ow & 1);
+      lds_mpool[mi] =
+          (int)((((long)(n * OH2 + oh / 2) * OW2 + ow / 2) * COp)) | q;
     }
     __syncthreads();
     for (int idx2 = tid; idx2 < NKF * MC * 2; idx2 += 256) {
@@ -1333,10 +1522,9 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_pool_bf16_k(
       int m = mc + mloc;
       bf16x8 v = bf8_zero();
       if (m < m_end && co0 + half * 8 < COp) {
-        int n = m / OHOW, r = m % OHOW;
-        int oh = r / OW, ow = r % OW;
-        unsigned char q = (unsigned char)((oh & 1) * 2 + (ow & 1));
-        long off = (((long)(n * OH2 + oh / 2) * OW2 + ow / 2) * COp) + co0 + half * 8;
+        int mp = lds_mpool[mloc];
+        unsigned char q = (unsigned char)(mp & 3);
+        long off = (long)(mp & ~3) + co0 + half * 8;
         bf16x8 dp = *reinterpret_cast<const bf16x8*>(&dpool[off]);
         bf16x8 py = *reinterpret_cast<const bf16x8*>(&pool_y[off]);
         unsigned long long am8 =

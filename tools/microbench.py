@@ -84,6 +84,17 @@ def main():
                                                   stride, pad, True))
         print(f"{name:14} fwd={fg:9.1f} dgrad={dg:9.1f} wgrad={wg:9.1f}")
 
+    # LeNet fused stem (conv1+pool+relu) fwd + its fused wgrad
+    N = 8192
+    xs = torch.randn(N, 28, 28, 1, device=dev).to(torch.bfloat16)
+    ws = torch.randn(10, 1, 5, 5, device=dev) * 0.1
+    bs = torch.randn(10, device=dev)
+    p, idx = ext.conv_pool_stem_bf16(xs, ws, bs, 16)
+    dpool = torch.randn_like(p)
+    sf = timeit(lambda: ext.conv_pool_stem_bf16(xs, ws, bs, 16))
+    sw = timeit(lambda: ext.conv2d_wgrad_pool_bf16(dpool, p, idx, xs, 1, 10, 5, 5))
+    print(f"\n{'lenet.stem':14} fwd={sf:9.1f} wgrad_pool={sw:9.1f}")
+
 
 if __name__ == "__main__":
     main()
