@@ -397,33 +397,28 @@ __global__ __launch_bounds__(256) void conv_smallk_bf16_k(
 
   f32x4 acc[MW][NW] = {};
 
-  // ---- barrier-free k loop ------------------------------------------------
+  // ---- barrier-free k loop, A prefetched one group ahead ------------------
+  // Per k-step each lane's A-gather feeds its MFMAs directly; without
+  // prefetch the wave's critical path is ngroups/4 dependent L1/L2 round
+  // trips. Loading group kg+4 before the MFMAs of group kg lets the next
+  // load's latency hide under the current step's matrix work.
   const int ngroups = K >> 3;
-  for (int kg = 0; kg < ngroups; kg += 4) {
+  auto gather_a = [&](int kg, bf16x8 (&a)[MW]) {
     bool gin = lk < (ngroups - kg);
     int gsel = gin ? kg + lk : 0;
     int kdec = lds_dec[gsel];
-    bf16x8 bw[NW];
-#pragma unroll
-    for (int ni = 0; ni < NW; ++ni)
-      bw[ni] = gin ? *reinterpret_cast<const bf16x8*>(
-                         &lds_w[(ni * 16 + lm) * Krow + gsel * 8])
-                   : bf8_zero();
     if (DGRAD) {
       int kh = kdec >> 20, kw = (kdec >> 10) & 0x3ff, c0g = kdec & 0x3ff;
 #pragma unroll
       for (int mi = 0; mi < MW; ++mi) {
-        bf16x8 a = bf8_zero();
+        a[mi] = bf8_zero();
         if (mval[mi] & gin) {
           int tH = ih0[mi] + pad - kh;
           int tW = iw0[mi] + pad - kw;
           if (tH >= 0 && tH < OH && tW >= 0 && tW < OW)
-            a = *reinterpret_cast<const bf16x8*>(
+            a[mi] = *reinterpret_cast<const bf16x8*>(
                 &x[((long)(nn[mi] * OH + tH) * OW + tW) * COp + c0g]);
         }
-#pragma unroll
-        for (int ni = 0; ni < NW; ++ni)
-          acc[mi][ni] = mfma16x16x32bf(a, bw[ni], acc[mi][ni]);
       }
     } else {
       int koff = lds_off[gsel];
@@ -435,13 +430,37 @@ __global__ __launch_bounds__(256) void conv_smallk_bf16_k(
           int ih = ih0[mi] + kh, iw = iw0[mi] + kw;
           ok = ok && ih >= 0 && ih < H && iw >= 0 && iw < W;
         }
-        bf16x8 a = ok ? *reinterpret_cast<const bf16x8*>(&x[xbase[mi] + koff])
-                      : bf8_zero();
-#pragma unroll
-        for (int ni = 0; ni < NW; ++ni)
-          acc[mi][ni] = mfma16x16x32bf(a, bw[ni], acc[mi][ni]);
+        a[mi] = ok ? *reinterpret_cast<const bf16x8*>(&x[xbase[mi] + koff])
+                   : bf8_zero();
       }
     }
+  };
+
+  bf16x8 a_cur[MW];
+  gather_a(0, a_cur);
+  for (int kg = 0; kg < ngroups; kg += 4) {
+    bf16x8 a_nxt[MW];
+    if (kg + 4 < ngroups) {
+      gather_a(kg + 4, a_nxt);
+    } else {
+#pragma unroll
+      for (int mi = 0; mi < MW; ++mi) a_nxt[mi] = bf8_zero();
+    }
+    bool gin = lk < (ngroups - kg);
+    int gsel = gin ? kg + lk : 0;
+    bf16x8 bw[NW];
+#pragma unroll
+    for (int ni = 0; ni < NW; ++ni)
+      bw[ni] = gin ? *reinterpret_cast<const bf16x8*>(
+                         &lds_w[(ni * 16 + lm) * Krow + gsel * 8])
+                   : bf8_zero();
+#pragma unroll
+    for (int mi = 0; mi < MW; ++mi)
+#pragma unroll
+      for (int ni = 0; ni < NW; ++ni)
+        acc[mi][ni] = mfma16x16x32bf(a_cur[mi], bw[ni], acc[mi][ni]);
+#pragma unroll
+    for (int mi = 0; mi < MW; ++mi) a_cur[mi] = a_nxt[mi];
   }
 
   // ---- epilogue -----------------------------------------------------------

@@ -34,27 +34,37 @@ def main() -> None:
             "select kernel_name, sum(duration) from counters_collection "
             "where counter_name='SQ_WAVE_CYCLES' group by kernel_name"):
         dur[name] = duration  # ns, one row per dispatch per counter
+    # every counter present in the db becomes a column, normalized by
+    # SQ_WAVE_CYCLES (the wave-parked/issue-stall split needs WAIT_ANY /
+    # WAIT_INST_ANY relative to wave cycles, not wall duration)
+    counters = sorted({cn for d in agg.values() for cn in d} - {"SQ_WAVE_CYCLES"})
     rows = []
     for name, d in agg.items():
         ns = dur[name]
         if ns <= 0:
             continue
         denom = ns * GHZ * SIMDS
-        rows.append((ns, len(calls[name]),
-                     100.0 * d.get("SQ_VALU_MFMA_BUSY_CYCLES", 0.0) / denom,
-                     100.0 * d.get("SQ_LDS_BANK_CONFLICT", 0.0) / denom,
-                     name))
+        wave = d.get("SQ_WAVE_CYCLES", 0.0) or 1.0
+        cols = []
+        for cn in counters:
+            if cn == "SQ_VALU_MFMA_BUSY_CYCLES":
+                cols.append(100.0 * d.get(cn, 0.0) / denom)   # % of wall
+            else:
+                cols.append(100.0 * d.get(cn, 0.0) / wave)    # % of wave cyc
+        rows.append((ns, len(calls[name]), cols, name))
     rows.sort(reverse=True)
+    short = {"SQ_VALU_MFMA_BUSY_CYCLES": "mfma%w", "SQ_LDS_BANK_CONFLICT": "ldsc",
+             "SQ_WAIT_ANY": "parked", "SQ_WAIT_INST_ANY": "isstal",
+             "SQ_ACTIVE_INST_ANY": "active"}
     with open(out, "w") as f:
         f.write(f"# {title}\n")
-        f.write("# rocprofv3 --pmc SQ_VALU_MFMA_BUSY_CYCLES SQ_WAVE_CYCLES "
-                "SQ_LDS_BANK_CONFLICT --kernel-trace\n")
-        f.write("# mfma%/ldsc% normalized by duration*2.4GHz*1024 SIMDs "
-                "(approximate, run-relative)\n")
-        f.write("  total_ms  calls  mfma%  ldsc%  kernel\n")
-        for ns, n, mfma, lds, name in rows[:30]:
-            f.write(f"{ns / 1e6:10.3f} {n:6d} {mfma:6.1f} {lds:6.2f}  "
-                    f"{name[:150]}\n")
+        f.write("# mfma%w = MFMA busy / wall (2.4GHz x 1024 SIMDs); other "
+                "counters as % of SQ_WAVE_CYCLES (quad-cycles)\n")
+        hdr = " ".join(f"{short.get(cn, cn[:6]):>7}" for cn in counters)
+        f.write(f"  total_ms  calls {hdr}  kernel\n")
+        for ns, n, cols, name in rows[:30]:
+            cs = " ".join(f"{v:7.1f}" for v in cols)
+            f.write(f"{ns / 1e6:10.3f} {n:6d} {cs}  {name[:140]}\n")
     print("wrote", out)
 
 
