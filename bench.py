@@ -130,10 +130,11 @@ def main():
             graphed.slots[slot][0].copy_(x, non_blocking=True)
             graphed.slots[slot][1].copy_(t, non_blocking=True)
             return graphed.step(slot)
-        if isinstance(model, DistributedDataParallel):
-            model.zero_grad()
-        else:
-            opt.zero_grad()
+        # no zero_grad: the fused SGD kernel resets grads to zero in the
+        # same stores that consume them (ops/sgd.py), so backward always
+        # accumulates onto zeros — the eager path carries no FillFunctor
+        # launches (first step: autograd assigns fresh grads / DDP installs
+        # zeroed bucket views)
         _, loss = model(x, t)
         loss.backward()
         if isinstance(model, DistributedDataParallel):

@@ -449,3 +449,162 @@ __global__ __launch_bounds__(256) void reduce_splits_wave_k(
   for (int d = 32; d > 0; d >>= 1) v += __shfl_xor(v, d);
   if (l == 0) out[e] = v;
 }
+
+// ---------------------------------------------------------------------------
+// Whole-image-LDS conv (fwd and dgrad), NCHW fp32 — the LeNet-class fp32
+// engine (fp32 at the bf16 engine's structural quality).
+// The generic f32 gather kernels above pay one SCALAR global gather per lane
+// per MFMA k-slice (NCHW has no contiguous channel vectors), measured
+// 478 us/call on the conv2 dgrad at batch 8192 — pure L1/L2 latency. At
+// LeNet scale whole input images are tiny (conv2 fwd: 5.8 KB/image fp32),
+// so a block stages its 2-3 images AND the whole repacked weight panel in
+// LDS once, then the k loop runs barrier-free: per mfma_f32_16x16x4 k-slice
+// each lane does one ds_read_b32 from the image (~2 cyc) instead of a
+// global gather. fp32 MFMA is exact f32 at the 155 TF vector rate
+// (guide §3), so the fp32 dtype contract is preserved bit-for-bit vs an
+// fmaf chain.
+//   A (lane l): A[m = l&15][k = l>>4] — one float
+//   B (lane l): B[k = l>>4][c = l&15] from lds_w[k][OCHp]
+//   D (lane l): col c = l&15, row m = (l>>4)*4 + reg
+// DGRAD consumes dy images with the transposed weight panel and the
+// (pad-kh, pad-kw) tap predicate; fwd is routed pad==0 only (LeNet).
+// ---------------------------------------------------------------------------
+template <bool DGRAD, bool HAS_BIAS, int MW, int NW, int IMG_MAX>
+__global__ __launch_bounds__(256) void conv_smallimg_f32_k(
+    const float* __restrict__ x, const float* __restrict__ w,
+    const float* __restrict__ bias, float* __restrict__ y,
+    int CI, int H, int W, int CO, int KH, int KW, int OH, int OW,
+    int pad, int M, int K, int N) {
+  const int ICH = DGRAD ? CO : CI;        // channels of the staged tensor
+  const int IH = DGRAD ? OH : H;
+  const int IW = DGRAD ? OW : W;
+  const int OCH = DGRAD ? CI : CO;        // valid output columns
+  const int OCHp = (OCH + 7) / 8 * 8;
+  const int OHo = DGRAD ? H : OH;         // output map dims
+  const int OWo = DGRAD ? W : OW;
+  const int ohow = OHo * OWo;             // out pixels per image
+  const int img_elems = ICH * IH * IW;
+  const int KHKW = KH * KW;
+
+  extern __shared__ __attribute__((aligned(16))) char smemc[];
+  float* lds_img = reinterpret_cast<float*>(smemc);      // [img][ICH][IH][IW]
+  float* lds_w = lds_img + IMG_MAX * img_elems;          // [K][OCHp]
+  int* lds_off = reinterpret_cast<int*>(lds_w + (long)K * OCHp);  // [K]
+  int* lds_dk = lds_off + K;                             // [K] (dgrad)
+
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6;
+  const int l = tid & 63;
+  const int lm = l & 15;
+  const int lk = l >> 4;
+  constexpr int BM = 4 * MW * 16;
+  const int m0 = blockIdx.x * BM;
+  const int img_lo = m0 / ohow;
+  const int img_n =
+      min(IMG_MAX, (min(m0 + BM, M) - 1) / ohow - img_lo + 1);
+
+  {  // stage images (linear copy; float4 when the image size permits)
+    const float* src = x + (long)img_lo * img_elems;
+    long total = (long)img_n * img_elems;
+    if ((img_elems & 3) == 0) {
+      for (long i = tid; i < (total >> 2); i += 256)
+        reinterpret_cast<float4*>(lds_img)[i] =
+            reinterpret_cast<const float4*>(src)[i];
+    } else {
+      for (long i = tid; i < total; i += 256) lds_img[i] = src[i];
+    }
+  }
+  for (int idx = tid; idx < K * OCHp; idx += 256) {  // weight panel
+    int k = idx / OCHp, c = idx % OCHp;
+    float v = 0.f;
+    if (c < OCH) {
+      if (DGRAD) {
+        v = w[((long)(k / KHKW) * CI + c) * KHKW + k % KHKW];
+      } else {
+        v = w[(long)c * K + k];   // w[CO][CI*KH*KW], k already canonical
+      }
+    }
+    lds_w[idx] = v;
+  }
+  for (int k = tid; k < K; k += 256) {  // per-k tables
+    int r = k % KHKW;
+    int kh = r / KW, kw = r % KW;
+    if (DGRAD) {
+      lds_off[k] = ((k / KHKW) * OH + pad - kh) * OW + (pad - kw);
+      lds_dk[k] = ((pad - kh) << 16) | ((pad - kw) & 0xffff);
+    } else {
+      lds_off[k] = ((k / KHKW) * H + kh) * W + kw;
+    }
+  }
+  __syncthreads();
+
+  // per-fragment m decode (this lane's A row is m = ... + lm)
+  int mbase[MW], mih[MW], miw[MW];
+  bool mok[MW];
+#pragma unroll
+  for (int mi = 0; mi < MW; ++mi) {
+    int m = m0 + (wid * MW + mi) * 16 + lm;
+    mok[mi] = m < M;
+    int mm = mok[mi] ? m : m0;
+    int nimg = mm / ohow, r = mm % ohow;
+    int oy = r / OWo, ox = r % OWo;
+    mih[mi] = oy;
+    miw[mi] = ox;
+    // fwd: element = img + xoff[k] + oy*W + ox   (pad 0, stride 1)
+    // dgrad: element = img + aoff[k] + ih*OW + iw
+    mbase[mi] = (nimg - img_lo) * img_elems +
+                (DGRAD ? oy * OW + ox : oy * W + ox);
+  }
+
+  f32x4 acc[MW][NW] = {};
+  for (int k0 = 0; k0 < K; k0 += 4) {
+    int k = k0 + lk;
+    bool kin = k < K;
+    int kc = kin ? k : 0;
+    int aoff = lds_off[kc];
+    float bw[NW];
+#pragma unroll
+    for (int ni = 0; ni < NW; ++ni) {
+      int c = ni * 16 + lm;
+      bw[ni] = (kin && c < OCHp) ? lds_w[kc * OCHp + c] : 0.f;
+    }
+    int dkh = 0, dkw = 0;
+    if (DGRAD) {
+      int dk = lds_dk[kc];
+      dkh = dk >> 16;
+      dkw = (short)(dk & 0xffff);
+    }
+#pragma unroll
+    for (int mi = 0; mi < MW; ++mi) {
+      bool ok = mok[mi] & kin;
+      if (DGRAD) {
+        int tH = mih[mi] + dkh, tW = miw[mi] + dkw;
+        ok = ok && (unsigned)tH < (unsigned)OH && (unsigned)tW < (unsigned)OW;
+      }
+      float a = lds_img[ok ? mbase[mi] + aoff : 0];
+      if (!ok) a = 0.f;
+#pragma unroll
+      for (int ni = 0; ni < NW; ++ni)
+        acc[mi][ni] = mfma16x16x4(a, bw[ni], acc[mi][ni]);
+    }
+  }
+
+  // epilogue: NCHW scattered stores (channel stride ohow)
+#pragma unroll
+  for (int mi = 0; mi < MW; ++mi) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      int m = m0 + (wid * MW + mi) * 16 + lk * 4 + reg;
+      if (m >= M) continue;
+      int nimg = m / ohow, r = m % ohow;
+#pragma unroll
+      for (int ni = 0; ni < NW; ++ni) {
+        int c = ni * 16 + lm;
+        if (c >= OCH) continue;
+        float v = acc[mi][ni][reg];
+        if (HAS_BIAS) v += bias[c];
+        y[((long)nimg * OCH + c) * ohow + r] = v;
+      }
+    }
+  }
+}

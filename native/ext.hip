@@ -61,6 +61,38 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias, long stride,
   int OW = (W + 2 * pad - KW) / stride + 1;
   auto y = at::empty({N, CO, OH, OW}, x.options());
   int M = N * OH * OW, K = CI * KH * KW;
+  // Whole-image-LDS engine (LeNet-class fp32 shapes): images + the whole
+  // weight panel stay LDS-resident, per-k gathers become ds_read_b32
+  // (conv_smallimg_f32_k) — replaces the scalar global gathers below.
+  {
+    int ohow = OH * OW;
+    long img_elems = (long)CI * H * W;
+    constexpr int IMGM = 4;
+    int span = (192 % ohow == 0) ? 192 / ohow : 192 / ohow + 2;
+    long lds = IMGM * img_elems * 4 + (long)K * ((CO + 7) / 8 * 8) * 4 +
+               2L * K * 4;
+    if (stride == 1 && pad == 0 && K <= 512 && CO <= 32 && span <= IMGM &&
+        lds <= 100000) {
+      int snw = CO > 16 ? 2 : 1;
+      dim3 g(cdiv(M, 192), 1);
+      auto go = [&](auto hb, auto nwc) {
+        hipLaunchKernelGGL((conv_smallimg_f32_k<false, decltype(hb)::value, 3,
+                                                decltype(nwc)::value, IMGM>),
+                           g, dim3(256), lds, cur_stream(),
+                           x.data_ptr<float>(), w.data_ptr<float>(),
+                           has_bias ? bias.data_ptr<float>() : nullptr,
+                           y.data_ptr<float>(), CI, H, W, CO, KH, KW, OH, OW,
+                           (int)pad, M, K, N);
+      };
+      using T = std::true_type;
+      using F = std::false_type;
+      using N1 = std::integral_constant<int, 1>;
+      using N2 = std::integral_constant<int, 2>;
+      if (has_bias) { if (snw == 2) go(T{}, N2{}); else go(T{}, N1{}); }
+      else { if (snw == 2) go(F{}, N2{}); else go(F{}, N1{}); }
+      return y;
+    }
+  }
   int nw = (CO > 16) ? 2 : 1;   // multi-co tiles amortize the im2col gather
   dim3 grid(cdiv(M, 128), cdiv(CO, nw * 16));
   bool bounds = pad > 0;
@@ -94,6 +126,38 @@ at::Tensor conv2d_bwd_input(at::Tensor dy, at::Tensor w, long H, long W,
   int CI = w.size(1), KH = w.size(2), KW = w.size(3);
   auto dx = at::empty({N, CI, H, W}, dy.options());
   int Mp = N * H * W, Kp = CO * KH * KW;
+  // Whole-image-LDS dgrad (LeNet-class): dy images + transposed weight
+  // panel LDS-resident (see conv_smallimg_f32_k; 478 us -> the scalar
+  // global gathers were the bound)
+  {
+    int ohow = H * W;
+    long img_elems = (long)CO * OH * OW;
+    constexpr int IMGM = 4;
+    int span = (256 % ohow == 0) ? 256 / ohow : 256 / ohow + 2;
+    long lds = IMGM * img_elems * 4 + (long)Kp * ((CI + 7) / 8 * 8) * 4 +
+               2L * Kp * 4;
+    // MEASURED NEUTRAL-TO-WORSE (516 vs 478 us on LeNet conv2 dgrad,
+    // gpurun_out/prof_lenet_fp32_r2h.txt): LDS-resident images do not move
+    // the bound (same pattern as the bf16 smallK experiments — the
+    // per-MFMA dependent gather chain, not where the gather lands).
+    // Env-gated for A/B only.
+    if (stride == 1 && Kp <= 512 && CI <= 32 && span <= IMGM &&
+        lds <= 100000 && getenv("DMA_F32_SMALLIMG_DGRAD")) {
+      int snw = CI > 16 ? 2 : 1;
+      dim3 g(cdiv(Mp, 256), 1);
+      auto go = [&](auto nwc) {
+        hipLaunchKernelGGL((conv_smallimg_f32_k<true, false, 4,
+                                                decltype(nwc)::value, IMGM>),
+                           g, dim3(256), lds, cur_stream(),
+                           dy.data_ptr<float>(), w.data_ptr<float>(), nullptr,
+                           dx.data_ptr<float>(), CI, (int)H, (int)W, CO, KH,
+                           KW, OH, OW, (int)pad, Mp, Kp, N);
+      };
+      if (snw == 2) go(std::integral_constant<int, 2>{});
+      else go(std::integral_constant<int, 1>{});
+      return dx;
+    }
+  }
   int mw = (Mp >= 1 << 16) ? 4 : 2;  // amortize LDS weight/kdec reads
   dim3 grid(cdiv(Mp, mw * 64), cdiv(CI, 16));
   auto launch = [&](auto s1c, auto mwc) {
