@@ -51,8 +51,13 @@ class Net(nn.Module):
     def forward(self, x, target=None):
         if self._bf16_native and x.is_cuda:
             return self._forward_bf16(x, target)
-        x = ops.conv2d(x, self.conv1.weight, self.conv1.bias)
-        x = ops.maxpool2x2_relu(x)
+        if x.is_cuda:
+            # fused stem (conv1 + pool + relu): the fp32 twin of the bf16
+            # path's fused stem — no conv-y round trip (2x189 MB at b8192)
+            x = ops.conv_pool_stem(x, self.conv1.weight, self.conv1.bias)
+        else:
+            x = ops.conv2d(x, self.conv1.weight, self.conv1.bias)
+            x = ops.maxpool2x2_relu(x)
         x = ops.conv2d(x, self.conv2.weight, self.conv2.bias)
         x = ops.dropout(x, p=self.conv2_drop.p, training=self.training, channelwise=True)
         x = ops.maxpool2x2_relu(x)

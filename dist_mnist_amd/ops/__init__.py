@@ -1,6 +1,7 @@
 from . import native  # noqa: F401
 from .functional import (  # noqa: F401
     conv2d,
+    conv_pool_stem,
     dropout,
     linear,
     log_softmax,

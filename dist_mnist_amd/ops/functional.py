@@ -72,6 +72,32 @@ def conv2d(x, w, b=None, stride=1, padding=0):
     return _Conv2dFn.apply(x, w, b, stride, padding)
 
 
+class _ConvPoolStemF32Fn(torch.autograd.Function):
+    """Fused Conv2d(CI==1, s1 p0) + MaxPool2x2 + ReLU, fp32 NCHW (the LeNet
+    stem, GPU only — the fp32 twin of ops.functional_bf16.conv_pool_stem).
+    Backward produces only (dw, db): the stem is the input layer."""
+
+    @staticmethod
+    def forward(ctx, x, w, b):
+        p, idx = native.ext().conv_pool_stem_f32(_nc(x), _nc(w), _nc(b))
+        ctx.save_for_backward(x, p, idx, w)
+        return p
+
+    @staticmethod
+    def backward(ctx, dpool):
+        x, p, idx, w = ctx.saved_tensors
+        if ctx.needs_input_grad[0]:
+            raise RuntimeError("conv_pool_stem has no dgrad (input layer only)")
+        dw, db = native.ext().conv2d_wgrad_pool_f32(
+            _nc(dpool), p, idx, _nc(x), w.shape[2], w.shape[3])
+        return None, dw, db
+
+
+def conv_pool_stem(x, w, b):
+    """GPU fp32 fused stem; callers fall back to conv2d+maxpool on CPU."""
+    return _ConvPoolStemF32Fn.apply(x, w, b)
+
+
 # ---------------------------------------------------------------------------
 # Fused MaxPool2d(2) + ReLU  (reference computes relu(max_pool2d(conv(x), 2));
 # relu is monotone so relu(max4(x)) == the same graph, one kernel fwd + bwd)

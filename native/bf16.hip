@@ -1530,8 +1530,16 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_pool_bf16_k(
       bf16x8 v = bf8_zero();
       if (m < m_end && k < K) {
         long a = (long)lds_mbase[mloc] + g_off[g];
-        if (a + 8 > x_elems) a = x_elems - 8;  // clamp: pad rows are discarded
-        v = *reinterpret_cast<const bf16x8*>(&x[a]);
+        if (a + 8 > x_elems) {
+          // per-element tail: a whole-vector clamp would SHIFT the
+          // still-valid low k slots for the last image's final rows
+          // (latent in round 1; caught by the f32 twin's tighter test)
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            v[j] = (a + j < x_elems) ? x[a + j] : to_bf16(0.f);
+        } else {
+          v = *reinterpret_cast<const bf16x8*>(&x[a]);
+        }
       }
       *reinterpret_cast<bf16x8*>(&lds_x[kf][mloc][half * 8]) = v;
     }

@@ -608,3 +608,202 @@ __global__ __launch_bounds__(256) void conv_smallimg_f32_k(
     }
   }
 }
+
+// ---------------------------------------------------------------------------
+// Fused fp32 stem: Conv2d(CI==1, s1 p0) + MaxPool2x2 + ReLU in one kernel
+// (the fp32-tier port of conv_pool_stem_bf16_k — VERDICT item 3). NCHW in
+// and out. Thread = (n, pool cell), all CO groups; x tile rows load as
+// float2 vectors (row base is an even pixel -> 8-B aligned). Writes only
+// the pooled output + argmax: the separate conv-y round trip at batch 8192
+// was 2x189 MB of fp32 traffic.
+// ---------------------------------------------------------------------------
+template <int KH, int KW>
+__global__ __launch_bounds__(256) void conv_pool_stem_f32_k(
+    const float* __restrict__ x, const float* __restrict__ w,
+    const float* __restrict__ bias, float* __restrict__ p,
+    unsigned char* __restrict__ idx,
+    int H, int W, int CO, int OH2, int OW2, int N) {
+  extern __shared__ float lds_wf[];   // [CO][KH*KW]
+  const int tid = threadIdx.x;
+  constexpr int Kc = KH * KW;
+  for (int i = tid; i < CO * Kc; i += 256) lds_wf[i] = w[i];
+  __syncthreads();
+
+  constexpr int TH = KH + 1, TW = KW + 1;
+  const long ohow2 = (long)OH2 * OW2;
+  long total = (long)N * ohow2;
+  for (long t = (long)blockIdx.x * blockDim.x + tid; t < total;
+       t += (long)gridDim.x * blockDim.x) {
+    int ow2 = t % OW2;
+    long r2 = t / OW2;
+    int oh2 = r2 % OH2;
+    int n = r2 / OH2;
+
+    float xt[TH * TW];
+    const float* xb = x + (long)n * H * W + (oh2 * 2) * W + ow2 * 2;
+#pragma unroll
+    for (int ty = 0; ty < TH; ++ty)
+#pragma unroll
+      for (int tx = 0; tx < TW; tx += 2) {
+        float2 v = *reinterpret_cast<const float2*>(&xb[ty * W + tx]);
+        xt[ty * TW + tx] = v.x;
+        xt[ty * TW + tx + 1] = v.y;
+      }
+
+    for (int co = 0; co < CO; ++co) {
+      float a0, a1, a2, a3;
+      float bv = bias[co];
+      a0 = a1 = a2 = a3 = bv;
+      const float* wr = &lds_wf[co * Kc];
+#pragma unroll
+      for (int kh = 0; kh < KH; ++kh)
+#pragma unroll
+        for (int kw = 0; kw < KW; ++kw) {
+          float wv = wr[kh * KW + kw];
+          const float* xr = &xt[kh * TW + kw];
+          a0 += wv * xr[0];
+          a1 += wv * xr[1];
+          a2 += wv * xr[TW];
+          a3 += wv * xr[TW + 1];
+        }
+      float m01 = fmaxf(a0, a1), m23 = fmaxf(a2, a3);
+      float mx = fmaxf(m01, m23);
+      unsigned char am;
+      if (mx == a0) am = 0; else if (mx == a1) am = 1;
+      else if (mx == a2) am = 2; else am = 3;
+      long off = ((long)n * CO + co) * ohow2 + oh2 * OW2 + ow2;
+      p[off] = fmaxf(mx, 0.f);
+      idx[off] = am;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Fused fp32 stem wgrad: reconstructs the conv-space dy from
+// (d_pool, pool_y, argmax) on the fly — dy[n,co,oh,ow] = (idx==quadrant &&
+// pool_y>0) ? d_pool : 0 — and reduces dw/db with exact-f32 MFMA
+// (16x16x4: D[k<=48][co<=16] over 4-m slices). Waves split the m chunk
+// (wave-local compute after one cooperative stage), each writing its own
+// partial slab row; reduce_splits folds ms*4 slabs deterministically.
+// k space is per-kh-row padded like the bf16 twin: K = KH*8 (CI==1).
+// ---------------------------------------------------------------------------
+template <int MC>
+__global__ __launch_bounds__(256) void conv2d_wgrad_pool_f32_k(
+    const float* __restrict__ dpool, const float* __restrict__ pool_y,
+    const unsigned char* __restrict__ pidx, const float* __restrict__ x,
+    float* __restrict__ dw_ws, float* __restrict__ db_ws,
+    int H, int W, int CO, int KH, int KW,
+    int OH, int OW, int M, int K, int Kcan, int m_per_split, long x_elems) {
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6;
+  const int l = tid & 63;
+  const int lm = l & 15;
+  const int lk = l >> 4;
+  const int ms = blockIdx.z;
+  const int co0 = blockIdx.y * 16;
+  const int OH2 = OH / 2, OW2 = OW / 2;
+  const int OHOW = OH * OW;
+  const int KROW = 8;                  // CI==1, KW<=8
+  const int NKF = 3;                   // 48 k slots >= K=40
+
+  __shared__ float lds_x[MC][NKF * 16 + 1];   // [m][48k] (+1: bank stagger)
+  __shared__ float lds_dy[MC][16 + 1];        // [m][16co]
+  __shared__ int lds_mb[MC];                  // x base of the patch
+  __shared__ int lds_mpq[MC];                 // pooled offset | quadrant
+
+  f32x4 acc[NKF] = {};
+  float db_part = 0.f;
+
+  const int m_begin = ms * m_per_split;
+  const int m_end = min(M, m_begin + m_per_split);
+  const int mw0 = wid * (MC / 4);      // this wave's m-slice of each chunk
+
+  for (int mc = m_begin; mc < m_end; mc += MC) {
+    __syncthreads();
+    for (int mi = tid; mi < MC; mi += 256) {
+      int m = mc + mi;
+      int mm = m < m_end ? m : 0;
+      int n = mm / OHOW, r = mm % OHOW;
+      int oh = r / OW, ow = r % OW;
+      lds_mb[mi] = (n * H + oh) * W + ow;
+      int q = (oh & 1) * 2 + (ow & 1);
+      // pooled offset is per-channel-plane: final index adds co*OH2*OW2
+      lds_mpq[mi] = ((((n * CO + co0) * OH2 + oh / 2) * OW2 + ow / 2) << 2) | q;
+    }
+    __syncthreads();
+    // x patch rows: [m][kh*8 + kw], 8-wide padded rows (3 valid beyond KW=5
+    // read in-row garbage, never written back)
+    for (int idx2 = tid; idx2 < MC * KH * 2; idx2 += 256) {
+      int half = idx2 & 1;
+      int rest = idx2 >> 1;
+      int kh = rest % KH;
+      int mloc = rest / KH;
+      int m = mc + mloc;
+      float4 v = {0.f, 0.f, 0.f, 0.f};
+      if (m < m_end) {
+        long a = (long)lds_mb[mloc] + kh * W + half * 4;
+        // per-ELEMENT tail bound: a whole-vector clamp would shift the
+        // still-valid kw=4 slot for the last image's final rows (measured:
+        // 7/250 dw elements off by ~2.4e-2 relative)
+        v.x = (a < x_elems) ? x[a] : 0.f;
+        v.y = (a + 1 < x_elems) ? x[a + 1] : 0.f;
+        v.z = (a + 2 < x_elems) ? x[a + 2] : 0.f;
+        v.w = (a + 3 < x_elems) ? x[a + 3] : 0.f;
+      }
+      float* dst = &lds_x[mloc][kh * 8 + half * 4];
+      dst[0] = v.x; dst[1] = v.y; dst[2] = v.z; dst[3] = v.w;
+    }
+    // gated conv-space dy [m][16co] (co plane stride OH2*OW2 in dpool)
+    for (int idx2 = tid; idx2 < MC * 16; idx2 += 256) {
+      int co = idx2 & 15;
+      int mloc = idx2 >> 4;
+      int m = mc + mloc;
+      float v = 0.f;
+      if (m < m_end && co0 + co < CO) {
+        int mpq = lds_mpq[mloc];
+        long off = (long)(mpq >> 2) + (long)co * (OH2 * OW2);
+        unsigned char q = (unsigned char)(mpq & 3);
+        if (pidx[off] == q && pool_y[off] > 0.f) v = dpool[off];
+      }
+      lds_dy[mloc][co] = v;
+    }
+    __syncthreads();
+
+    // wave-local MFMA over its MC/4 m rows: A[k][m-slice], B[m-slice][co]
+    for (int mstep = 0; mstep < MC / 4 / 4; ++mstep) {
+      int mrow = mw0 + mstep * 4 + lk;
+      float b = lds_dy[mrow][lm];
+      db_part += b;
+#pragma unroll
+      for (int kf = 0; kf < NKF; ++kf) {
+        float a = lds_x[mrow][kf * 16 + lm];
+        acc[kf] = mfma16x16x4(a, b, acc[kf]);
+      }
+    }
+  }
+
+  // per-wave partial slab: split index ms*4 + wid
+  {
+    float* slab = dw_ws + (long)(ms * 4 + wid) * (CO * (long)Kcan);
+#pragma unroll
+    for (int kf = 0; kf < NKF; ++kf) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        int kk = kf * 16 + lk * 4 + reg;
+        int co = co0 + lm;
+        if (kk < K && co < CO) {
+          int kh = kk / KROW, r = kk % KROW;
+          if (r < KW)
+            slab[(long)co * Kcan + kh * KW + r] = acc[kf][reg];
+        }
+      }
+    }
+  }
+  {
+    // db: each lane holds partials for co=lm over its (lk, wave) m rows
+    db_part += __shfl_down(db_part, 16);
+    db_part += __shfl_down(db_part, 32);
+    if (lk == 0 && co0 + lm < CO)
+      db_ws[(long)(ms * 4 + wid) * CO + co0 + lm] = db_part;
+  }
+}

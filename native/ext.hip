@@ -229,6 +229,70 @@ std::vector<at::Tensor> conv2d_bwd_weight(at::Tensor dy, at::Tensor x, long KH,
 }
 
 // ---------------------------------------------------------------------------
+std::vector<at::Tensor> conv_pool_stem_f32(at::Tensor x, at::Tensor w,
+                                           at::Tensor bias) {
+  CHECK_F32(x);
+  CHECK_F32(w);
+  CHECK_F32(bias);
+  int N = x.size(0), CI = x.size(1), H = x.size(2), W = x.size(3);
+  int CO = w.size(0), KH = w.size(2), KW = w.size(3);
+  TORCH_CHECK(CI == 1 && KH == 5 && KW == 5,
+              "f32 stem specialized for CI=1 5x5 (LeNet conv1)");
+  int OH = H - KH + 1, OW = W - KW + 1;
+  TORCH_CHECK(OH % 2 == 0 && OW % 2 == 0);
+  int OH2 = OH / 2, OW2 = OW / 2;
+  auto p = at::empty({N, CO, OH2, OW2}, x.options());
+  auto idx = at::empty({N, CO, OH2, OW2}, x.options().dtype(at::kByte));
+  long total = (long)N * OH2 * OW2;
+  int blocks = (int)std::min<long>(4096, cdiv(total, 256));
+  size_t lds = (size_t)CO * KH * KW * sizeof(float);
+  hipLaunchKernelGGL((conv_pool_stem_f32_k<5, 5>), dim3(blocks), dim3(256),
+                     lds, cur_stream(), x.data_ptr<float>(),
+                     w.data_ptr<float>(), bias.data_ptr<float>(),
+                     p.data_ptr<float>(), idx.data_ptr<unsigned char>(),
+                     H, W, CO, OH2, OW2, N);
+  return {p, idx};
+}
+
+std::vector<at::Tensor> conv2d_wgrad_pool_f32(at::Tensor dpool,
+                                              at::Tensor pool_y,
+                                              at::Tensor pidx, at::Tensor x,
+                                              long KH, long KW) {
+  CHECK_F32(dpool);
+  CHECK_F32(pool_y);
+  CHECK_F32(x);
+  int N = x.size(0), H = x.size(2), W = x.size(3);
+  int CO = dpool.size(1), OH2 = dpool.size(2), OW2 = dpool.size(3);
+  int OH = OH2 * 2, OW = OW2 * 2;
+  TORCH_CHECK(x.size(1) == 1 && KW <= 8 && KH * 8 <= 48,
+              "f32 wgrad_pool: CI=1, KW<=8");
+  auto opts = at::TensorOptions().dtype(at::kFloat).device(x.device());
+  long Kcan = KH * KW;
+  auto dw = at::empty({CO, 1, KH, KW}, opts);
+  auto db = at::empty({CO}, opts);
+  int M = N * OH * OW;
+  int K = (int)KH * 8;
+  constexpr int MC = 128;
+  int ct = cdiv(CO, 16);
+  int ms = std::max(1, std::min(2048 / ct, cdiv(M, 256)));
+  int m_per_split = cdiv(cdiv(M, ms), MC) * MC;
+  ms = cdiv(M, m_per_split);
+  // each WAVE writes its own slab row -> ms*4 splits for the fold
+  auto dw_ws = at::empty({ms * 4, CO * Kcan}, opts);
+  auto db_ws = at::empty({ms * 4, CO}, opts);
+  hipLaunchKernelGGL((conv2d_wgrad_pool_f32_k<MC>), dim3(1, ct, ms),
+                     dim3(256), 0, cur_stream(), dpool.data_ptr<float>(),
+                     pool_y.data_ptr<float>(),
+                     pidx.data_ptr<unsigned char>(), x.data_ptr<float>(),
+                     dw_ws.data_ptr<float>(), db_ws.data_ptr<float>(),
+                     H, W, CO, (int)KH, (int)KW, OH, OW, M, K, (int)Kcan,
+                     m_per_split, (long)x.numel());
+  launch_reduce_splits(dw_ws, dw, CO * Kcan, ms * 4);
+  launch_reduce_splits(db_ws, db, CO, ms * 4);
+  return {dw, db};
+}
+
+// ---------------------------------------------------------------------------
 at::Tensor linear_fwd(at::Tensor x, at::Tensor w, at::Tensor bias, bool relu) {
   CHECK_F32(x);
   CHECK_F32(w);
@@ -481,6 +545,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dropout_fwd", &dropout_fwd);
   m.def("dropout_bwd", &dropout_bwd);
   m.def("fused_sgd_step", &fused_sgd_step);
+  m.def("conv_pool_stem_f32", &conv_pool_stem_f32);
+  m.def("conv2d_wgrad_pool_f32", &conv2d_wgrad_pool_f32);
   // bf16 NHWC engine
   m.def("mfma_probe", &mfma_probe);
   m.def("tr16_probe", &tr16_probe);
