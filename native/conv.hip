@@ -557,6 +557,7 @@ __global__ __launch_bounds__(256) void conv_smallimg_f32_k(
   }
 
   f32x4 acc[MW][NW] = {};
+#pragma unroll 2
   for (int k0 = 0; k0 < K; k0 += 4) {
     int k = k0 + lk;
     bool kin = k < K;

@@ -141,9 +141,26 @@ at::Tensor conv2d_bwd_input(at::Tensor dy, at::Tensor w, long H, long W,
     // the bound (same pattern as the bf16 smallK experiments — the
     // per-MFMA dependent gather chain, not where the gather lands).
     // Env-gated for A/B only.
+    // Default: MW=8 whole-image-LDS dgrad for CI<=16 — measured 431 vs the
+    // 470 us gather and 510 us MW=4 (gpurun_out/ab_dgrad.txt; the wider
+    // per-wave m-tile amortizes the per-k table reads + predicates).
+    // DMA_F32_SMALLIMG_DGRAD=0 forces the gather engine for A/B.
+    const char* smv = getenv("DMA_F32_SMALLIMG_DGRAD");
+    int smw = smv ? atoi(smv) : 8;
     if (stride == 1 && Kp <= 512 && CI <= 32 && span <= IMGM &&
-        lds <= 100000 && getenv("DMA_F32_SMALLIMG_DGRAD")) {
+        lds <= 100000 && smw != 0) {
       int snw = CI > 16 ? 2 : 1;
+      if (smw == 8 && snw == 1) {
+        long lds8 = 5 * img_elems * 4 + (long)Kp * ((CI + 7) / 8 * 8) * 4 +
+                    2L * Kp * 4;
+        dim3 g(cdiv(Mp, 512), 1);
+        hipLaunchKernelGGL((conv_smallimg_f32_k<true, false, 8, 1, 5>),
+                           g, dim3(256), lds8, cur_stream(),
+                           dy.data_ptr<float>(), w.data_ptr<float>(), nullptr,
+                           dx.data_ptr<float>(), CI, (int)H, (int)W, CO, KH,
+                           KW, OH, OW, (int)pad, Mp, Kp, N);
+        return dx;
+      }
       dim3 g(cdiv(Mp, 256), 1);
       auto go = [&](auto nwc) {
         hipLaunchKernelGGL((conv_smallimg_f32_k<true, false, 4,
@@ -301,13 +318,20 @@ at::Tensor linear_fwd(at::Tensor x, at::Tensor w, at::Tensor bias, bool relu) {
   int M = x.size(0), CI = x.size(1), CO = w.size(0);
   TORCH_CHECK(w.size(1) == CI, "linear shape mismatch");
   auto y = at::empty({M, CO}, x.options());
-  dim3 grid(cdiv(M, 64), cdiv(CO, 64));
+  // BN=16 when BN=64 would underfill the chip (skinny CO, e.g. fc1)
+  int bn = (cdiv(M, 64) * cdiv(CO, 64) < 512 && CO <= 64) ? 16 : 64;
+  dim3 grid(cdiv(M, 64), cdiv(CO, bn));
   auto launch = [&](auto hb, auto rl) {
-    hipLaunchKernelGGL((linear_fwd_f32_k<decltype(hb)::value, decltype(rl)::value>),
-                       grid, dim3(256), 0, cur_stream(), x.data_ptr<float>(),
-                       w.data_ptr<float>(),
-                       has_bias ? bias.data_ptr<float>() : nullptr,
-                       y.data_ptr<float>(), M, CI, CO);
+    auto go = [&](auto bnc) {
+      hipLaunchKernelGGL((linear_fwd_f32_k<decltype(hb)::value, decltype(rl)::value,
+                                           decltype(bnc)::value>),
+                         grid, dim3(256), 0, cur_stream(), x.data_ptr<float>(),
+                         w.data_ptr<float>(),
+                         has_bias ? bias.data_ptr<float>() : nullptr,
+                         y.data_ptr<float>(), M, CI, CO);
+    };
+    if (bn == 16) go(std::integral_constant<int, 16>{});
+    else go(std::integral_constant<int, 64>{});
   };
   using T = std::true_type;
   using F = std::false_type;

@@ -13,12 +13,15 @@
 // ---------------------------------------------------------------------------
 // Forward: D[m][co]; block = 4 waves x 16 m-rows = BM 64, BN 64 co columns.
 // ---------------------------------------------------------------------------
-template <bool HAS_BIAS, bool RELU>
+template <bool HAS_BIAS, bool RELU, int BN = 64>
 __global__ __launch_bounds__(256) void linear_fwd_f32_k(
     const float* __restrict__ x, const float* __restrict__ w,
     const float* __restrict__ bias, float* __restrict__ y,
     int M, int CI, int CO) {
-  constexpr int BM = 64, BN = 64, BK = 32;
+  // BN=16 variant: skinny-CO shapes (LeNet fc1 CO=50, M=8192) give only
+  // 128 blocks at BN=64 — half the CUs idle; 16-wide co tiles restore the
+  // grid (x re-reads stay L2-hot)
+  constexpr int BM = 64, BK = 32;
   const int m0 = blockIdx.x * BM;
   const int co0 = blockIdx.y * BN;
   const int tid = threadIdx.x;
