@@ -620,13 +620,16 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_bf16_k(
           for (int j = 0; j < 8; ++j) db_part[ni] += to_f32(b[ni][j]);
         }
       }
+      // all KT A tr-reads issue before the MFMA block (wait collapse)
+      bf16x8 av[KT];
 #pragma unroll
-      for (int kf = 0; kf < KT; ++kf) {
-        bf16x8 a = tr_read_8m(&lds_x[wid * KT + kf][mm][0], l);
+      for (int kf = 0; kf < KT; ++kf)
+        av[kf] = tr_read_8m(&lds_x[wid * KT + kf][mm][0], l);
+#pragma unroll
+      for (int kf = 0; kf < KT; ++kf)
 #pragma unroll
         for (int ni = 0; ni < NW; ++ni)
-          acc[kf][ni] = mfma16x16x32bf(a, b[ni], acc[kf][ni]);
-      }
+          acc[kf][ni] = mfma16x16x32bf(av[kf], b[ni], acc[kf][ni]);
     }
   }
 
@@ -802,13 +805,15 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_rs_bf16_k(
     for (int tap = 0; tap < 9; ++tap) {
       int buf = tap & 1;
       if (tap + 1 < 9) build(tap + 1, buf ^ 1);  // overlap with compute
+      // paired tr-reads BEFORE their MFMAs (and alternating accumulators):
+      // the per-f read-then-consume order exposed a full lgkmcnt drain per
+      // MFMA and chained same-acc MFMAs (dep latency 40 > issue 17)
 #pragma unroll
-      for (int f = 0; f < 2; ++f) {
-#pragma unroll
-        for (int s = 0; s < MSTEPS; ++s) {
-          bf16x8 a = tr_read_8m(plane(buf * 2 + f) + s * 32 * 16, l);
-          acc[f][tap] = mfma16x16x32bf(a, bfr[s], acc[f][tap]);
-        }
+      for (int s = 0; s < MSTEPS; ++s) {
+        bf16x8 a0 = tr_read_8m(plane(buf * 2 + 0) + s * 32 * 16, l);
+        bf16x8 a1 = tr_read_8m(plane(buf * 2 + 1) + s * 32 * 16, l);
+        acc[0][tap] = mfma16x16x32bf(a0, bfr[s], acc[0][tap]);
+        acc[1][tap] = mfma16x16x32bf(a1, bfr[s], acc[1][tap]);
       }
       __syncthreads();
     }
@@ -1715,24 +1720,59 @@ __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
 #pragma unroll
       for (int ks = 0; ks < CC; ks += 32) {
         bool gin = lk * 8 < cc - ks;
-        bf16x8 bw[NW];
+        if constexpr (NIMG != 2) {
+          // ALL MW A-reads issue BEFORE the MFMA block: the round-1
+          // read-then-consume-per-fragment order made hipcc drain
+          // lgkmcnt(0) before every 4-MFMA group (~50 exposed LDS cycles
+          // per ~70 cycles of matrix work — the measured 19% MfmaUtil;
+          // asm audit in the commit). Loads are unconditional at a
+          // clamped offset + post-select. Measured: l1 fwd 230->209,
+          // l2 208, dgrad l1 175->160 us. The NIMG==2 instantiation
+          // REGRESSED with this order (194->234) and keeps the old body.
+          int bsel = gin ? ks + lk * 8 : 0;
+          bf16x8 bw[NW];
 #pragma unroll
-        for (int ni = 0; ni < NW; ++ni)
-          bw[ni] = gin ? *reinterpret_cast<const bf16x8*>(
-                             &wtap[(long)(ni * 16 + lm) * CIp + ks + lk * 8])
-                       : bf8_zero();
+          for (int ni = 0; ni < NW; ++ni) {
+            bf16x8 t = *reinterpret_cast<const bf16x8*>(
+                &wtap[(long)(ni * 16 + lm) * CIp + bsel]);
+            bw[ni] = gin ? t : bf8_zero();
+          }
+          bf16x8 av[MW];
 #pragma unroll
-        for (int mi = 0; mi < MW; ++mi) {
-          int px = (mi % (MW / NIMG)) * 8 + (lm & 7);
-          int img = mi / (MW / NIMG);
-          const bf16* xpix =
-              &lds_x[(img * RH * RW + (py * S + kh) * RW + (px * S + kw)) *
-                     ccp];
-          bf16x8 a = gin ? *reinterpret_cast<const bf16x8*>(&xpix[ks + lk * 8])
-                         : bf8_zero();
+          for (int mi = 0; mi < MW; ++mi) {
+            int px = (mi % (MW / NIMG)) * 8 + (lm & 7);
+            int img = mi / (MW / NIMG);
+            const bf16* xpix =
+                &lds_x[(img * RH * RW + (py * S + kh) * RW + (px * S + kw)) *
+                       ccp];
+            bf16x8 t = *reinterpret_cast<const bf16x8*>(&xpix[bsel]);
+            av[mi] = gin ? t : bf8_zero();
+          }
+#pragma unroll
+          for (int mi = 0; mi < MW; ++mi)
+#pragma unroll
+            for (int ni = 0; ni < NW; ++ni)
+              acc[mi][ni] = mfma16x16x32bf(av[mi], bw[ni], acc[mi][ni]);
+        } else {
+          bf16x8 bw[NW];
 #pragma unroll
           for (int ni = 0; ni < NW; ++ni)
-            acc[mi][ni] = mfma16x16x32bf(a, bw[ni], acc[mi][ni]);
+            bw[ni] = gin ? *reinterpret_cast<const bf16x8*>(
+                               &wtap[(long)(ni * 16 + lm) * CIp + ks + lk * 8])
+                         : bf8_zero();
+#pragma unroll
+          for (int mi = 0; mi < MW; ++mi) {
+            int px = (mi % (MW / NIMG)) * 8 + (lm & 7);
+            int img = mi / (MW / NIMG);
+            const bf16* xpix =
+                &lds_x[(img * RH * RW + (py * S + kh) * RW + (px * S + kw)) *
+                       ccp];
+            bf16x8 a = gin ? *reinterpret_cast<const bf16x8*>(&xpix[ks + lk * 8])
+                           : bf8_zero();
+#pragma unroll
+            for (int ni = 0; ni < NW; ++ni)
+              acc[mi][ni] = mfma16x16x32bf(a, bw[ni], acc[mi][ni]);
+          }
         }
       }
     }
@@ -1887,13 +1927,14 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_stem_bf16_k(
     for (int kh = 0; kh < 7; ++kh) {
       int buf = kh & 1;
       if (kh + 1 < 7) build(kh + 1, buf ^ 1);
+      // paired tr-reads before their MFMAs + alternating accumulators
+      // (wait-collapse; same fix as conv2d_wgrad_rs_bf16_k)
 #pragma unroll
-      for (int f = 0; f < 2; ++f) {
-#pragma unroll
-        for (int s = 0; s < MSTEPS; ++s) {
-          bf16x8 a = tr_read_8m(plane(f * 2 + buf) + s * 32 * 16, l);
-          acc[f][kh] = mfma16x16x32bf(a, bfr[s], acc[f][kh]);
-        }
+      for (int s = 0; s < MSTEPS; ++s) {
+        bf16x8 a0 = tr_read_8m(plane(0 * 2 + buf) + s * 32 * 16, l);
+        bf16x8 a1 = tr_read_8m(plane(1 * 2 + buf) + s * 32 * 16, l);
+        acc[0][kh] = mfma16x16x32bf(a0, bfr[s], acc[0][kh]);
+        acc[1][kh] = mfma16x16x32bf(a1, bfr[s], acc[1][kh]);
       }
       __syncthreads();
     }
