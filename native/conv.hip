@@ -734,11 +734,15 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_pool_f32_k(
     __syncthreads();
     // x patch rows: [m][kh*8 + kw], 8-wide padded rows (3 valid beyond KW=5
     // read in-row garbage, never written back)
+    // mloc FASTEST across threads: consecutive lanes then read consecutive
+    // conv pixels (stride-1 x / stride-0-or-1 pooled offsets) — with co or
+    // kh fastest every gating load is a separate NCHW-strided cacheline
+    // (measured 394 us/step, worse than the unfused pair it replaced)
     for (int idx2 = tid; idx2 < MC * KH * 2; idx2 += 256) {
-      int half = idx2 & 1;
-      int rest = idx2 >> 1;
-      int kh = rest % KH;
-      int mloc = rest / KH;
+      int mloc = idx2 % MC;
+      int rest = idx2 / MC;
+      int kh = rest >> 1;
+      int half = rest & 1;
       int m = mc + mloc;
       float4 v = {0.f, 0.f, 0.f, 0.f};
       if (m < m_end) {
@@ -756,8 +760,8 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_pool_f32_k(
     }
     // gated conv-space dy [m][16co] (co plane stride OH2*OW2 in dpool)
     for (int idx2 = tid; idx2 < MC * 16; idx2 += 256) {
-      int co = idx2 & 15;
-      int mloc = idx2 >> 4;
+      int mloc = idx2 % MC;
+      int co = idx2 / MC;
       int m = mc + mloc;
       float v = 0.f;
       if (m < m_end && co0 + co < CO) {

@@ -291,7 +291,9 @@ std::vector<at::Tensor> conv2d_wgrad_pool_f32(at::Tensor dpool,
   int K = (int)KH * 8;
   constexpr int MC = 128;
   int ct = cdiv(CO, 16);
-  int ms = std::max(1, std::min(2048 / ct, cdiv(M, 256)));
+  // 1024 blocks fill the chip at the kernel's 4-blocks/CU LDS occupancy;
+  // each wave writes its own slab so the fold sees ms*4 splits
+  int ms = std::max(1, std::min(1024 / ct, cdiv(M, 256)));
   int m_per_split = cdiv(cdiv(M, ms), MC) * MC;
   ms = cdiv(M, m_per_split);
   // each WAVE writes its own slab row -> ms*4 splits for the fold
