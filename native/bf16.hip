@@ -403,6 +403,20 @@ __global__ __launch_bounds__(256) void conv_smallk_bf16_k(
   // trips. Loading group kg+4 before the MFMAs of group kg lets the next
   // load's latency hide under the current step's matrix work.
   const int ngroups = K >> 3;
+  // Gathers are hardware-bounds-checked buffer loads (guide T8): an
+  // exec-predicated gather is uncountable for hipcc's s_waitcnt
+  // bookkeeping, so every MFMA group drained vmcnt(0) — 34 full drains per
+  // loop body in the .s — serializing the one-group-ahead prefetch this
+  // loop is built around (guide §5 ".s-level traps" (c)). A raw buffer
+  // load needs NO predication: a dead lane's voffset is pushed past
+  // num_records and the hardware returns 0.
+  // x extent in bytes (host gates the smallK route to < 2 GB tensors):
+  // fwd M = N*OH*OW -> images N = M/(OH*OW); dgrad x==dy with M = N*H*W
+  const long n_imgs = DGRAD ? (long)M / (H * W) : (long)M / (OH * OW);
+  const long xbytes =
+      (DGRAD ? n_imgs * OH * OW * COp : n_imgs * H * W * CIp) * 2;
+  auto xrsrc = __builtin_amdgcn_make_buffer_rsrc((void*)x, (short)0,
+                                                 (int)xbytes, 0x00020000);
   auto gather_a = [&](int kg, bf16x8 (&a)[MW]) {
     bool gin = lk < (ngroups - kg);
     int gsel = gin ? kg + lk : 0;
@@ -411,14 +425,13 @@ __global__ __launch_bounds__(256) void conv_smallk_bf16_k(
       int kh = kdec >> 20, kw = (kdec >> 10) & 0x3ff, c0g = kdec & 0x3ff;
 #pragma unroll
       for (int mi = 0; mi < MW; ++mi) {
-        a[mi] = bf8_zero();
-        if (mval[mi] & gin) {
-          int tH = ih0[mi] + pad - kh;
-          int tW = iw0[mi] + pad - kw;
-          if (tH >= 0 && tH < OH && tW >= 0 && tW < OW)
-            a[mi] = *reinterpret_cast<const bf16x8*>(
-                &x[((long)(nn[mi] * OH + tH) * OW + tW) * COp + c0g]);
-        }
+        int tH = ih0[mi] + pad - kh;
+        int tW = iw0[mi] + pad - kw;
+        bool ok = mval[mi] & gin && tH >= 0 && tH < OH && tW >= 0 && tW < OW;
+        long off = ((long)(nn[mi] * OH + tH) * OW + tW) * COp + c0g;
+        int boff = ok ? (int)(off * 2) : 0x7ffffff0;
+        f32x4 t = __builtin_amdgcn_raw_buffer_load_b128(xrsrc, boff, 0, 0);
+        a[mi] = __builtin_bit_cast(bf16x8, t);
       }
     } else {
       int koff = lds_off[gsel];
@@ -430,37 +443,50 @@ __global__ __launch_bounds__(256) void conv_smallk_bf16_k(
           int ih = ih0[mi] + kh, iw = iw0[mi] + kw;
           ok = ok && ih >= 0 && ih < H && iw >= 0 && iw < W;
         }
-        a[mi] = ok ? *reinterpret_cast<const bf16x8*>(&x[xbase[mi] + koff])
-                   : bf8_zero();
+        int boff = ok ? (int)(((long)xbase[mi] + koff) * 2) : 0x7ffffff0;
+        f32x4 t = __builtin_amdgcn_raw_buffer_load_b128(xrsrc, boff, 0, 0);
+        a[mi] = __builtin_bit_cast(bf16x8, t);
       }
     }
   };
 
-  bf16x8 a_cur[MW];
+  // B fragments prefetch one group ahead too (unconditional reads at the
+  // clamped gsel=0 row, zero-selected after — keeps the LDS reads off the
+  // MFMA block's critical wait)
+  auto load_bw = [&](int kg, bf16x8 (&bw)[NW]) {
+    bool gin = lk < (ngroups - kg);
+    int gsel = gin ? kg + lk : 0;
+#pragma unroll
+    for (int ni = 0; ni < NW; ++ni) {
+      bf16x8 t = *reinterpret_cast<const bf16x8*>(
+          &lds_w[(ni * 16 + lm) * Krow + gsel * 8]);
+      bw[ni] = gin ? t : bf8_zero();
+    }
+  };
+
+  bf16x8 a_cur[MW], bw_cur[NW];
   gather_a(0, a_cur);
+  load_bw(0, bw_cur);
   for (int kg = 0; kg < ngroups; kg += 4) {
-    bf16x8 a_nxt[MW];
+    bf16x8 a_nxt[MW], bw_nxt[NW];
     if (kg + 4 < ngroups) {
       gather_a(kg + 4, a_nxt);
+      load_bw(kg + 4, bw_nxt);
     } else {
 #pragma unroll
       for (int mi = 0; mi < MW; ++mi) a_nxt[mi] = bf8_zero();
-    }
-    bool gin = lk < (ngroups - kg);
-    int gsel = gin ? kg + lk : 0;
-    bf16x8 bw[NW];
 #pragma unroll
-    for (int ni = 0; ni < NW; ++ni)
-      bw[ni] = gin ? *reinterpret_cast<const bf16x8*>(
-                         &lds_w[(ni * 16 + lm) * Krow + gsel * 8])
-                   : bf8_zero();
+      for (int ni = 0; ni < NW; ++ni) bw_nxt[ni] = bf8_zero();
+    }
 #pragma unroll
     for (int mi = 0; mi < MW; ++mi)
 #pragma unroll
       for (int ni = 0; ni < NW; ++ni)
-        acc[mi][ni] = mfma16x16x32bf(a_cur[mi], bw[ni], acc[mi][ni]);
+        acc[mi][ni] = mfma16x16x32bf(a_cur[mi], bw_cur[ni], acc[mi][ni]);
 #pragma unroll
     for (int mi = 0; mi < MW; ++mi) a_cur[mi] = a_nxt[mi];
+#pragma unroll
+    for (int ni = 0; ni < NW; ++ni) bw_cur[ni] = bw_nxt[ni];
   }
 
   // ---- epilogue -----------------------------------------------------------
