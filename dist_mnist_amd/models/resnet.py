@@ -35,29 +35,37 @@ class BasicBlock(nn.Module):
         self.stride = stride
 
     def forward(self, x):
+        # BN-apply fusion (docs/ROADMAP.md round 2): bn1's apply+ReLU runs
+        # inside conv2's tiled staging, bn2's (and the downsample bn's)
+        # inside the residual add — no normalized activation is ever
+        # written/re-read. bn_virtual computes stats + the per-channel
+        # affine only; backward recomputes the relu gate from (x, affine).
         t = self.training
         out = fb.conv2d_bf16(x, self.conv1.weight, stride=self.stride, pad=1,
                              cop=_pad8(self.conv1.out_channels))
-        out = fr.batchnorm_bf16(out, self.bn1.weight, self.bn1.bias,
-                                self.bn1.running_mean, self.bn1.running_var,
-                                training=t, momentum=self.bn1.momentum,
-                                eps=self.bn1.eps, relu=True)
+        out, aff1 = fr.bn_virtual_bf16(out, self.bn1.weight, self.bn1.bias,
+                                       self.bn1.running_mean,
+                                       self.bn1.running_var, training=t,
+                                       momentum=self.bn1.momentum,
+                                       eps=self.bn1.eps, relu=True)
         out = fb.conv2d_bf16(out, self.conv2.weight, pad=1,
-                             cop=_pad8(self.conv2.out_channels))
-        out = fr.batchnorm_bf16(out, self.bn2.weight, self.bn2.bias,
-                                self.bn2.running_mean, self.bn2.running_var,
-                                training=t, momentum=self.bn2.momentum,
-                                eps=self.bn2.eps)
+                             cop=_pad8(self.conv2.out_channels), aff=aff1)
+        out, aff2 = fr.bn_virtual_bf16(out, self.bn2.weight, self.bn2.bias,
+                                       self.bn2.running_mean,
+                                       self.bn2.running_var, training=t,
+                                       momentum=self.bn2.momentum,
+                                       eps=self.bn2.eps)
         if self.downsample is not None:
             conv, bn = self.downsample[0], self.downsample[1]
             idn = fb.conv2d_bf16(x, conv.weight, stride=conv.stride[0],
                                  cop=_pad8(conv.out_channels))
-            idn = fr.batchnorm_bf16(idn, bn.weight, bn.bias, bn.running_mean,
-                                    bn.running_var, training=t,
-                                    momentum=bn.momentum, eps=bn.eps)
+            idn, affd = fr.bn_virtual_bf16(idn, bn.weight, bn.bias,
+                                           bn.running_mean, bn.running_var,
+                                           training=t, momentum=bn.momentum,
+                                           eps=bn.eps)
         else:
-            idn = x
-        return fr.add_relu_bf16(out, idn)
+            idn, affd = x, None
+        return fr.add_aff_relu_bf16(out, idn, aff2, affd)
 
 
 class ResNet18(nn.Module):

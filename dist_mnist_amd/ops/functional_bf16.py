@@ -69,26 +69,32 @@ def _use_tiled(x_cip, w, stride, dgrad=False):
 
 class _Conv2dBF16Fn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, w, b, stride, pad, cop, relu):
+    def forward(ctx, x, w, b, stride, pad, cop, relu, aff):
         xc, wc = x.contiguous(), w.contiguous()
         bc = b.contiguous() if b is not None else torch.Tensor()
-        if _use_tiled(x.shape[3], w, stride, dgrad=False):
+        if aff is not None:
+            # x is the RAW previous conv output; the BN+ReLU apply runs
+            # inside the tiled staging (fused-BN route: 3x3 s1 only)
+            y = native.ext().conv_tiled_s1_aff_bf16(xc, wc, aff, cop, pad)
+        elif _use_tiled(x.shape[3], w, stride, dgrad=False):
             y = native.ext().conv_tiled_s1_bf16(xc, wc, bc, cop, pad, relu, False)
         else:
             y = native.ext().conv2d_fwd_bf16(xc, wc, bc, cop, stride, pad, relu)
-        ctx.save_for_backward(x, w)
+        ctx.save_for_backward(x, w, aff if aff is not None else torch.Tensor())
         ctx.meta = (stride, pad, b is not None)
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        x, w = ctx.saved_tensors
+        x, w, aff = ctx.saved_tensors
         stride, pad, has_bias = ctx.meta
         dy = dy.contiguous()
         m = native.ext()
         dx = dw = db = None
         if ctx.needs_input_grad[0]:
-            # dgrad reads dy (channels = COp of the conv): route on that
+            # dgrad reads dy (channels = COp of the conv): route on that.
+            # With fused BN, dx here is d(y_bn) — autograd routes it into
+            # _BnVirtualFn.backward, which turns it into d(x_raw).
             if _use_tiled(dy.shape[3], w, stride, dgrad=True):
                 # s1 dgrad of a KxK conv is a KxK s1 conv with pad' = K-1-pad
                 dx = m.conv_tiled_s1_bf16(dy, w.contiguous(), torch.Tensor(),
@@ -98,17 +104,19 @@ class _Conv2dBF16Fn(torch.autograd.Function):
                 dx = m.conv2d_dgrad_bf16(dy, w, x.shape[1], x.shape[2],
                                          x.shape[3], stride, pad)
         if ctx.needs_input_grad[1] or (has_bias and ctx.needs_input_grad[2]):
+            # with fused BN the wgrad's region staging re-applies the affine,
+            # reproducing the never-materialized y_bn
             dw, db = m.conv2d_wgrad_bf16(dy, x.contiguous(), w.shape[1],
                                          w.shape[0], w.shape[2], w.shape[3],
-                                         stride, pad, has_bias)
+                                         stride, pad, has_bias, aff)
         if not has_bias:
             db = None
-        return dx, dw, db, None, None, None, None
+        return dx, dw, db, None, None, None, None, None
 
 
-def conv2d_bf16(x, w, b=None, stride=1, pad=0, cop=None, relu=False):
+def conv2d_bf16(x, w, b=None, stride=1, pad=0, cop=None, relu=False, aff=None):
     cop = cop or pad8(w.shape[0])
-    return _Conv2dBF16Fn.apply(x, w, b, stride, pad, cop, relu)
+    return _Conv2dBF16Fn.apply(x, w, b, stride, pad, cop, relu, aff)
 
 
 class _MaxPoolReluBF16Fn(torch.autograd.Function):

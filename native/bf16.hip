@@ -720,11 +720,15 @@ constexpr int WGRAD_BF16_KTILE = 4 * 2 * 16;  // KT=2 -> 128
 // H==2*OH, W==2*OW so the input global row for output row g is the LINEAR
 // 2g+kh-1 — the same contiguous-row staging then works with a 2x-wide
 // region and the tap predicate S*oy+kh in [0, S*OH).
-template <int MC, int S = 1>
+template <int MC, int S = 1, bool AFF = false>
 __global__ __launch_bounds__(256) void conv2d_wgrad_rs_bf16_k(
     const bf16* __restrict__ dy, const bf16* __restrict__ x,
     float* __restrict__ dw_ws, float* __restrict__ db_ws, int CIp, int CI,
-    int COp, int CO, int OH, int OW, int M, int Kcan, int m_per_split) {
+    int COp, int CO, int OH, int OW, int M, int Kcan, int m_per_split,
+    const float* __restrict__ aff = nullptr) {
+  // AFF: x is the raw conv output whose BN(+ReLU) apply is fused into the
+  // region staging (same transform as the fused forward — the conv's true
+  // input y_bn is never materialized, so wgrad reproduces it on load)
   const int cic0 = blockIdx.x * 32;       // ci tile
   const int co0 = blockIdx.y * 64;        // co tile (4 waves x 16)
   const int ms = blockIdx.z;
@@ -774,9 +778,17 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_rs_bf16_k(
       int gi = S * g0 - 1 + r;             // input global row
       int iw = c - 1;
       bf16x8 v = bf8_zero();
-      if (gi >= 0 && gi < S * (M / OW) && iw >= 0 && iw < Wi)
+      if (gi >= 0 && gi < S * (M / OW) && iw >= 0 && iw < Wi) {
         v = *reinterpret_cast<const bf16x8*>(
             &x[((long)gi * Wi + iw) * CIp + cic0 + g8 * 8]);
+        if (AFF) {
+          const float* sc = &aff[cic0 + g8 * 8];
+          const float* sh = &aff[CIp + cic0 + g8 * 8];
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            v[j] = to_bf16(fmaxf(to_f32(v[j]) * sc[j] + sh[j], 0.f));
+        }
+      }
       *reinterpret_cast<bf16x8*>(&lds_reg[(p << 5) + g8 * 8]) = v;
     }
     // dy planes [ni][256][16co]
@@ -1676,12 +1688,17 @@ __global__ void repack_w_k(const float* __restrict__ w, bf16* __restrict__ w_rp,
 }
 
 template <int KH, bool HAS_BIAS, bool RELU, int NW, int MW, int NIMG = 1,
-          int S = 1>
+          int S = 1, bool AFF = false>
 __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
     const bf16* __restrict__ x, const bf16* __restrict__ w_rp,
     const float* __restrict__ bias, bf16* __restrict__ y,
     int CIp, int H, int W, int COp, int COcan,
-    int OH, int OW, int pad, int N, int tiles_h, int tiles_w) {
+    int OH, int OW, int pad, int N, int tiles_h, int tiles_w,
+    const float* __restrict__ aff = nullptr) {
+  // AFF: x is a RAW conv output whose BatchNorm(+ReLU) apply is fused into
+  // this staging pass — v = bf16(max(aff[c]*v + aff[CIp+c], 0)); padded
+  // channels have aff == 0 so they stay zero. Saves the separate bn_apply
+  // write + re-read of the whole activation (docs/ROADMAP.md BN fusion).
   // output tile: 8 rows x (8*MW/NIMG) cols x NIMG images; a wave owns MW
   // 16-pixel fragments (split across columns and images), so every
   // B-fragment read feeds MW MFMAs. NIMG>1 keeps the reuse on small maps
@@ -1729,9 +1746,17 @@ __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
       int img = rest2 / RH;
       int ih = oh0 * S - pad + ry, iw = ow0 * S - pad + rx;
       bf16x8 v = bf8_zero();
-      if (ih >= 0 && ih < H && iw >= 0 && iw < W && n + img < N)
+      if (ih >= 0 && ih < H && iw >= 0 && iw < W && n + img < N) {
         v = *reinterpret_cast<const bf16x8*>(
             &x[((long)((n + img) * H + ih) * W + iw) * CIp + cic + g8 * 8]);
+        if (AFF) {
+          const float* sc = &aff[cic + g8 * 8];
+          const float* sh = &aff[CIp + cic + g8 * 8];
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            v[j] = to_bf16(fmaxf(to_f32(v[j]) * sc[j] + sh[j], 0.f));
+        }
+      }
       *reinterpret_cast<bf16x8*>(
           &lds_x[(img * RH * RW + ry * RW + rx) * ccp + g8 * 8]) = v;
     }

@@ -601,6 +601,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("comm_world", &comm_world);
   m.def("bn_fwd_bf16", &bn_fwd_bf16);
   m.def("bn_bwd_bf16", &bn_bwd_bf16);
+  m.def("bn_stats_bf16", &bn_stats_bf16);
+  m.def("bn_bwd_fused_bf16", &bn_bwd_fused_bf16);
+  m.def("add_aff_relu_bf16", &add_aff_relu_bf16);
+  m.def("conv_tiled_s1_aff_bf16", &conv_tiled_s1_aff_bf16);
   m.def("maxpool3x3s2_fwd_bf16", &maxpool3x3s2_fwd_bf16);
   m.def("maxpool3x3s2_bwd_bf16", &maxpool3x3s2_bwd_bf16);
   m.def("gap_fwd_bf16", &gap_fwd_bf16);

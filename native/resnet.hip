@@ -98,7 +98,13 @@ __global__ __launch_bounds__(256) void bn_finalize_fold_k(
     const float* __restrict__ ws, float* __restrict__ mean,
     float* __restrict__ invstd, float* __restrict__ running_mean,
     float* __restrict__ running_var, long npix, int C, int Cp, float eps,
-    float momentum, int update_running, int S) {
+    float momentum, int update_running, int S,
+    const float* __restrict__ gamma = nullptr,
+    const float* __restrict__ beta = nullptr,
+    float* __restrict__ aff = nullptr) {
+  // aff != nullptr: additionally emit the per-channel affine the BN-apply
+  // FUSION consumes (aff[c] = gamma*invstd, aff[Cp+c] = beta - mean*scale;
+  // pads -> 0 so fused staging zeroes padded channels) — no extra launch.
   int c = blockIdx.x * 4 + (threadIdx.x >> 6);
   if (c >= Cp) return;
   int l = threadIdx.x & 63;
@@ -106,6 +112,10 @@ __global__ __launch_bounds__(256) void bn_finalize_fold_k(
     if (l == 0) {
       mean[c] = 0.f;
       invstd[c] = 0.f;
+      if (aff != nullptr) {
+        aff[c] = 0.f;
+        aff[Cp + c] = 0.f;
+      }
     }
     return;
   }
@@ -124,8 +134,14 @@ __global__ __launch_bounds__(256) void bn_finalize_fold_k(
   if (l != 0) return;
   float m = s / (float)npix;
   float var = fmaxf(ss / (float)npix - m * m, 0.f);
+  float is = rsqrtf(var + eps);
   mean[c] = m;
-  invstd[c] = rsqrtf(var + eps);
+  invstd[c] = is;
+  if (aff != nullptr) {
+    float sc = gamma[c] * is;
+    aff[c] = sc;
+    aff[Cp + c] = beta[c] - m * sc;
+  }
   if (update_running) {
     float unbiased = var * (float)npix / (float)max(1L, npix - 1);
     running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * m;
@@ -210,25 +226,35 @@ __global__ void bn_apply_k(const bf16* __restrict__ x, bf16* __restrict__ y,
 //  reduce: dbeta[c] = sum dy ; dgamma[c] = sum dy * xhat
 //  apply : dx = gamma*invstd * (dy - dbeta/npix - xhat * dgamma/npix)
 // ---------------------------------------------------------------------------
-template <bool RELU>
+// GATE: 0 = none, 1 = read the saved relu output (yact), 2 = RECOMPUTE the
+// relu gate from (x, affine) — the fused-apply path never materializes
+// y_bn, and recomputing (scale*x+shift > 0) is 2 VALU instead of a whole
+// extra tensor read.
+template <int GATE>
 __global__ void bn_bwd_reduce_k(const bf16* __restrict__ dy,
                                 const bf16* __restrict__ x,
                                 const bf16* __restrict__ yact,
                                 const float* __restrict__ mean,
                                 const float* __restrict__ invstd,
                                 float* __restrict__ red_out,  // [C][2] partial
-                                long npix, int Cp) {
+                                long npix, int Cp,
+                                const float* __restrict__ gamma = nullptr,
+                                const float* __restrict__ beta = nullptr) {
   int ncg = Cp / 8;
   int cg = threadIdx.x % ncg;
   int pl = threadIdx.x / ncg;
   int lanes = blockDim.x / ncg;
   __shared__ float red[256][8];
   float db[8] = {}, dg[8] = {};
-  float mu[8], is[8];
+  float mu[8], is[8], sc[8], sh[8];
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
     mu[j] = mean[cg * 8 + j];
     is[j] = invstd[cg * 8 + j];
+    if (GATE == 2) {
+      sc[j] = gamma[cg * 8 + j] * is[j];
+      sh[j] = beta[cg * 8 + j] - mu[j] * sc[j];
+    }
   }
   if (pl < lanes) {
     for (long i = (long)blockIdx.x * lanes + pl; i < npix;
@@ -236,13 +262,17 @@ __global__ void bn_bwd_reduce_k(const bf16* __restrict__ dy,
       bf16x8 g = *reinterpret_cast<const bf16x8*>(&dy[i * Cp + cg * 8]);
       bf16x8 v = *reinterpret_cast<const bf16x8*>(&x[i * Cp + cg * 8]);
       bf16x8 ya;
-      if (RELU) ya = *reinterpret_cast<const bf16x8*>(&yact[i * Cp + cg * 8]);
+      if (GATE == 1) ya = *reinterpret_cast<const bf16x8*>(&yact[i * Cp + cg * 8]);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         float gf = to_f32(g[j]);
-        if (RELU && !(to_f32(ya[j]) > 0.f)) gf = 0.f;
+        float xf = to_f32(v[j]);
+        if (GATE == 1 && !(to_f32(ya[j]) > 0.f)) gf = 0.f;
+        // the fused forward stored bf16(max(sc*x+sh, 0)): gate on the SAME
+        // bf16-rounded value so backward matches the forward bit-for-bit
+        if (GATE == 2 && !(to_f32(to_bf16(sc[j] * xf + sh[j])) > 0.f)) gf = 0.f;
         db[j] += gf;
-        dg[j] += gf * (to_f32(v[j]) - mu[j]) * is[j];
+        dg[j] += gf * (xf - mu[j]) * is[j];
       }
     }
   }
@@ -270,7 +300,7 @@ __global__ void bn_bwd_reduce_k(const bf16* __restrict__ dy,
   }
 }
 
-template <bool RELU>
+template <int GATE>
 __global__ void bn_bwd_apply_k(const bf16* __restrict__ dy,
                                const bf16* __restrict__ x,
                                const bf16* __restrict__ yact,
@@ -278,13 +308,14 @@ __global__ void bn_bwd_apply_k(const bf16* __restrict__ dy,
                                const float* __restrict__ invstd,
                                const float* __restrict__ gamma,
                                const float* __restrict__ red,
-                               bf16* __restrict__ dx, long npix, int Cp, int C) {
+                               bf16* __restrict__ dx, long npix, int Cp, int C,
+                               const float* __restrict__ beta = nullptr) {
   int ncg = Cp / 8;
   int cg = threadIdx.x % ncg;
   int pl = threadIdx.x / ncg;
   int lanes = blockDim.x / ncg;
   if (pl >= lanes) return;
-  float mu[8], is[8], gsc[8], mdb[8], mdg[8];
+  float mu[8], is[8], gsc[8], mdb[8], mdg[8], sh[8];
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
     int c = cg * 8 + j;
@@ -294,8 +325,9 @@ __global__ void bn_bwd_apply_k(const bf16* __restrict__ dy,
       gsc[j] = gamma[c] * is[j];
       mdb[j] = red[c * 2] / (float)npix;
       mdg[j] = red[c * 2 + 1] / (float)npix;
+      if (GATE == 2) sh[j] = beta[c] - mu[j] * gsc[j];
     } else {
-      mu[j] = is[j] = gsc[j] = mdb[j] = mdg[j] = 0.f;
+      mu[j] = is[j] = gsc[j] = mdb[j] = mdg[j] = sh[j] = 0.f;
     }
   }
   for (long i = (long)blockIdx.x * lanes + pl; i < npix;
@@ -303,13 +335,15 @@ __global__ void bn_bwd_apply_k(const bf16* __restrict__ dy,
     bf16x8 g = *reinterpret_cast<const bf16x8*>(&dy[i * Cp + cg * 8]);
     bf16x8 v = *reinterpret_cast<const bf16x8*>(&x[i * Cp + cg * 8]);
     bf16x8 ya;
-    if (RELU) ya = *reinterpret_cast<const bf16x8*>(&yact[i * Cp + cg * 8]);
+    if (GATE == 1) ya = *reinterpret_cast<const bf16x8*>(&yact[i * Cp + cg * 8]);
     bf16x8 o;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       float gf = to_f32(g[j]);
-      if (RELU && !(to_f32(ya[j]) > 0.f)) gf = 0.f;
-      float xhat = (to_f32(v[j]) - mu[j]) * is[j];
+      float xf = to_f32(v[j]);
+      if (GATE == 1 && !(to_f32(ya[j]) > 0.f)) gf = 0.f;
+      if (GATE == 2 && !(to_f32(to_bf16(gsc[j] * xf + sh[j])) > 0.f)) gf = 0.f;
+      float xhat = (xf - mu[j]) * is[j];
       float val = gsc[j] * (gf - mdb[j] - xhat * mdg[j]);
       o[j] = to_bf16(val);
     }
@@ -465,6 +499,34 @@ __global__ void add_bf16_k(const bf16* __restrict__ a, const bf16* __restrict__ 
       float f = to_f32(va[j]) + to_f32(vb[j]);
       if (RELU) f = fmaxf(f, 0.f);
       o[j] = to_bf16(f);
+    }
+    *reinterpret_cast<bf16x8*>(out + i * 8) = o;
+  }
+}
+
+// Residual add with the BN-APPLY of one or both inputs FUSED (a = bn2's raw
+// conv output, b = the downsample conv's raw output): the separate
+// bn_apply passes (write + re-read of each whole tensor) disappear. Inputs
+// round to bf16 after their affine so the result is bit-identical to the
+// unfused apply-then-add.
+template <bool AFFA, bool AFFB>
+__global__ void add_aff_relu_bf16_k(
+    const bf16* __restrict__ a, const bf16* __restrict__ b,
+    bf16* __restrict__ out, long n8, int Cp8,
+    const float* __restrict__ affa, const float* __restrict__ affb, int Cp) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n8;
+       i += (long)gridDim.x * blockDim.x) {
+    int cg = (int)(i % Cp8) * 8;
+    bf16x8 va = *reinterpret_cast<const bf16x8*>(a + i * 8);
+    bf16x8 vb = *reinterpret_cast<const bf16x8*>(b + i * 8);
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float fa = to_f32(va[j]);
+      float fb = to_f32(vb[j]);
+      if (AFFA) fa = to_f32(to_bf16(fa * affa[cg + j] + affa[Cp + cg + j]));
+      if (AFFB) fb = to_f32(to_bf16(fb * affb[cg + j] + affb[Cp + cg + j]));
+      o[j] = to_bf16(fmaxf(fa + fb, 0.f));
     }
     *reinterpret_cast<bf16x8*>(out + i * 8) = o;
   }

@@ -59,7 +59,8 @@ def main():
                                                    KH - 1 - pad, False, True)) \
             if stride == 1 else float("nan")
         wg = timeit(lambda: ext.conv2d_wgrad_bf16(dy, x, CIcan, CO, KH, KH,
-                                                  stride, pad, True))
+                                                  stride, pad, True,
+                                                  torch.Tensor()))
         print(f"{name:14} {fg:11.1f} {ft:10.1f} {dg:10.1f} {dt:9.1f} {wg:8.1f}")
 
     # stride-2 shapes: the host dispatchers route these to the stem /
@@ -81,7 +82,8 @@ def main():
         fg = timeit(lambda: ext.conv2d_fwd_bf16(x, w, b, COp, stride, pad, False))
         dg = timeit(lambda: ext.conv2d_dgrad_bf16(dy, w, H, W, CIp, stride, pad))
         wg = timeit(lambda: ext.conv2d_wgrad_bf16(dy, x, CIcan, CO, KH, KH,
-                                                  stride, pad, True))
+                                                  stride, pad, True,
+                                                  torch.Tensor()))
         print(f"{name:14} fwd={fg:9.1f} dgrad={dg:9.1f} wgrad={wg:9.1f}")
 
     # LeNet fused stem (conv1+pool+relu) fwd + its fused wgrad
