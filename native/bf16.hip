@@ -34,6 +34,32 @@ DEV bf16x8 bf8_zero() {
   return z;
 }
 
+// Hardware-bounds-checked 16-B gather (guide T8): a dead lane's offset is
+// pushed past num_records and the load returns 0 IN HARDWARE — no exec
+// predication, so hipcc can COUNT the load in its s_waitcnt bookkeeping.
+// (An if-guarded gather is uncountable: the compiler wraps it in
+// s_and_saveexec and drains vmcnt(0) at every consumer — measured as the
+// dominant stall of every gather/staging loop this engine has.)
+using buf_rsrc_t = __amdgpu_buffer_rsrc_t;
+DEV buf_rsrc_t make_bf16_rsrc(const bf16* p, long n_elems) {
+  // tensors >= 2 GB fall back to predicated loads (u32 offset limit);
+  // callers gate on rsrc_ok()
+  return __builtin_amdgcn_make_buffer_rsrc((void*)p, (short)0,
+                                           (int)(n_elems * 2), 0x00020000);
+}
+DEV bool rsrc_fits(long n_elems) { return n_elems * 2 < (1LL << 31); }
+DEV bf16x8 buf_load8(buf_rsrc_t r, long elem_off, bool ok) {
+  int boff = ok ? (int)(elem_off * 2) : 0x7ffffff0;
+  f32x4 t = __builtin_amdgcn_raw_buffer_load_b128(r, boff, 0, 0);
+  return __builtin_bit_cast(bf16x8, t);
+}
+using f32x2_v = __attribute__((ext_vector_type(2))) float;
+DEV bf16x4 buf_load4(buf_rsrc_t r, long elem_off, bool ok) {
+  int boff = ok ? (int)(elem_off * 2) : 0x7ffffff0;
+  f32x2_v t = __builtin_amdgcn_raw_buffer_load_b64(r, boff, 0, 0);
+  return __builtin_bit_cast(bf16x4, t);
+}
+
 // ---------------------------------------------------------------------------
 // Mapping probe: one wave computes D = A(16x32) @ B(32x16) via the assumed
 // fragment mapping; the GPU test compares against torch.matmul.
@@ -586,6 +612,12 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_bf16_k(
   int m_begin = ms * m_per_split;
   int m_end = min(M, m_begin + m_per_split);
 
+  const long x_elems = (long)(M / OHOW) * H * W * CIp;   // N*H*W*CIp
+  const long dy_elems = (long)M * COp;
+  const bool use_buf = rsrc_fits(x_elems) && rsrc_fits(dy_elems);
+  buf_rsrc_t xr = make_bf16_rsrc(x, x_elems);
+  buf_rsrc_t dyr = make_bf16_rsrc(dy, dy_elems);
+
   for (int mc = m_begin; mc < m_end; mc += MC) {
     __syncthreads();
     if (tid < MC) {
@@ -606,18 +638,21 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_bf16_k(
       int g = kf * 2 + half;
       int m = mc + mloc;
       int k = k0 + g * 8;
-      bf16x8 v = bf8_zero();
-      if (m < m_end && k < K) {
-        bool ok = true;
-        if (BOUNDS) {
-          int hw = lds_mhw[mloc];
-          int khw = g_khw[g];
-          int ihh = (hw >> 16) - 16384 + (khw >> 16);
-          int iww = (hw & 0xffff) - 16384 + (khw & 0xffff);
-          ok = ihh >= 0 && ihh < H && iww >= 0 && iww < W;
-        }
-        if (ok)
-          v = *reinterpret_cast<const bf16x8*>(&x[lds_mbase[mloc] + g_off[g]]);
+      bool ok = m < m_end && k < K;
+      if (BOUNDS && ok) {
+        int hw = lds_mhw[mloc];
+        int khw = g_khw[g];
+        int ihh = (hw >> 16) - 16384 + (khw >> 16);
+        int iww = (hw & 0xffff) - 16384 + (khw & 0xffff);
+        ok = ihh >= 0 && ihh < H && iww >= 0 && iww < W;
+      }
+      long off = (long)lds_mbase[mloc] + g_off[g];
+      bf16x8 v;
+      if (use_buf) {
+        v = buf_load8(xr, off, ok);   // countable: no vmcnt(0) per stage
+      } else {
+        v = bf8_zero();
+        if (ok) v = *reinterpret_cast<const bf16x8*>(&x[off]);
       }
       *reinterpret_cast<bf16x8*>(&lds_x[kf][mloc][half * 8]) = v;
     }
@@ -627,10 +662,15 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_bf16_k(
       int mloc = rest % MC;
       int ni = rest / MC;
       int m = mc + mloc;
-      bf16x8 v = bf8_zero();
       int cb = co0 + ni * 16 + half * 8;
-      if (m < m_end && cb < COp)
-        v = *reinterpret_cast<const bf16x8*>(&dy[(long)m * COp + cb]);
+      bool ok = m < m_end && cb < COp;
+      bf16x8 v;
+      if (use_buf) {
+        v = buf_load8(dyr, (long)m * COp + cb, ok);
+      } else {
+        v = bf8_zero();
+        if (ok) v = *reinterpret_cast<const bf16x8*>(&dy[(long)m * COp + cb]);
+      }
       *reinterpret_cast<bf16x8*>(&lds_dy[ni][mloc][half * 8]) = v;
     }
     __syncthreads();
@@ -755,6 +795,12 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_rs_bf16_k(
   f32x4 acc[2][9] = {};   // [ci-frag][tap]
   float db_part = 0.f;    // bias grad: colsum of this wave's dy fragments
 
+  const long x_elems = (long)(M / OW) * S * Wi * CIp;  // N*H*W*CIp
+  const long dy_elems = (long)M * COp;
+  const bool use_buf = rsrc_fits(x_elems) && rsrc_fits(dy_elems);
+  buf_rsrc_t xr = make_bf16_rsrc(x, x_elems);
+  buf_rsrc_t dyr = make_bf16_rsrc(dy, dy_elems);
+
   for (int mc = m_begin; mc < m_end; mc += MC) {
     const int g0 = mc / OW;               // first OUTPUT global row of chunk
     const int mlast = min(mc + MC - 1, m_end - 1);
@@ -777,17 +823,21 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_rs_bf16_k(
       int r = p / W2;
       int gi = S * g0 - 1 + r;             // input global row
       int iw = c - 1;
-      bf16x8 v = bf8_zero();
-      if (gi >= 0 && gi < S * (M / OW) && iw >= 0 && iw < Wi) {
-        v = *reinterpret_cast<const bf16x8*>(
-            &x[((long)gi * Wi + iw) * CIp + cic0 + g8 * 8]);
-        if (AFF) {
-          const float* sc = &aff[cic0 + g8 * 8];
-          const float* sh = &aff[CIp + cic0 + g8 * 8];
+      bool ok = gi >= 0 && gi < S * (M / OW) && iw >= 0 && iw < Wi;
+      long off = ((long)gi * Wi + iw) * CIp + cic0 + g8 * 8;
+      bf16x8 v;
+      if (use_buf) {
+        v = buf_load8(xr, off, ok);   // countable: no vmcnt(0) per stage
+      } else {
+        v = bf8_zero();
+        if (ok) v = *reinterpret_cast<const bf16x8*>(&x[off]);
+      }
+      if (AFF && ok) {
+        const float* sc = &aff[cic0 + g8 * 8];
+        const float* sh = &aff[CIp + cic0 + g8 * 8];
 #pragma unroll
-          for (int j = 0; j < 8; ++j)
-            v[j] = to_bf16(fmaxf(to_f32(v[j]) * sc[j] + sh[j], 0.f));
-        }
+        for (int j = 0; j < 8; ++j)
+          v[j] = to_bf16(fmaxf(to_f32(v[j]) * sc[j] + sh[j], 0.f));
       }
       *reinterpret_cast<bf16x8*>(&lds_reg[(p << 5) + g8 * 8]) = v;
     }
@@ -798,10 +848,15 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_rs_bf16_k(
       int mloc = rest % MC;
       int ni = rest / MC;
       int m = mc + mloc;
-      bf16x8 v = bf8_zero();
       int cb = co0 + ni * 16 + half * 8;
-      if (m < m_end && cb < COp)
-        v = *reinterpret_cast<const bf16x8*>(&dy[(long)m * COp + cb]);
+      bool ok = m < m_end && cb < COp;
+      bf16x8 v;
+      if (use_buf) {
+        v = buf_load8(dyr, (long)m * COp + cb, ok);
+      } else {
+        v = bf8_zero();
+        if (ok) v = *reinterpret_cast<const bf16x8*>(&dy[(long)m * COp + cb]);
+      }
       *reinterpret_cast<bf16x8*>(&plane(ni)[mloc * 16 + half * 8]) = v;
     }
     __syncthreads();
@@ -1549,6 +1604,13 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_pool_bf16_k(
   int m_begin = ms * m_per_split;
   int m_end = min(M, m_begin + m_per_split);
 
+  // buffer-load staging measured SLOWER here (128 -> 138 us,
+  // gpurun_out/micro_r2u.txt) unlike every other staging loop: keep the
+  // predicated path with the per-element tail
+  const bool use_buf = false;
+  buf_rsrc_t xr = make_bf16_rsrc(x, 8);
+  (void)xr;
+
   for (int mc = m_begin; mc < m_end; mc += MC) {
     __syncthreads();
     for (int mi = tid; mi < MC; mi += 256) {
@@ -1570,18 +1632,24 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_pool_bf16_k(
       int g = kf * 2 + half;
       int m = mc + mloc;
       int k = k0 + g * 8;
-      bf16x8 v = bf8_zero();
-      if (m < m_end && k < K) {
-        long a = (long)lds_mbase[mloc] + g_off[g];
-        if (a + 8 > x_elems) {
-          // per-element tail: a whole-vector clamp would SHIFT the
-          // still-valid low k slots for the last image's final rows
-          // (latent in round 1; caught by the f32 twin's tighter test)
+      bool ok = m < m_end && k < K;
+      long a = (long)lds_mbase[mloc] + g_off[g];
+      bf16x8 v;
+      if (use_buf) {
+        // hardware bounds-check handles BOTH the dead-m mask and the
+        // per-element tensor-tail cut (the round-1 whole-vector clamp
+        // shifted still-valid k slots); loads stay countable
+        v = buf_load8(xr, a, ok);
+      } else {
+        v = bf8_zero();
+        if (ok) {
+          if (a + 8 > x_elems) {
 #pragma unroll
-          for (int j = 0; j < 8; ++j)
-            v[j] = (a + j < x_elems) ? x[a + j] : to_bf16(0.f);
-        } else {
-          v = *reinterpret_cast<const bf16x8*>(&x[a]);
+            for (int j = 0; j < 8; ++j)
+              v[j] = (a + j < x_elems) ? x[a + j] : to_bf16(0.f);
+          } else {
+            v = *reinterpret_cast<const bf16x8*>(&x[a]);
+          }
         }
       }
       *reinterpret_cast<bf16x8*>(&lds_x[kf][mloc][half * 8]) = v;
@@ -1733,6 +1801,9 @@ __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
   f32x4 acc[MW][NW] = {};
 
   constexpr int KHKW = KH * KH;
+  const long x_elems = (long)N * H * W * CIp;
+  const bool use_buf = rsrc_fits(x_elems);
+  buf_rsrc_t xr = make_bf16_rsrc(x, x_elems);
   for (int cic = 0; cic < CIp; cic += CC) {
     int cc = min(CC, CIp - cic);
     int ccp = cc + 8;
@@ -1745,17 +1816,23 @@ __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
       int ry = rest2 % RH;
       int img = rest2 / RH;
       int ih = oh0 * S - pad + ry, iw = ow0 * S - pad + rx;
-      bf16x8 v = bf8_zero();
-      if (ih >= 0 && ih < H && iw >= 0 && iw < W && n + img < N) {
-        v = *reinterpret_cast<const bf16x8*>(
-            &x[((long)((n + img) * H + ih) * W + iw) * CIp + cic + g8 * 8]);
-        if (AFF) {
-          const float* sc = &aff[cic + g8 * 8];
-          const float* sh = &aff[CIp + cic + g8 * 8];
+      bool ok = ih >= 0 && ih < H && iw >= 0 && iw < W && n + img < N;
+      long off = ((long)((n + img) * H + ih) * W + iw) * CIp + cic + g8 * 8;
+      bf16x8 v;
+      if (use_buf) {
+        // unconditional hardware-bounds-checked load: keeps the staging
+        // loop countable (no per-iteration vmcnt(0) drain — see buf_load8)
+        v = buf_load8(xr, off, ok);
+      } else {
+        v = bf8_zero();
+        if (ok) v = *reinterpret_cast<const bf16x8*>(&x[off]);
+      }
+      if (AFF && ok) {
+        const float* sc = &aff[cic + g8 * 8];
+        const float* sh = &aff[CIp + cic + g8 * 8];
 #pragma unroll
-          for (int j = 0; j < 8; ++j)
-            v[j] = to_bf16(fmaxf(to_f32(v[j]) * sc[j] + sh[j], 0.f));
-        }
+        for (int j = 0; j < 8; ++j)
+          v[j] = to_bf16(fmaxf(to_f32(v[j]) * sc[j] + sh[j], 0.f));
       }
       *reinterpret_cast<bf16x8*>(
           &lds_x[(img * RH * RW + ry * RW + rx) * ccp + g8 * 8]) = v;
@@ -1894,6 +1971,12 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_stem_bf16_k(
   f32x4 acc[2][7] = {};  // [kf][kh]
   float db_part = 0.f;
 
+  const long x_elems = 2L * (M / OW) * Wi * 8;
+  const long dy_elems = (long)M * COp;
+  const bool use_buf = rsrc_fits(x_elems) && rsrc_fits(dy_elems);
+  buf_rsrc_t xr = make_bf16_rsrc(x, x_elems);
+  buf_rsrc_t dyr = make_bf16_rsrc(dy, dy_elems);
+
   for (int mc = m_begin; mc < m_end; mc += MC) {
     const int g0 = mc / OW;
     const int mlast = min(mc + MC - 1, m_end - 1);
@@ -1908,14 +1991,21 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_stem_bf16_k(
       lds_mm[tid] = (px << 8) | (m < m_end ? oy + 1 : 0);
     }
     // x region rows gi in [2*g0-3, ...]: 4-packed channels, b64 stores
+    // (buffer loads: countable, no per-stage vmcnt(0) — see buf_load8)
     for (int idx = tid; idx < rows * W2; idx += 256) {
       int c = idx % W2;
       int r = idx / W2;
       int gi = 2 * g0 - 3 + r;
       int iw = c - 3;
-      bf16x4 v = {};
-      if (gi >= 0 && gi < 2 * (M / OW) && iw >= 0 && iw < Wi)
-        v = *reinterpret_cast<const bf16x4*>(&x[((long)gi * Wi + iw) * 8]);
+      bool ok = gi >= 0 && gi < 2 * (M / OW) && iw >= 0 && iw < Wi;
+      long off = ((long)gi * Wi + iw) * 8;
+      bf16x4 v;
+      if (use_buf) {
+        v = buf_load4(xr, off, ok);
+      } else {
+        v = bf16x4{};
+        if (ok) v = *reinterpret_cast<const bf16x4*>(&x[off]);
+      }
       *reinterpret_cast<bf16x4*>(&lds_reg[(r * PXCOLS + c) * 4]) = v;
     }
     // dy planes [ni][MC][16co]
@@ -1925,10 +2015,15 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_stem_bf16_k(
       int mloc = rest % MC;
       int ni = rest / MC;
       int m = mc + mloc;
-      bf16x8 v = bf8_zero();
       int cb = co0 + ni * 16 + half * 8;
-      if (m < m_end && cb < COp)
-        v = *reinterpret_cast<const bf16x8*>(&dy[(long)m * COp + cb]);
+      bool ok = m < m_end && cb < COp;
+      bf16x8 v;
+      if (use_buf) {
+        v = buf_load8(dyr, (long)m * COp + cb, ok);
+      } else {
+        v = bf8_zero();
+        if (ok) v = *reinterpret_cast<const bf16x8*>(&dy[(long)m * COp + cb]);
+      }
       *reinterpret_cast<bf16x8*>(&plane(ni)[mloc * 16 + half * 8]) = v;
     }
     __syncthreads();
@@ -2062,6 +2157,9 @@ __global__ __launch_bounds__(256) void conv_tiled_dgrad_s2_bf16_k(
   // a class with no live taps (1x1 downsample, classes 1-3) skips straight
   // to the zero-writing epilogue
   const int k_end = ((tap_mask >> (blockIdx.z * 4)) & 15) ? COp : 0;
+  const long dy_elems = (long)N * OH * OW * COp;
+  const bool use_buf = rsrc_fits(dy_elems);
+  buf_rsrc_t dyr = make_bf16_rsrc(dy, dy_elems);
   for (int cic = 0; cic < k_end; cic += 64) {
     int cc = min(64, COp - cic);
     int ccp = cc + 8;
@@ -2074,11 +2172,15 @@ __global__ __launch_bounds__(256) void conv_tiled_dgrad_s2_bf16_k(
       int ry = rest2 % RH;
       int img = rest2 / RH;
       int ih = oh0 + ry, iw = ow0 + rx;
-      bf16x8 v = bf8_zero();
-      if (ih < OH && iw < OW && n + img < N)
-        v = *reinterpret_cast<const bf16x8*>(
-            &dy[((long)((n + img) * OH + ih) * OW + iw) * COp + cic +
-                g8 * 8]);
+      bool ok = ih < OH && iw < OW && n + img < N;
+      long off = ((long)((n + img) * OH + ih) * OW + iw) * COp + cic + g8 * 8;
+      bf16x8 v;
+      if (use_buf) {
+        v = buf_load8(dyr, off, ok);   // countable, no per-stage vmcnt(0)
+      } else {
+        v = bf8_zero();
+        if (ok) v = *reinterpret_cast<const bf16x8*>(&dy[off]);
+      }
       *reinterpret_cast<bf16x8*>(
           &lds_y[(img * RH * RW + ry * RW + rx) * ccp + g8 * 8]) = v;
     }
@@ -2093,25 +2195,32 @@ __global__ __launch_bounds__(256) void conv_tiled_dgrad_s2_bf16_k(
       const bf16* wtap = wcls + ((long)tap * CIp + ci0) * COp + cic;
 #pragma unroll
       for (int ks = 0; ks < 64; ks += 32) {
+        // hoisted A reads + clamped unconditional loads (wait-collapse,
+        // same fix as conv_tiled_s1_bf16_k)
         bool gin = lk * 8 < cc - ks;
+        int bsel = gin ? ks + lk * 8 : 0;
         bf16x8 bw[NW];
 #pragma unroll
-        for (int ni = 0; ni < NW; ++ni)
-          bw[ni] = gin ? *reinterpret_cast<const bf16x8*>(
-                             &wtap[(long)(ni * 16 + lm) * COp + ks + lk * 8])
-                       : bf8_zero();
+        for (int ni = 0; ni < NW; ++ni) {
+          bf16x8 t = *reinterpret_cast<const bf16x8*>(
+              &wtap[(long)(ni * 16 + lm) * COp + bsel]);
+          bw[ni] = gin ? t : bf8_zero();
+        }
+        bf16x8 av[MW];
 #pragma unroll
         for (int mi = 0; mi < MW; ++mi) {
           int px = (mi % (MW / NIMG)) * 8 + (lm & 7);
           int img = mi / (MW / NIMG);
           const bf16* ypix =
               &lds_y[(img * RH * RW + (py + a) * RW + (px + b)) * ccp];
-          bf16x8 av = gin ? *reinterpret_cast<const bf16x8*>(&ypix[ks + lk * 8])
-                          : bf8_zero();
+          bf16x8 t = *reinterpret_cast<const bf16x8*>(&ypix[bsel]);
+          av[mi] = gin ? t : bf8_zero();
+        }
+#pragma unroll
+        for (int mi = 0; mi < MW; ++mi)
 #pragma unroll
           for (int ni = 0; ni < NW; ++ni)
-            acc[mi][ni] = mfma16x16x32bf(av, bw[ni], acc[mi][ni]);
-        }
+            acc[mi][ni] = mfma16x16x32bf(av[mi], bw[ni], acc[mi][ni]);
       }
     }
   }
@@ -2203,15 +2312,23 @@ __global__ __launch_bounds__(256) void conv_stem_s2_bf16_k(
   __shared__ bf16 lds_x[RH * RW * 4 + 8];
 
   // stage the input region ONCE, dropping input channels 4..7 (pure pad):
-  // one 8-byte store per region pixel
+  // one 8-byte store per region pixel (buffer loads: countable waits)
   const int ih0 = oh0 * 2 - pad, iw0 = ow0 * 2 - pad;
+  const long x_elems = (long)N * H * W * 8;
+  const bool use_buf = rsrc_fits(x_elems);
+  buf_rsrc_t xr = make_bf16_rsrc(x, x_elems);
   for (int idx = tid; idx < RH * RW; idx += 256) {
     int rx = idx % RW, ry = idx / RW;
     int ih = ih0 + ry, iw = iw0 + rx;
-    bf16x4 v = {};
-    if (ih >= 0 && ih < H && iw >= 0 && iw < W)
-      v = *reinterpret_cast<const bf16x4*>(
-          &x[((long)(n * H + ih) * W + iw) * 8]);
+    bool ok = ih >= 0 && ih < H && iw >= 0 && iw < W;
+    long off = ((long)(n * H + ih) * W + iw) * 8;
+    bf16x4 v;
+    if (use_buf) {
+      v = buf_load4(xr, off, ok);
+    } else {
+      v = bf16x4{};
+      if (ok) v = *reinterpret_cast<const bf16x4*>(&x[off]);
+    }
     *reinterpret_cast<bf16x4*>(&lds_x[idx * 4]) = v;
   }
   __syncthreads();
