@@ -40,7 +40,6 @@ DEV bf16x8 bf8_zero() {
 // (An if-guarded gather is uncountable: the compiler wraps it in
 // s_and_saveexec and drains vmcnt(0) at every consumer — measured as the
 // dominant stall of every gather/staging loop this engine has.)
-using buf_rsrc_t = __amdgpu_buffer_rsrc_t;
 DEV buf_rsrc_t make_bf16_rsrc(const bf16* p, long n_elems) {
   // tensors >= 2 GB fall back to predicated loads (u32 offset limit);
   // callers gate on rsrc_ok()

@@ -29,6 +29,29 @@ DEV f32x4 mfma16x16x4(float a, float b, f32x4 c) {
   return __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, c, 0, 0, 0);
 }
 
+// ---------------------------------------------------------------------------
+// Hardware-bounds-checked buffer loads (guide T8): an if-guarded gather is
+// exec-predicated and UNCOUNTABLE for hipcc's s_waitcnt bookkeeping, which
+// forces a vmcnt(0) full drain at every consumer. A raw buffer load needs no
+// predication — a dead lane's offset is pushed past num_records and the
+// hardware returns 0. Callers gate on rsrc*_fits (u32 num_records limit).
+// ---------------------------------------------------------------------------
+using buf_rsrc_t = __amdgpu_buffer_rsrc_t;
+DEV buf_rsrc_t make_f32_rsrc(const float* p, long n_elems) {
+  return __builtin_amdgcn_make_buffer_rsrc((void*)p, (short)0,
+                                           (int)(n_elems * 4), 0x00020000);
+}
+DEV bool rsrc_fits_f32(long n_elems) { return n_elems * 4 < (1LL << 31); }
+DEV float buf_loadf(buf_rsrc_t r, long elem_off, bool ok) {
+  int boff = ok ? (int)(elem_off * 4) : 0x7ffffff0;
+  // bit_cast, NOT assignment: the b32 builtin's return type is integral on
+  // this clang, and an implicit int->float VALUE conversion of the loaded
+  // bits silently corrupts every gather
+  auto t = __builtin_amdgcn_raw_buffer_load_b32(r, boff, 0, 0);
+  static_assert(sizeof(t) == 4);
+  return __builtin_bit_cast(float, t);
+}
+
 // counter-based RNG (splitmix64 finalizer) for dropout masks: pure function
 // of (seed, stream, index) — reproducible across world sizes and replays.
 DEV unsigned long long mix64(unsigned long long z) {

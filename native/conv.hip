@@ -308,6 +308,14 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_f32_k(
   int m_begin = ms * m_per_split;
   int m_end = min(M, m_begin + m_per_split);
 
+  // countable (buffer) loads for the x gathers + dy staging: the previous
+  // exec-predicated forms forced vmcnt(0) drains per consumer (common.h)
+  const long x_elems = (long)(M / OHOW) * CI * HW;
+  const long dy_elems = (long)(M / OHOW) * CO * OHOW;
+  const bool use_buf = rsrc_fits_f32(x_elems) && rsrc_fits_f32(dy_elems);
+  buf_rsrc_t xr = make_f32_rsrc(x, x_elems);
+  buf_rsrc_t dyrs = make_f32_rsrc(dy, dy_elems);
+
   for (int mc = m_begin; mc < m_end; mc += MC) {
     __syncthreads();
     // stage dy[mc:mc+256][co0:+16] transposed (reads coalesced over m)
@@ -315,10 +323,16 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_f32_k(
       int co = idx >> 8;  // idx = co*256 + mloc
       int mloc = idx & 255;
       int m = mc + mloc;
-      float v = 0.f;
-      if (m < m_end && co0 + co < CO) {
-        int n = m / OHOW, r = m % OHOW;
-        v = dy[(n * CO + co0 + co) * OHOW + r];
+      bool ok = m < m_end && co0 + co < CO;
+      int mm = ok ? m : 0;
+      int n = mm / OHOW, r = mm % OHOW;
+      long off = (long)(n * CO + co0 + co) * OHOW + r;
+      float v;
+      if (use_buf) {
+        v = buf_loadf(dyrs, off, ok);
+      } else {
+        v = 0.f;
+        if (ok) v = dy[off];
       }
       lds_dy[mloc][co] = v;
     }
@@ -346,14 +360,17 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_f32_k(
       int hw = lds_mhw[msel];
 #pragma unroll
       for (int kf = 0; kf < KT; ++kf) {
-        float a = 0.f;
-        if (kval[kf] & min_ok) {
-          bool ok = true;
-          if (BOUNDS) {
-            int ih = (hw >> 16) - 16384 + khh[kf];
-            int iw = (hw & 0xffff) - 16384 + kww[kf];
-            ok = ih >= 0 && ih < H && iw >= 0 && iw < W;
-          }
+        bool ok = kval[kf] & min_ok;
+        if (BOUNDS && ok) {
+          int ih = (hw >> 16) - 16384 + khh[kf];
+          int iw = (hw & 0xffff) - 16384 + kww[kf];
+          ok = ih >= 0 && ih < H && iw >= 0 && iw < W;
+        }
+        float a;
+        if (use_buf) {
+          a = buf_loadf(xr, (long)mb + koff[kf], ok);
+        } else {
+          a = 0.f;
           if (ok) a = x[mb + koff[kf]];
         }
         acc[kf] = mfma16x16x4(a, b, acc[kf]);
