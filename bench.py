@@ -130,11 +130,14 @@ def main():
             graphed.slots[slot][0].copy_(x, non_blocking=True)
             graphed.slots[slot][1].copy_(t, non_blocking=True)
             return graphed.step(slot)
-        # no zero_grad: the fused SGD kernel resets grads to zero in the
-        # same stores that consume them (ops/sgd.py), so backward always
-        # accumulates onto zeros — the eager path carries no FillFunctor
-        # launches (first step: autograd assigns fresh grads / DDP installs
-        # zeroed bucket views)
+        # Single-GPU eager: ASSIGN-mode grads (p.grad = None) so backward
+        # allocates instead of accumulating — no ATen add/Fill launches in
+        # the step. The caching allocator hands back the same blocks, so
+        # the fused-SGD pointer-table cache still hits. DDP keeps its
+        # persistent bucket views (the SGD kernel zeroes them in-kernel).
+        if not isinstance(model, DistributedDataParallel):
+            for p in model.parameters():
+                p.grad = None
         _, loss = model(x, t)
         loss.backward()
         if isinstance(model, DistributedDataParallel):
