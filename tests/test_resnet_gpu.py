@@ -194,3 +194,74 @@ def test_batchnorm_relu_bwd_fused_gate():
                                rtol=5e-2, atol=5e-2)
     torch.testing.assert_close(gamma.grad, gr.grad, rtol=3e-2, atol=3e-2)
     torch.testing.assert_close(beta.grad, br.grad, rtol=3e-2, atol=3e-2)
+
+
+def test_bn_fusion_matches_unfused_block():
+    """The BN-apply fusion (bn_virtual + conv aff staging / add_aff) must
+    reproduce the unfused batchnorm_bf16 -> conv / add path bit-for-bit on
+    the forward (the staging rounds to bf16 exactly where bn_apply did) and
+    to fp32-accumulation tolerance on the gradients."""
+    import dist_mnist_amd.ops.functional_bf16 as fb
+    import dist_mnist_amd.ops.functional_resnet as fr
+
+    torch.manual_seed(77)
+    N, C, Hs = 8, 64, 28
+    x = torch.randn(N, Hs, Hs, C, device=DEV).to(torch.bfloat16)
+    gamma = torch.randn(C, device=DEV).abs() + 0.5
+    beta = torch.randn(C, device=DEV)
+    rm = torch.zeros(C, device=DEV)
+    rv = torch.ones(C, device=DEV)
+    w = (torch.randn(C, C, 3, 3, device=DEV) * 0.05).requires_grad_(True)
+
+    # fused: stats-only BN -> conv with affine staging
+    xa = x.clone().requires_grad_(True)
+    t_tok, aff = fr.bn_virtual_bf16(xa, gamma, beta, rm.clone(), rv.clone(),
+                                    training=True, relu=True)
+    y_fused = fb.conv2d_bf16(t_tok, w, pad=1, cop=C, aff=aff)
+
+    # unfused: materialized bn_apply -> plain conv
+    xb = x.clone().requires_grad_(True)
+    wb = w.detach().clone().requires_grad_(True)
+    ybn = fr.batchnorm_bf16(xb, gamma, beta, rm.clone(), rv.clone(),
+                            training=True, relu=True)
+    y_ref = fb.conv2d_bf16(ybn, wb, pad=1, cop=C)
+
+    torch.testing.assert_close(y_fused.float(), y_ref.float(), rtol=0, atol=0)
+
+    dy = torch.randn_like(y_fused.float()).to(torch.bfloat16)
+    y_fused.backward(dy)
+    y_ref.backward(dy)
+    torch.testing.assert_close(w.grad, wb.grad, rtol=1e-4, atol=1e-4)
+    torch.testing.assert_close(xa.grad.float(), xb.grad.float(),
+                               rtol=1e-4, atol=1e-4)
+
+
+def test_add_aff_fusion_matches_unfused():
+    import dist_mnist_amd.ops.functional_resnet as fr
+
+    torch.manual_seed(78)
+    N, C, Hs = 8, 64, 14
+    a = torch.randn(N, Hs, Hs, C, device=DEV).to(torch.bfloat16)
+    b = torch.randn(N, Hs, Hs, C, device=DEV).to(torch.bfloat16)
+    gamma = torch.randn(C, device=DEV).abs() + 0.5
+    beta = torch.randn(C, device=DEV)
+    rm = torch.zeros(C, device=DEV)
+    rv = torch.ones(C, device=DEV)
+
+    a1 = a.clone().requires_grad_(True)
+    tok, aff = fr.bn_virtual_bf16(a1, gamma, beta, rm.clone(), rv.clone(),
+                                  training=True, relu=False)
+    out_fused = fr.add_aff_relu_bf16(tok, b, aff)
+
+    a2 = a.clone().requires_grad_(True)
+    ybn = fr.batchnorm_bf16(a2, gamma, beta, rm.clone(), rv.clone(),
+                            training=True, relu=False)
+    out_ref = fr.add_relu_bf16(ybn, b)
+
+    torch.testing.assert_close(out_fused.float(), out_ref.float(),
+                               rtol=0, atol=0)
+    dy = torch.randn_like(out_fused.float()).to(torch.bfloat16)
+    out_fused.backward(dy)
+    out_ref.backward(dy)
+    torch.testing.assert_close(a1.grad.float(), a2.grad.float(),
+                               rtol=1e-4, atol=1e-4)
