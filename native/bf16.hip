@@ -58,6 +58,15 @@ DEV bf16x4 buf_load4(buf_rsrc_t r, long elem_off, bool ok) {
   f32x2_v t = __builtin_amdgcn_raw_buffer_load_b64(r, boff, 0, 0);
   return __builtin_bit_cast(bf16x4, t);
 }
+DEV buf_rsrc_t make_u8_rsrc(const unsigned char* p, long n_bytes) {
+  return __builtin_amdgcn_make_buffer_rsrc((void*)p, (short)0, (int)n_bytes,
+                                           0x00020000);
+}
+DEV unsigned long long buf_load_u64(buf_rsrc_t r, long byte_off, bool ok) {
+  int boff = ok ? (int)byte_off : 0x7ffffff0;
+  f32x2_v t = __builtin_amdgcn_raw_buffer_load_b64(r, boff, 0, 0);
+  return __builtin_bit_cast(unsigned long long, t);
+}
 
 // ---------------------------------------------------------------------------
 // Mapping probe: one wave computes D = A(16x32) @ B(32x16) via the assumed

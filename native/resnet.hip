@@ -416,17 +416,39 @@ __global__ void maxpool3x3s2_bwd_k(const bf16* __restrict__ dy,
     int ih = r2 % H;
     int n = r2 / H;
     float acc[8] = {};
-    // windows (oh,ow) with oh*2-1 <= ih <= oh*2+1 => oh in [(ih-1)/2, (ih+1)/2]
-    for (int oh = max(0, (ih - 1) / 2); oh <= min(OH - 1, (ih + 1) / 2); ++oh) {
-      int kh = ih - (oh * 2 - 1);
-      if (kh < 0 || kh > 2) continue;
-      for (int ow = max(0, (iw - 1) / 2); ow <= min(OW - 1, (iw + 1) / 2); ++ow) {
-        int kw = iw - (ow * 2 - 1);
-        if (kw < 0 || kw > 2) continue;
+    // windows (oh,ow) with oh*2-1 <= ih <= oh*2+1 => oh in [(ih-1)/2, (ih+1)/2].
+    // FLATTENED to a fixed unrolled 2x2 candidate set with hardware-bounds-
+    // checked loads: the runtime-bounded loop form put every dy/idx load
+    // behind divergent branches (uncountable -> vmcnt(0) drains, see
+    // buf_load8). Dead candidates read 0 and (a8==0, want==0) pairs add a
+    // zero g — harmless.
+    const long dy_elems = (long)N * OH * OW * Cp;
+    const bool use_buf = rsrc_fits(dy_elems);
+    buf_rsrc_t dyr = make_bf16_rsrc(dy, dy_elems);
+    buf_rsrc_t idxr = make_u8_rsrc(idx, dy_elems);
+    int oh0 = (ih - 1) / 2, ow0 = (iw - 1) / 2;
+#pragma unroll
+    for (int a = 0; a < 2; ++a) {
+#pragma unroll
+      for (int b = 0; b < 2; ++b) {
+        int oh = oh0 + a, ow = ow0 + b;
+        int kh = ih - (oh * 2 - 1), kw = iw - (ow * 2 - 1);
+        bool ok = oh >= 0 && oh < OH && ow >= 0 && ow < OW &&
+                  kh >= 0 && kh <= 2 && kw >= 0 && kw <= 2;
         long off = (((long)(n * OH + oh) * OW + ow) * Cp) + cg * 8;
-        unsigned long long a8 =
-            *reinterpret_cast<const unsigned long long*>(&idx[off]);
-        bf16x8 g = *reinterpret_cast<const bf16x8*>(&dy[off]);
+        unsigned long long a8;
+        bf16x8 g;
+        if (use_buf) {
+          a8 = buf_load_u64(idxr, off, ok);
+          g = buf_load8(dyr, off, ok);
+        } else {
+          a8 = 0;
+          g = bf8_zero();
+          if (ok) {
+            a8 = *reinterpret_cast<const unsigned long long*>(&idx[off]);
+            g = *reinterpret_cast<const bf16x8*>(&dy[off]);
+          }
+        }
         unsigned char want = (unsigned char)(kh * 3 + kw);
 #pragma unroll
         for (int j = 0; j < 8; ++j)
