@@ -598,20 +598,22 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_bf16_k(
   __shared__ bf16 lds_dy[NW][MC][16];       // [ni][m][16co]
   __shared__ int lds_mbase[MC];
   __shared__ int lds_mhw[MC];
+  // per-8k decode tables in LDS: per-thread int[NKF*2] arrays are
+  // runtime-indexed in the staging loop -> scratch spills (guide rule 20)
+  __shared__ int lds_goff[NKF * 2];
+  __shared__ int lds_gkhw[NKF * 2];
 
   const int KWC = KW * CIp;
   const int OHOW = OH * OW;
 
-  int g_off[NKF * 2];
-  int g_khw[NKF * 2];
-#pragma unroll
-  for (int g = 0; g < NKF * 2; ++g) {
+  if (tid < NKF * 2) {
+    int g = tid;
     int k = k0 + g * 8;
     int kcl = k < K ? k : 0;
     int kh = kcl / KWC, kr = kcl % KWC;
     int kw = kr / CIp, ci0 = kr % CIp;
-    g_off[g] = (kh * W + kw) * CIp + ci0;
-    g_khw[g] = (kh << 16) | kw;
+    lds_goff[g] = (kh * W + kw) * CIp + ci0;
+    lds_gkhw[g] = (kh << 16) | kw;
   }
 
   f32x4 acc[KT][NW] = {};
@@ -649,12 +651,12 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_bf16_k(
       bool ok = m < m_end && k < K;
       if (BOUNDS && ok) {
         int hw = lds_mhw[mloc];
-        int khw = g_khw[g];
+        int khw = lds_gkhw[g];
         int ihh = (hw >> 16) - 16384 + (khw >> 16);
         int iww = (hw & 0xffff) - 16384 + (khw & 0xffff);
         ok = ihh >= 0 && ihh < H && iww >= 0 && iww < W;
       }
-      long off = (long)lds_mbase[mloc] + g_off[g];
+      long off = (long)lds_mbase[mloc] + lds_goff[g];
       bf16x8 v;
       if (use_buf) {
         v = buf_load8(xr, off, ok);   // countable: no vmcnt(0) per stage
@@ -1592,18 +1594,20 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_pool_bf16_k(
   __shared__ bf16 lds_dy[MC][16];
   __shared__ int lds_mbase[MC];
   __shared__ int lds_mpool[MC];   // pooled-offset | quadrant (COp mult of 8)
+  // decode table in LDS (a per-thread array would be runtime-indexed in
+  // the staging loop -> scratch, guide rule 20)
+  __shared__ int lds_goff[NKF * 2];
 
   const int KROWV = KW * CIp;                 // valid entries per kh row
   const int KROW = ((KROWV + 7) / 8) * 8;     // padded row length
   const int OHOW = OH * OW;
 
-  int g_off[NKF * 2];
-#pragma unroll
-  for (int g = 0; g < NKF * 2; ++g) {
+  if (tid < NKF * 2) {
+    int g = tid;
     int k = k0 + g * 8;
     int kcl = k < K ? k : 0;
     int kh = kcl / KROW, r = kcl % KROW;
-    g_off[g] = kh * W * CIp + r;              // r may run past the real row
+    lds_goff[g] = kh * W * CIp + r;           // r may run past the real row
   }
 
   f32x4 acc[KT] = {};
@@ -1641,7 +1645,7 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_pool_bf16_k(
       int m = mc + mloc;
       int k = k0 + g * 8;
       bool ok = m < m_end && k < K;
-      long a = (long)lds_mbase[mloc] + g_off[g];
+      long a = (long)lds_mbase[mloc] + lds_goff[g];
       bf16x8 v;
       if (use_buf) {
         // hardware bounds-check handles BOTH the dead-m mask and the
