@@ -1768,13 +1768,21 @@ __global__ void repack_w_k(const float* __restrict__ w, bf16* __restrict__ w_rp,
 }
 
 template <int KH, bool HAS_BIAS, bool RELU, int NW, int MW, int NIMG = 1,
-          int S = 1, bool AFF = false>
+          int S = 1, bool AFF = false, bool BGLDS = false>
 __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
     const bf16* __restrict__ x, const bf16* __restrict__ w_rp,
     const float* __restrict__ bias, bf16* __restrict__ y,
     int CIp, int H, int W, int COp, int COcan,
     int OH, int OW, int pad, int N, int tiles_h, int tiles_w,
     const float* __restrict__ aff = nullptr) {
+  // BGLDS: the next tap's 8 KB B panel streams into a 2-deep LDS ring via
+  // global_load_lds WHILE the current tap's MFMAs run — the per-tap B
+  // loads' L2 latency was 70% wave-parked time (pmc_micro_r2.txt). The
+  // panel image is chunk-XOR-swizzled (chunk ^= (co>>1)&7) on BOTH the
+  // glds source index and the read so the b128 reads stay conflict-free
+  // without per-row padding (glds destinations must be lane-linear).
+  // Costs one LDS ring (16 KB -> occupancy 3->2 blocks/CU) + a raw
+  // barrier per tap; requires full 64-ci chunks and NW==4.
   // AFF: x is a RAW conv output whose BatchNorm(+ReLU) apply is fused into
   // this staging pass — v = bf16(max(aff[c]*v + aff[CIp+c], 0)); padded
   // channels have aff == 0 so they stay zero. Saves the separate bn_apply
@@ -1802,8 +1810,13 @@ __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
   const int oh0 = th * TH, ow0 = tw * TW;
   const int co0 = blockIdx.y * (NW * 16);
 
-  // per-pixel stride padded +8 bf16 (bank spread for the 16-pixel b128 reads)
-  __shared__ bf16 lds_x[NIMG * RH * RW * (CC + 8) + 8];
+  // per-pixel stride padded +8 bf16 (bank spread for the 16-pixel b128
+  // reads); ONE shared object (a second one would force vmcnt(0) before
+  // every ds_read beside an in-flight glds — guide §5 trap 4a): the BGLDS
+  // B-panel ring is carved off the tail
+  __shared__ bf16 lds_x[NIMG * RH * RW * (CC + 8) + 8 +
+                        (BGLDS ? 2 * 64 * CC : 0)];
+  bf16* ldsB = &lds_x[NIMG * RH * RW * (CC + 8) + 8];
 
   // wave pixel map: fragment mi covers a 2-row x 8-col patch; the block's 4
   // waves cover the 8 rows. pixel (py, px) for (wave, mi, lm):
@@ -1850,10 +1863,40 @@ __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
           &lds_x[(img * RH * RW + ry * RW + rx) * ccp + g8 * 8]) = v;
     }
     __syncthreads();
+    // glds one 8 KB tap panel: each wave DMAs two 1 KiB pieces (dest =
+    // wave-uniform base + lane*16; the SOURCE index carries the XOR
+    // swizzle so reads stay conflict-free on a pad-free panel)
+    auto stage_B = [&](int tap, int buf) {
+      if constexpr (BGLDS) {
+        constexpr int NCH = CC / 8;            // 16-B chunks per co row
+        const bf16* wt = w_rp + ((long)tap * COp + co0) * CIp + cic;
+        bf16* panel = ldsB + buf * (64 * CC);
+#pragma unroll
+        for (int i = 0; i < CC / 32; ++i) {
+          int s = (wid * (CC / 32) + i) * 64 + l;  // 16-B slot index
+          int co = s / NCH;
+          int chlog = (s % NCH) ^ ((co >> 1) & (NCH - 1));
+          const bf16* src = &wt[(long)co * CIp + chlog * 8];
+          __builtin_amdgcn_global_load_lds(
+              (const __attribute__((address_space(1))) unsigned int*)src,
+              (__attribute__((address_space(3))) unsigned int*)(panel +
+                                                               (wid * (CC / 32) +
+                                                                i) * 512),
+              16, 0, 0);
+        }
+      }
+    };
+    if (BGLDS) {
+      stage_B(0, 0);
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+    }
 #pragma unroll
     for (int tap = 0; tap < KHKW; ++tap) {
       int kh = tap / KH, kw = tap % KH;
       const bf16* wtap = w_rp + ((long)tap * COp + co0) * CIp + cic;
+      const bf16* panel = BGLDS ? ldsB + (tap & 1) * (64 * CC) : nullptr;
+      if (BGLDS && tap + 1 < KHKW) stage_B(tap + 1, (tap + 1) & 1);
       // compile-time trip count (remainder chunks handled by the gin
       // predicate): the whole tap x ks nest fully unrolls, so the compiler
       // can hoist and software-pipeline the A/B loads across taps
@@ -1871,11 +1914,23 @@ __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
           // REGRESSED with this order (194->234) and keeps the old body.
           int bsel = gin ? ks + lk * 8 : 0;
           bf16x8 bw[NW];
+          if constexpr (BGLDS) {
+            // B from the prefetched LDS panel (chunk-XOR-swizzled read)
+            int chlog = (ks >> 3) + lk;
 #pragma unroll
-          for (int ni = 0; ni < NW; ++ni) {
-            bf16x8 t = *reinterpret_cast<const bf16x8*>(
-                &wtap[(long)(ni * 16 + lm) * CIp + bsel]);
-            bw[ni] = gin ? t : bf8_zero();
+            for (int ni = 0; ni < NW; ++ni) {
+              int co = ni * 16 + lm;
+              int chst = chlog ^ ((co >> 1) & (CC / 8 - 1));
+              bw[ni] = *reinterpret_cast<const bf16x8*>(
+                  &panel[co * CC + chst * 8]);
+            }
+          } else {
+#pragma unroll
+            for (int ni = 0; ni < NW; ++ni) {
+              bf16x8 t = *reinterpret_cast<const bf16x8*>(
+                  &wtap[(long)(ni * 16 + lm) * CIp + bsel]);
+              bw[ni] = gin ? t : bf8_zero();
+            }
           }
           bf16x8 av[MW];
 #pragma unroll
@@ -1895,11 +1950,22 @@ __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
               acc[mi][ni] = mfma16x16x32bf(av[mi], bw[ni], acc[mi][ni]);
         } else {
           bf16x8 bw[NW];
+          if constexpr (BGLDS) {
+            int chlog = (ks >> 3) + lk;
 #pragma unroll
-          for (int ni = 0; ni < NW; ++ni)
-            bw[ni] = gin ? *reinterpret_cast<const bf16x8*>(
-                               &wtap[(long)(ni * 16 + lm) * CIp + ks + lk * 8])
-                         : bf8_zero();
+            for (int ni = 0; ni < NW; ++ni) {
+              int co = ni * 16 + lm;
+              int chst = chlog ^ ((co >> 1) & (CC / 8 - 1));
+              bw[ni] = *reinterpret_cast<const bf16x8*>(
+                  &panel[co * CC + chst * 8]);
+            }
+          } else {
+#pragma unroll
+            for (int ni = 0; ni < NW; ++ni)
+              bw[ni] = gin ? *reinterpret_cast<const bf16x8*>(
+                                 &wtap[(long)(ni * 16 + lm) * CIp + ks + lk * 8])
+                           : bf8_zero();
+          }
 #pragma unroll
           for (int mi = 0; mi < MW; ++mi) {
             int px = (mi % (MW / NIMG)) * 8 + (lm & 7);
@@ -1914,6 +1980,14 @@ __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
               acc[mi][ni] = mfma16x16x32bf(a, bw[ni], acc[mi][ni]);
           }
         }
+      }
+      if (BGLDS) {
+        // raw barrier + explicit drain: __syncthreads() would emit
+        // vmcnt(0) anyway with a glds in flight, but hipcc must not
+        // reorder it; after this every wave's next-tap panel piece has
+        // landed for every consumer
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
       }
     }
   }
@@ -2138,7 +2212,7 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_stem_bf16_k(
 // per-pixel loads). Epilogue writes dx[2y+p][2x+q] directly.
 // blockIdx.z = class; per-class output dims Hc x Wc derived in-kernel.
 // ---------------------------------------------------------------------------
-template <int NW, int MW, int NIMG>
+template <int NW, int MW, int NIMG, bool BGLDS = false>
 __global__ __launch_bounds__(256) void conv_tiled_dgrad_s2_bf16_k(
     const bf16* __restrict__ dy, const bf16* __restrict__ wc,
     bf16* __restrict__ dx, int COp, int OH, int OW, int CIp, int CIcan,
@@ -2161,7 +2235,8 @@ __global__ __launch_bounds__(256) void conv_tiled_dgrad_s2_bf16_k(
   const int Hc = (H - p + 1) >> 1, Wc = (W - q + 1) >> 1;
   const bf16* wcls = wc + (long)blockIdx.z * 4 * CIp * COp;
 
-  __shared__ bf16 lds_y[NIMG * RH * RW * 72 + 8];
+  __shared__ bf16 lds_y[NIMG * RH * RW * 72 + 8 + (BGLDS ? 2 * 64 * 64 : 0)];
+  bf16* ldsB = &lds_y[NIMG * RH * RW * 72 + 8];
 
   const int py = wid * 2 + (lm >> 3);
   f32x4 acc[MW][NW] = {};
@@ -2197,14 +2272,48 @@ __global__ __launch_bounds__(256) void conv_tiled_dgrad_s2_bf16_k(
           &lds_y[(img * RH * RW + ry * RW + rx) * ccp + g8 * 8]) = v;
     }
     __syncthreads();
+    // glds-pipelined tap weight panels (same recipe as conv_tiled_s1's
+    // BGLDS; dead taps still stage/barrier so the loop stays uniform)
+    auto stage_B = [&](int tap, int buf) {
+      if constexpr (BGLDS) {
+        const bf16* wt = wcls + ((long)tap * CIp + ci0) * COp + cic;
+        bf16* panel = ldsB + buf * (64 * 64);
+#pragma unroll
+        for (int i = 0; i < 2; ++i) {
+          int sl = (wid * 2 + i) * 64 + l;
+          int ci = sl >> 3;
+          int chlog = (sl & 7) ^ ((ci >> 1) & 7);
+          const bf16* src = &wt[(long)ci * COp + chlog * 8];
+          __builtin_amdgcn_global_load_lds(
+              (const __attribute__((address_space(1))) unsigned int*)src,
+              (__attribute__((address_space(3))) unsigned int*)(panel +
+                                                               (wid * 2 + i) *
+                                                                   512),
+              16, 0, 0);
+        }
+      }
+    };
+    if (BGLDS) {
+      stage_B(0, 0);
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+    }
 #pragma unroll
     for (int tap = 0; tap < 4; ++tap) {
+      const bf16* panel = BGLDS ? ldsB + (tap & 1) * (64 * 64) : nullptr;
+      if (BGLDS && tap + 1 < 4) stage_B(tap + 1, (tap + 1) & 1);
       // live-tap mask (host bit cls*4+tap): classes carry 1/2/2/4 live
       // taps for 3x3 pad-1 and a single live tap for 1x1 — skipping the
       // zero-weight ones is ~44% of the MFMA work (block-uniform branch)
-      if (!((tap_mask >> (blockIdx.z * 4 + tap)) & 1)) continue;
+      bool live = ((tap_mask >> (blockIdx.z * 4 + tap)) & 1) != 0;
+      if (!BGLDS && !live) continue;
       int a = tap >> 1, b = tap & 1;
       const bf16* wtap = wcls + ((long)tap * CIp + ci0) * COp + cic;
+      if (!live) {  // BGLDS: keep the stage/barrier tail uniform
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+        continue;
+      }
 #pragma unroll
       for (int ks = 0; ks < 64; ks += 32) {
         // hoisted A reads + clamped unconditional loads (wait-collapse,
@@ -2212,11 +2321,22 @@ __global__ __launch_bounds__(256) void conv_tiled_dgrad_s2_bf16_k(
         bool gin = lk * 8 < cc - ks;
         int bsel = gin ? ks + lk * 8 : 0;
         bf16x8 bw[NW];
+        if constexpr (BGLDS) {
+          int chlog = (ks >> 3) + lk;
 #pragma unroll
-        for (int ni = 0; ni < NW; ++ni) {
-          bf16x8 t = *reinterpret_cast<const bf16x8*>(
-              &wtap[(long)(ni * 16 + lm) * COp + bsel]);
-          bw[ni] = gin ? t : bf8_zero();
+          for (int ni = 0; ni < NW; ++ni) {
+            int ci = ni * 16 + lm;
+            int chst = chlog ^ ((ci >> 1) & 7);
+            bw[ni] = *reinterpret_cast<const bf16x8*>(
+                &panel[ci * 64 + chst * 8]);
+          }
+        } else {
+#pragma unroll
+          for (int ni = 0; ni < NW; ++ni) {
+            bf16x8 t = *reinterpret_cast<const bf16x8*>(
+                &wtap[(long)(ni * 16 + lm) * COp + bsel]);
+            bw[ni] = gin ? t : bf8_zero();
+          }
         }
         bf16x8 av[MW];
 #pragma unroll
@@ -2233,6 +2353,10 @@ __global__ __launch_bounds__(256) void conv_tiled_dgrad_s2_bf16_k(
 #pragma unroll
           for (int ni = 0; ni < NW; ++ni)
             acc[mi][ni] = mfma16x16x32bf(av[mi], bw[ni], acc[mi][ni]);
+      }
+      if (BGLDS) {
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
       }
     }
   }
