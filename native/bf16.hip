@@ -1887,6 +1887,9 @@ __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
       }
     };
     if (BGLDS) {
+      // 3-deep ring: panels for taps t+1 and t+2 stay in flight across
+      // the barrier (counted vmcnt at the tail, guide T3/T4 — never
+      // drain to 0 in the main loop)
       stage_B(0, 0);
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       __builtin_amdgcn_s_barrier();
@@ -1982,10 +1985,9 @@ __global__ __launch_bounds__(256) void conv_tiled_s1_bf16_k(
         }
       }
       if (BGLDS) {
-        // raw barrier + explicit drain: __syncthreads() would emit
-        // vmcnt(0) anyway with a glds in flight, but hipcc must not
-        // reorder it; after this every wave's next-tap panel piece has
-        // landed for every consumer
+        // depth 1: drain + raw barrier per tap (a 3-deep ring with
+        // counted vmcnt measured WORSE — 12.98 vs 12.53 ms on the b256
+        // step; the next tap's panel has the whole MFMA block to fly)
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         __builtin_amdgcn_s_barrier();
       }
