@@ -102,14 +102,16 @@ __global__ __launch_bounds__(256) void bn_finalize_fold_k(
     const float* __restrict__ gamma = nullptr,
     const float* __restrict__ beta = nullptr,
     float* __restrict__ aff = nullptr) {
+  // ONE channel per 256-lane block (round 2: the 4-channels-per-block form
+  // ran on <=128 blocks — chip underfilled, 15.7 us/call at trivial work).
   // aff != nullptr: additionally emit the per-channel affine the BN-apply
   // FUSION consumes (aff[c] = gamma*invstd, aff[Cp+c] = beta - mean*scale;
   // pads -> 0 so fused staging zeroes padded channels) — no extra launch.
-  int c = blockIdx.x * 4 + (threadIdx.x >> 6);
+  int c = blockIdx.x;
+  int t = threadIdx.x;
   if (c >= Cp) return;
-  int l = threadIdx.x & 63;
   if (c >= C) {  // zero the channel pads (downstream kernels multiply by them)
-    if (l == 0) {
+    if (t == 0) {
       mean[c] = 0.f;
       invstd[c] = 0.f;
       if (aff != nullptr) {
@@ -121,7 +123,7 @@ __global__ __launch_bounds__(256) void bn_finalize_fold_k(
   }
   long E = (long)Cp * 2;
   float s = 0.f, ss = 0.f;
-  for (int sp = l; sp < S; sp += 64) {
+  for (int sp = t; sp < S; sp += 256) {
     const float* p = ws + (long)sp * E + (long)c * 2;
     s += p[0];
     ss += p[1];
@@ -131,7 +133,15 @@ __global__ __launch_bounds__(256) void bn_finalize_fold_k(
     s += __shfl_xor(s, d);
     ss += __shfl_xor(ss, d);
   }
-  if (l != 0) return;
+  __shared__ float red[4][2];
+  if ((t & 63) == 0) {
+    red[t >> 6][0] = s;
+    red[t >> 6][1] = ss;
+  }
+  __syncthreads();
+  if (t != 0) return;
+  s = red[0][0] + red[1][0] + red[2][0] + red[3][0];
+  ss = red[0][1] + red[1][1] + red[2][1] + red[3][1];
   float m = s / (float)npix;
   float var = fmaxf(ss / (float)npix - m * m, 0.f);
   float is = rsqrtf(var + eps);
@@ -153,15 +163,17 @@ __global__ __launch_bounds__(256) void bn_finalize_fold_k(
 // and ALSO writes dbeta/dgamma densely (their previous extraction from
 // red[Cp][2] cost two strided-copy kernels per BN layer).
 __global__ __launch_bounds__(256) void bn_bwd_fold_k(
-    const float* __restrict__ ws, float* __restrict__ red,
+    const float* __restrict__ ws, float* __restrict__ red_out,
     float* __restrict__ dbeta, float* __restrict__ dgamma, int C, int Cp,
     int S) {
-  int c = blockIdx.x * 4 + (threadIdx.x >> 6);
+  // one channel per 256-lane block (see bn_finalize_fold_k); also writes
+  // dbeta/dgamma densely
+  int c = blockIdx.x;
+  int t = threadIdx.x;
   if (c >= Cp) return;
-  int l = threadIdx.x & 63;
   long E = (long)Cp * 2;
   float s0 = 0.f, s1 = 0.f;
-  for (int sp = l; sp < S; sp += 64) {
+  for (int sp = t; sp < S; sp += 256) {
     const float* p = ws + (long)sp * E + (long)c * 2;
     s0 += p[0];
     s1 += p[1];
@@ -171,9 +183,17 @@ __global__ __launch_bounds__(256) void bn_bwd_fold_k(
     s0 += __shfl_xor(s0, d);
     s1 += __shfl_xor(s1, d);
   }
-  if (l != 0) return;
-  red[c * 2] = s0;
-  red[c * 2 + 1] = s1;
+  __shared__ float red[4][2];
+  if ((t & 63) == 0) {
+    red[t >> 6][0] = s0;
+    red[t >> 6][1] = s1;
+  }
+  __syncthreads();
+  if (t != 0) return;
+  s0 = red[0][0] + red[1][0] + red[2][0] + red[3][0];
+  s1 = red[0][1] + red[1][1] + red[2][1] + red[3][1];
+  red_out[c * 2] = s0;
+  red_out[c * 2 + 1] = s1;
   if (c < C) {
     dbeta[c] = s0;
     dgamma[c] = s1;
