@@ -16,18 +16,18 @@ class _BatchNormBF16Fn(torch.autograd.Function):
         y, mean, invstd = native.ext().bn_fwd_bf16(
             x.contiguous(), gamma.contiguous(), beta.contiguous(),
             running_mean, running_var, bool(training), momentum, eps, relu)
-        ctx.save_for_backward(x, gamma, mean, invstd, y if relu else None)
+        ctx.save_for_backward(x, gamma, beta, mean, invstd)
         ctx.relu = relu
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        x, gamma, mean, invstd, yrelu = ctx.saved_tensors
+        x, gamma, beta, mean, invstd = ctx.saved_tensors
         dy = dy.contiguous()
-        # the relu gate (yact) folds into both bn backward passes
+        # relu gate is RECOMPUTED from (x, affine) — bitwise the forward's
+        # yact>0 (bn_bwd_bf16 docstring) — so y is never saved or re-read
         dx, dgamma, dbeta = native.ext().bn_bwd_bf16(
-            dy, x, mean, invstd, gamma,
-            yrelu if ctx.relu else torch.Tensor())
+            dy, x, mean, invstd, gamma, beta, ctx.relu)
         return dx, dgamma, dbeta, None, None, None, None, None, None
 
 
