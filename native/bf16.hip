@@ -789,7 +789,8 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_rs_bf16_k(
   const int Wi = S * OW;   // input width (host guarantees W == S*OW)
   const int W2 = Wi + 2;
   // region capacity: worst-case rows*(Wi+2) over the routed OW range
-  constexpr int PX = (S == 2) ? 528 : (MC <= 64 ? 296 : MC <= 128 ? 352 : 464);
+  constexpr int PX = (S == 2) ? (MC <= 64 ? 528 : 640)
+                             : (MC <= 64 ? 296 : MC <= 128 ? 352 : 464);
 
   __shared__ bf16 lds_reg[PX * 32];           // [px][32ci]
   // dy planes then A-image slots; +16-elem stagger per plane so the build's

@@ -422,63 +422,84 @@ __global__ void maxpool3x3s2_fwd_k(const bf16* __restrict__ x,
   }
 }
 
+template <int PH>
 __global__ void maxpool3x3s2_bwd_k(const bf16* __restrict__ dy,
                                    const unsigned char* __restrict__ idx,
                                    bf16* __restrict__ dx, int N, int H, int W,
                                    int Cp, int OH, int OW) {
-  long total = (long)N * H * W * (Cp / 8);
+  // PAIRED-COLUMN gather (k3/s2/p1). A thread owns the adjacent input pixel
+  // pair (iw=2k, 2k+1) of one row: the pair's column candidates collapse to
+  // ow=k (serves both: kw=1 for the even pixel, kw=2 for the odd) and
+  // ow=k+1 (kw=0, odd pixel only) — 2 window loads produce 2 outputs, and
+  // the two 16-byte stores are CONTIGUOUS (a stride-2 per-pixel parity
+  // split was measured write-bound: every 128 B line half-covered).
+  // Rows keep the parity split (even ih: 1 candidate row, odd: 2), so the
+  // avg load count is 1.5/pixel vs 4 for the flat 2x2 set. Lower bounds
+  // never trip; ow=k+1 and the odd class's second row are hardware-bounds-
+  // checked buffer loads (OOB reads 0 and idx 0 pairs with want!=0 rows).
+  int Hc = (H - PH + 1) / 2, W2 = (W + 1) / 2;
+  long total = (long)N * Hc * W2 * (Cp / 8);
+  const long dy_elems = (long)N * OH * OW * Cp;
+  const bool use_buf = rsrc_fits(dy_elems);
+  buf_rsrc_t dyr = make_bf16_rsrc(dy, dy_elems);
+  buf_rsrc_t idxr = make_u8_rsrc(idx, dy_elems);
   for (long t = (long)blockIdx.x * blockDim.x + threadIdx.x; t < total;
        t += (long)gridDim.x * blockDim.x) {
     int cg = t % (Cp / 8);
     long r = t / (Cp / 8);
-    int iw = r % W;
-    long r2 = r / W;
-    int ih = r2 % H;
-    int n = r2 / H;
-    float acc[8] = {};
-    // windows (oh,ow) with oh*2-1 <= ih <= oh*2+1 => oh in [(ih-1)/2, (ih+1)/2].
-    // FLATTENED to a fixed unrolled 2x2 candidate set with hardware-bounds-
-    // checked loads: the runtime-bounded loop form put every dy/idx load
-    // behind divergent branches (uncountable -> vmcnt(0) drains, see
-    // buf_load8). Dead candidates read 0 and (a8==0, want==0) pairs add a
-    // zero g — harmless.
-    const long dy_elems = (long)N * OH * OW * Cp;
-    const bool use_buf = rsrc_fits(dy_elems);
-    buf_rsrc_t dyr = make_bf16_rsrc(dy, dy_elems);
-    buf_rsrc_t idxr = make_u8_rsrc(idx, dy_elems);
-    int oh0 = (ih - 1) / 2, ow0 = (iw - 1) / 2;
+    int k = r % W2;
+    long r2 = r / W2;
+    int ihc = r2 % Hc;
+    int n = r2 / Hc;
+    int ih = 2 * ihc + PH;
+    float acce[8] = {}, acco[8] = {};
 #pragma unroll
-    for (int a = 0; a < 2; ++a) {
-#pragma unroll
-      for (int b = 0; b < 2; ++b) {
-        int oh = oh0 + a, ow = ow0 + b;
-        int kh = ih - (oh * 2 - 1), kw = iw - (ow * 2 - 1);
-        bool ok = oh >= 0 && oh < OH && ow >= 0 && ow < OW &&
-                  kh >= 0 && kh <= 2 && kw >= 0 && kw <= 2;
-        long off = (((long)(n * OH + oh) * OW + ow) * Cp) + cg * 8;
-        unsigned long long a8;
-        bf16x8 g;
-        if (use_buf) {
-          a8 = buf_load_u64(idxr, off, ok);
-          g = buf_load8(dyr, off, ok);
-        } else {
-          a8 = 0;
-          g = bf8_zero();
-          if (ok) {
-            a8 = *reinterpret_cast<const unsigned long long*>(&idx[off]);
-            g = *reinterpret_cast<const bf16x8*>(&dy[off]);
-          }
+    for (int a = 0; a < (PH ? 2 : 1); ++a) {
+      int oh = PH ? (ih - 1) / 2 + a : ih / 2;
+      int kh = ih - (oh * 2 - 1);
+      bool okr = (a == 0 || oh < OH);
+      long off0 = (((long)(n * OH + oh) * OW + k) * Cp) + cg * 8;
+      unsigned long long a0, a1;
+      bf16x8 g0, g1;
+      bool ok1 = okr && (k + 1 < OW);
+      if (use_buf) {
+        a0 = buf_load_u64(idxr, off0, okr);
+        g0 = buf_load8(dyr, off0, okr);
+        a1 = buf_load_u64(idxr, off0 + Cp, ok1);
+        g1 = buf_load8(dyr, off0 + Cp, ok1);
+      } else {
+        a0 = a1 = 0;
+        g0 = g1 = bf8_zero();
+        if (okr) {
+          a0 = *reinterpret_cast<const unsigned long long*>(&idx[off0]);
+          g0 = *reinterpret_cast<const bf16x8*>(&dy[off0]);
         }
-        unsigned char want = (unsigned char)(kh * 3 + kw);
+        if (ok1) {
+          a1 = *reinterpret_cast<const unsigned long long*>(&idx[off0 + Cp]);
+          g1 = *reinterpret_cast<const bf16x8*>(&dy[off0 + Cp]);
+        }
+      }
+      unsigned char we = (unsigned char)(kh * 3 + 1);   // even pixel: kw=1@k
+      unsigned char wo0 = (unsigned char)(kh * 3 + 2);  // odd: kw=2@k
+      unsigned char wo1 = (unsigned char)(kh * 3 + 0);  // odd: kw=0@k+1
 #pragma unroll
-        for (int j = 0; j < 8; ++j)
-          if (((a8 >> (8 * j)) & 0xff) == want) acc[j] += to_f32(g[j]);
+      for (int j = 0; j < 8; ++j) {
+        unsigned char b0 = (a0 >> (8 * j)) & 0xff;
+        float f0 = to_f32(g0[j]);
+        if (b0 == we) acce[j] += f0;
+        if (b0 == wo0) acco[j] += f0;
+        if (((a1 >> (8 * j)) & 0xff) == wo1) acco[j] += to_f32(g1[j]);
       }
     }
-    bf16x8 o;
+    bf16x8 oe, oo;
 #pragma unroll
-    for (int j = 0; j < 8; ++j) o[j] = to_bf16(acc[j]);
-    *reinterpret_cast<bf16x8*>(&dx[((long)(n * H + ih) * W + iw) * Cp + cg * 8]) = o;
+    for (int j = 0; j < 8; ++j) {
+      oe[j] = to_bf16(acce[j]);
+      oo[j] = to_bf16(acco[j]);
+    }
+    long doff = ((long)(n * H + ih) * W + 2 * k) * Cp + cg * 8;
+    *reinterpret_cast<bf16x8*>(&dx[doff]) = oe;
+    if (2 * k + 1 < W) *reinterpret_cast<bf16x8*>(&dx[doff + Cp]) = oo;
   }
 }
 
