@@ -97,4 +97,9 @@ class SGD(torch.optim.Optimizer):
                     else:
                         p.add_(p.grad, alpha=-group["lr"])
                     p.grad.zero_()  # same consume-and-reset as the GPU kernel
+        if torch.cuda.is_available() and torch.cuda.is_initialized():
+            # one batched launch re-packs every cached conv-weight buffer the
+            # update just invalidated (native/ext_pack.inc); no-op when the
+            # model has no packed weights (e.g. LeNet's smallK engines)
+            native.ext().repack_refresh_all()
         return loss

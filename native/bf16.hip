@@ -2549,6 +2549,78 @@ __global__ void repack_w_stem_k(const float* __restrict__ w,
 }
 
 // ---------------------------------------------------------------------------
+// Batched weight repack: ONE launch refreshes every cached packed-weight
+// buffer after the optimizer step (grid.y = cache entry). Replaces ~2 small
+// repack launches per conv per step (mostly launch tail at these extents)
+// with a single kernel. Kinds mirror repack_w_k / repack_w_stem_k /
+// repack_w_dgrad_s2_k above — same indexing, same zero-padding rules.
+// ---------------------------------------------------------------------------
+struct PackDesc {
+  const float* w;
+  bf16* rp;
+  long extent;
+  long base;  // prefix offset in the flattened batched-refresh index space
+  int kind;  // 0 = repack_w (fwd/dgrad s1), 1 = stem, 2 = dgrad_s2 parity
+  int CO, CI, KH, KW, Ap, Bp, fp;  // fp: kind0 dgrad flag / kind2 pad
+};
+
+__global__ void repack_all_k(const PackDesc* __restrict__ ds, int n,
+                             long total) {
+  // flattened 1:1 thread<->element over the SUM of entry extents (a
+  // blocks-per-entry grid.y layout measured 200 us: the per-entry block cap
+  // starved the big entries while small entries idled). Entry found by
+  // binary search on the prefix offsets — 6 scalar-ish steps, block-uniform
+  // except at entry seams.
+  // bases are 256-aligned (host pads the flattened space), so a BLOCK maps
+  // to exactly one entry: search on the block-uniform offset -> the whole
+  // lookup compiles to scalar loads, one chain per wave not per lane
+  long i0 = (long)blockIdx.x * blockDim.x;
+  if (i0 >= total) return;
+  int lo = 0, hi = n - 1;
+  while (lo < hi) {
+    int mid = (lo + hi + 1) >> 1;
+    if (ds[mid].base <= i0) lo = mid;
+    else hi = mid - 1;
+  }
+  const PackDesc d = ds[lo];
+  long j = i0 + threadIdx.x - d.base;
+  if (j >= d.extent) return;  // alignment gap past the entry tail
+  bf16 v = to_bf16(0.f);
+  if (d.kind == 0) {
+    int KHKW = d.KH * d.KW;
+    int b = j % d.Bp;
+    long r = j / d.Bp;
+    int a = r % d.Ap;
+    int tap_d = r / d.Ap;
+    int co = d.fp ? b : a, ci = d.fp ? a : b;
+    int tap = d.fp ? KHKW - 1 - tap_d : tap_d;
+    if (co < d.CO && ci < d.CI)
+      v = to_bf16(d.w[((long)co * d.CI + ci) * KHKW + tap]);
+  } else if (d.kind == 1) {
+    int kwci = j & 31;
+    long r = j >> 5;
+    int co = r % d.Ap;
+    int kh = r / d.Ap;
+    int kw = kwci >> 2, ci = kwci & 3;
+    if (co < d.CO && ci < d.CI && kw < d.KW)
+      v = to_bf16(d.w[(((long)co * d.CI + ci) * d.KH + kh) * d.KW + kw]);
+  } else {
+    int co = j % d.Bp;
+    long r = j / d.Bp;
+    int ci = r % d.Ap;
+    long r2 = r / d.Ap;
+    int tap = r2 & 3;
+    int cls = r2 >> 2;
+    int kh = (cls >> 1) + d.fp - 2 * (tap >> 1);
+    int kw = (cls & 1) + d.fp - 2 * (tap & 1);
+    if (kh >= 0 && kh < d.KH && kw >= 0 && kw < d.KW && co < d.CO &&
+        ci < d.CI)
+      v = to_bf16(d.w[(((long)co * d.CI + ci) * d.KH + kh) * d.KW + kw]);
+  }
+  d.rp[j] = v;
+}
+
+// ---------------------------------------------------------------------------
 // Wide-class LogSoftmax+NLL (ResNet: C=1000): one WAVE per row, lanes split
 // the class axis, shfl tree reductions. (The thread-per-row kernel above is
 // for tiny C like LeNet's 10.)

@@ -548,6 +548,7 @@ void fused_sgd_step(at::Tensor ptrs, at::Tensor chunks, long chunk_elems,
                      (int)chunk_elems, (float)lr, (float)momentum);
 }
 
+#include "ext_pack.inc"
 #include "ext_bf16.inc"
 #include "ext_resnet.inc"
 #include "ext_comm.inc"
@@ -577,6 +578,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_probe", &mfma_probe);
   m.def("tr16_probe", &tr16_probe);
   m.def("conv2d_fwd_bf16", &conv2d_fwd_bf16);
+  m.def("repack_refresh_all", &repack_refresh_all);
+  m.def("repack_cache_size", &repack_cache_size);
+  m.def("repack_cache_clear", &repack_cache_clear);
   m.def("conv_tiled_s1_bf16", &conv_tiled_s1_bf16);
   m.def("conv2d_dgrad_bf16", &conv2d_dgrad_bf16);
   m.def("conv2d_wgrad_bf16", &conv2d_wgrad_bf16);
