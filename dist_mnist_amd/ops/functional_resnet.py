@@ -54,6 +54,9 @@ class _BnVirtualFn(torch.autograd.Function):
         ctx.save_for_backward(x, gamma, beta, mean, invstd)
         ctx.relu = relu
         ctx.mark_non_differentiable(aff)
+        # no zero-tensor materialization for d_aff (it is ignored below) —
+        # saves a fill launch per BN backward inside the captured step
+        ctx.set_materialize_grads(False)
         return x.view_as(x), aff
 
     @staticmethod
