@@ -55,3 +55,21 @@ def test_prefetcher_cpu_passthrough():
     x, y = next(pf)
     assert x.shape == (4, 1, 28, 28)
     assert y.shape == (4,)
+
+
+def test_synthetic_stream_rgb224_layout():
+    """ResNet bench input (BASELINE config 5): NHWC bf16 [N,224,224,8] with
+    RGB in channels 0-2 and the 5 pad channels exactly zero."""
+    from dist_mnist_amd.data import SyntheticBatchStream
+
+    s1 = SyntheticBatchStream(batch_size=4, n_pool=2, seed=7, pin=False,
+                              layout="rgb224_bf16", n_classes=1000)
+    s2 = SyntheticBatchStream(batch_size=4, n_pool=2, seed=7, pin=False,
+                              layout="rgb224_bf16", n_classes=1000)
+    x1, y1 = next(iter(s1))
+    x2, y2 = next(iter(s2))
+    assert x1.shape == (4, 224, 224, 8) and x1.dtype == torch.bfloat16
+    assert torch.equal(x1, x2) and torch.equal(y1, y2)
+    assert torch.all(x1[..., 3:] == 0)
+    assert x1[..., :3].float().abs().sum() > 0
+    assert y1.dtype == torch.int64 and y1.max() < 1000 and y1.min() >= 0
