@@ -61,6 +61,46 @@ def _worker_bucketed_vs_reference(rank, world, port, q):
         dist.destroy_process_group()
 
 
+def _worker_bucket_size_edges(rank, world, port, q):
+    """bucket_bytes=1 (every param its own bucket) and 1 GiB (one bucket
+    for the whole model) must both reproduce reference gradients — the
+    bucket assembly edge cases."""
+    from dist_mnist_amd.parallel import DistributedDataParallel, average_gradients
+
+    _init(rank, world, port)
+    try:
+        torch.manual_seed(300)
+        base = Net().eval()
+        g = torch.Generator().manual_seed(3000 + rank)
+        x = torch.randn(8, 1, 28, 28, generator=g)
+        t = torch.randint(0, 10, (8,), generator=g)
+
+        ref = Net().eval()
+        ref.load_state_dict(base.state_dict())
+        _, loss = ref(x, t)
+        loss.backward()
+        average_gradients(ref)
+        ref_grads = {k: p.grad.clone() for k, p in ref.named_parameters()}
+
+        for bb in (1, 1 << 30):
+            ddp = DistributedDataParallel(Net().eval(), broadcast_params=False,
+                                          bucket_bytes=bb)
+            ddp.module.load_state_dict(base.state_dict())
+            ddp.zero_grad()
+            _, loss = ddp(x, t)
+            loss.backward()
+            ddp.sync()
+            for k, p in ddp.module.named_parameters():
+                torch.testing.assert_close(
+                    p.grad, ref_grads[k], rtol=1e-6, atol=1e-7,
+                    msg=f"bucket_bytes={bb} != reference: {k}")
+        q.put((rank, "ok"))
+    except Exception as e:  # pragma: no cover
+        q.put((rank, f"FAIL: {type(e).__name__}: {e}"))
+    finally:
+        dist.destroy_process_group()
+
+
 def _worker_dp_equals_large_batch(rank, world, port, q):
     from dist_mnist_amd.ops import SGD
     from dist_mnist_amd.parallel import DistributedDataParallel
@@ -161,6 +201,11 @@ def _worker_resync_after_fallback(rank, world, port, q):
 @pytest.mark.timeout(180)
 def test_bucketed_matches_reference_semantics(free_port):
     _run_world(_worker_bucketed_vs_reference, 2, free_port)
+
+
+@pytest.mark.timeout(180)
+def test_bucket_size_edges_match_reference(free_port):
+    _run_world(_worker_bucket_size_edges, 2, free_port)
 
 
 @pytest.mark.timeout(180)
