@@ -94,7 +94,11 @@ class DistributedDataParallel(nn.Module):
         size = 0
         for p in params:
             nbytes = p.numel() * p.element_size()
-            if group and size + nbytes > bucket_bytes:
+            # a bucket is one flat buffer: close it on size, and on any
+            # dtype/device boundary (mixed-precision param lists)
+            if group and (size + nbytes > bucket_bytes
+                          or p.dtype != group[0].dtype
+                          or p.device != group[0].device):
                 self._finish_bucket(group)
                 group, size = [], 0
             group.append(p)

@@ -101,6 +101,53 @@ def _worker_bucket_size_edges(rank, world, port, q):
         dist.destroy_process_group()
 
 
+def _worker_mixed_dtype_buckets(rank, world, port, q):
+    """Params with different dtypes must land in separate buckets (one flat
+    buffer per bucket); gradient averaging still matches the reference."""
+    import torch.nn as nn
+    from dist_mnist_amd.parallel import DistributedDataParallel
+
+    _init(rank, world, port)
+    try:
+        torch.manual_seed(400)
+
+        class Mixed(nn.Module):
+            def __init__(self):
+                super().__init__()
+                self.a = nn.Linear(8, 8)                      # fp32
+                self.b = nn.Linear(8, 4).to(torch.bfloat16)   # bf16
+
+            def forward(self, x):
+                return self.b(self.a(x).to(torch.bfloat16)).float().sum()
+
+        ddp = DistributedDataParallel(Mixed(), broadcast_params=True)
+        dtypes = {b.flat.dtype for b in ddp.buckets}
+        assert dtypes == {torch.float32, torch.bfloat16}, dtypes
+        for b in ddp.buckets:
+            assert all(p.dtype == b.flat.dtype for p in b.params)
+
+        g = torch.Generator().manual_seed(4000 + rank)
+        x = torch.randn(4, 8, generator=g)
+        ddp.zero_grad()
+        ddp(x).backward()
+        ddp.sync()
+        # oracle: same model, reference per-param averaging
+        ref = Mixed()
+        ref.load_state_dict(ddp.module.state_dict())
+        ref(x).backward()
+        from dist_mnist_amd.parallel import average_gradients
+        average_gradients(ref)
+        for (k, p), (_, pr) in zip(ddp.module.named_parameters(),
+                                   ref.named_parameters()):
+            torch.testing.assert_close(p.grad, pr.grad, rtol=1e-2, atol=1e-3,
+                                       msg=k)
+        q.put((rank, "ok"))
+    except Exception as e:  # pragma: no cover
+        q.put((rank, f"FAIL: {type(e).__name__}: {e}"))
+    finally:
+        dist.destroy_process_group()
+
+
 def _worker_dp_equals_large_batch(rank, world, port, q):
     from dist_mnist_amd.ops import SGD
     from dist_mnist_amd.parallel import DistributedDataParallel
@@ -206,6 +253,11 @@ def test_bucketed_matches_reference_semantics(free_port):
 @pytest.mark.timeout(180)
 def test_bucket_size_edges_match_reference(free_port):
     _run_world(_worker_bucket_size_edges, 2, free_port)
+
+
+@pytest.mark.timeout(180)
+def test_mixed_dtype_params_bucket_separately(free_port):
+    _run_world(_worker_mixed_dtype_buckets, 2, free_port)
 
 
 @pytest.mark.timeout(180)
