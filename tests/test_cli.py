@@ -54,6 +54,37 @@ def test_cli_two_rank_gloo(tmp_path, free_port):
 
 
 @pytest.mark.timeout(300)
+def test_cli_ddp_reference_mode_matches_bucketed(tmp_path, free_port):
+    """--ddp reference (the literal per-param all-reduce) must
+    train to the same loss as the default bucketed engine — same seed, same
+    shards, same SUM/world semantics."""
+    def run_mode(mode, port):
+        env_base = dict(os.environ, MASTER_ADDR="127.0.0.1",
+                        MASTER_PORT=str(port), WORLD_SIZE="2")
+        procs = []
+        for rank in range(2):
+            env = dict(env_base, RANK=str(rank))
+            procs.append(subprocess.Popen(
+                [sys.executable, os.path.join(REPO, "dist_mnist.py"),
+                 "--backend", "gloo", "--epochs", "1", "--batch-size", "32",
+                 "--dataset-size", "64", "--no-cuda", "--seed", "5",
+                 "--ddp", mode],
+                env=env, stdout=subprocess.PIPE, stderr=subprocess.PIPE,
+                text=True, cwd=REPO))
+        outs = [p.communicate(timeout=280) for p in procs]
+        for i, p in enumerate(procs):
+            assert p.returncode == 0, outs[i][1][-2000:]
+        m = re.search(r"Rank\s+0\s*,\s*epoch\s+0\s*:\s*([0-9.]+)",
+                      outs[0][0])
+        assert m, outs[0][0]
+        return float(m.group(1))
+
+    l_buck = run_mode("bucketed", free_port)
+    l_ref = run_mode("reference", free_port + 1)
+    assert abs(l_buck - l_ref) <= 1e-6 * max(1.0, abs(l_buck)), (l_buck, l_ref)
+
+
+@pytest.mark.timeout(300)
 def test_cli_resume(tmp_path, free_port):
     """--resume loads a prior checkpoint and continues (framework extension
     on the reference's save-only surface, SURVEY.md §5 checkpoint/resume)."""
